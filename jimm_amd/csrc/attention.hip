@@ -1,0 +1,235 @@
+// K5 — flash-style attention forward for CDNA4 (gfx950), bf16, head_dim 64.
+//
+// Covers every attention in the model zoo: self-attention
+// L in {50..1024} (ViT/CLIP/SigLIP towers, all with head_dim 64 since
+// heads = width/64 — clip.py:60), causal masking for the CLIP text tower
+// (clip.py:62), and the MAP head's Lq=1 cross-attention (K9).
+//
+// Design (one workgroup = 4 waves = 64 q rows; grid = (ceil(Lq/64), B*H)):
+//   * "swapped" QK^T: S^T = mfma(A=K_tile, B=Q^T) so each lane's softmax
+//     row stats live in-lane + 2 shuffles (no serial-lane softmax — CDNA
+//     guide common-mistake #6);
+//   * K staged row-major in LDS, V staged TRANSPOSED (V^T) so the PV
+//     mfma's B-fragment is a contiguous ds_read_b128;
+//   * P routed through a per-wave LDS tile (padded rows, +8 shorts) to
+//     re-shape from the S^T C-layout into the PV A-fragment layout;
+//   * online softmax with running (m, l) per q row; lse = m + log(l)
+//     saved for the backward (recompute) pass.
+//
+// MFMA: v_mfma_f32_16x16x32_bf16. Fragment layouts (verified on hardware by
+// csrc/probe.hip + tests/test_kernels_gpu.py::test_mfma_probe):
+//   A[i][k]: lane l holds A[l&15][(l>>4)*8 + j], j=0..7
+//   B[k][j]: lane l holds B[(l>>4)*8 + j][l&15]
+//   C[i][j]: lane l holds rows (l>>4)*4+r (r=0..3), col l&15
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+#define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
+
+constexpr int QBLK = 64;    // q rows per workgroup (16 per wave)
+constexpr int KVBLK = 64;   // keys per LDS tile
+constexpr int D = 64;       // head_dim (checked host-side)
+constexpr int LDS_PITCH = D + 8;  // +8 shorts: bank-conflict pad for b128 reads
+
+template <bool CAUSAL>
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale) {
+  // LDS carve: K [64][72], V^T [64][72], P per-wave 4x[16][72] (shorts)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = reinterpret_cast<short*>(smem);                       // 64*72
+  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // 64*72
+  short* p_lds = vt_lds + D * LDS_PITCH;                               // 4*16*72
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lo = lane & 15;   // column index inside a 16-wide fragment
+  const int hi = lane >> 4;   // 0..3
+  const int64_t bh = blockIdx.y;
+  const int q0 = blockIdx.x * QBLK + wave * 16;  // this wave's first q row
+
+  const bf16* qp = q + bh * Lq * D;
+  const bf16* kp = k + bh * Lk * D;
+  const bf16* vp = v + bh * Lk * D;
+  bf16* op = o + bh * Lq * D;
+
+  // ---- load Q fragments (B-operand of the swapped QK^T) -------------------
+  // B[d][q] = Q[q0+lo][d = 32*s + hi*8 + j]
+  bf16x8_t qb[2];
+  {
+    const int qrow = min(q0 + lo, Lq - 1);
+#pragma unroll
+    for (int s = 0; s < 2; ++s)
+      qb[s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * D + 32 * s + hi * 8);
+  }
+
+  short* my_p = p_lds + wave * 16 * LDS_PITCH;
+
+  f32x4_t acc_o[4] = {};  // O tile: rows q = hi*4+r, cols d = 16*dt + lo
+  float m_run = -INFINITY;
+  float l_run = 0.f;
+
+  const int kv_end = CAUSAL ? min(Lk, blockIdx.x * QBLK + QBLK) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
+    // ---- stage K tile (row-major) and V tile (transposed) ----------------
+    {
+      const int row = tid / 4;          // 0..63 (key within tile)
+      const int c0 = (tid % 4) * 16;    // 16 shorts per thread
+      const int key = kv0 + row;
+      if (key < Lk) {
+        const bf16x8_t k0 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * D + c0);
+        const bf16x8_t k1 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * D + c0 + 8);
+        *reinterpret_cast<bf16x8_t*>(k_lds + row * LDS_PITCH + c0) = k0;
+        *reinterpret_cast<bf16x8_t*>(k_lds + row * LDS_PITCH + c0 + 8) = k1;
+        const bf16x8_t v0 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * D + c0);
+        const bf16x8_t v1 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * D + c0 + 8);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) vt_lds[(c0 + i) * LDS_PITCH + row] = v0[i];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) vt_lds[(c0 + 8 + i) * LDS_PITCH + row] = v1[i];
+      } else {
+        // zero-fill so mfma on the padded tail contributes nothing
+        for (int i = 0; i < 16; ++i) k_lds[row * LDS_PITCH + c0 + i] = 0;
+        for (int i = 0; i < 16; ++i) vt_lds[(c0 + i) * LDS_PITCH + row] = 0;
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K . Q^T : 4 key tiles x 2 d-steps -------------------------
+    f32x4_t sc[4] = {};
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const bf16x8_t ka =
+            *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * t + lo) * LDS_PITCH + 32 * s + hi * 8);
+        sc[t] = MFMA16(ka, qb[s], sc[t]);
+      }
+    }
+
+    // ---- masked, scaled scores; per-row (q = lo) online softmax ----------
+    float sv[16];
+    const int q_idx = q0 + lo;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = kv0 + 16 * t + hi * 4 + r;
+        float x = sc[t][r] * scale;
+        if (key >= Lk || (CAUSAL && key > q_idx)) x = -INFINITY;
+        sv[4 * t + r] = x;
+      }
+    }
+    float mt = sv[0];
+#pragma unroll
+    for (int i = 1; i < 16; ++i) mt = fmaxf(mt, sv[i]);
+    mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE));
+    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE));
+
+    const float m_new = fmaxf(m_run, mt);
+    // alpha=0 on the first tile (m_run = -inf) starts O from zero
+    const float alpha = (m_new == -INFINITY) ? 0.f : __expf(m_run - m_new);
+    m_run = m_new;
+
+    float psum = 0.f;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) {
+      const float p = (sv[i] == -INFINITY) ? 0.f : __expf(sv[i] - m_new);
+      sv[i] = p;
+      psum += p;
+    }
+    psum += __shfl_xor(psum, 16, WAVE);
+    psum += __shfl_xor(psum, 32, WAVE);
+    l_run = l_run * alpha + psum;
+
+    // ---- write P (bf16) into this wave's LDS tile: [q=lo][k=16t+hi*4+r] --
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      bf16x4 pk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) pk[r] = f2bfs(sv[4 * t + r]);
+      *reinterpret_cast<bf16x4*>(my_p + lo * LDS_PITCH + 16 * t + hi * 4) = pk;
+    }
+
+    // ---- rescale O accumulators (rows q = hi*4+r need alpha from lane q) -
+    float alpha_r[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) alpha_r[r] = __shfl(alpha, hi * 4 + r, WAVE);
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[dt][r] *= alpha_r[r];
+
+    // ---- O += P . V : A = P (from LDS), B = V^T reads --------------------
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      const bf16x8_t pa =
+          *reinterpret_cast<const bf16x8_t*>(my_p + lo * LDS_PITCH + 32 * s + hi * 8);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const bf16x8_t vb =
+            *reinterpret_cast<const bf16x8_t*>(vt_lds + (16 * dt + lo) * LDS_PITCH + 32 * s + hi * 8);
+        acc_o[dt] = MFMA16(pa, vb, acc_o[dt]);
+      }
+    }
+    __syncthreads();  // K/V^T tiles reused next iteration
+  }
+
+  // ---- epilogue: O /= l, store O and lse ----------------------------------
+  const float invl = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  float invl_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) invl_r[r] = __shfl(invl, hi * 4 + r, WAVE);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + hi * 4 + r;
+    if (qrow >= Lq) continue;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      op[(int64_t)qrow * D + 16 * dt + lo] = f2bf(acc_o[dt][r] * invl_r[r]);
+  }
+  if (hi == 0 && q0 + lo < Lq && wave * 16 + lo < QBLK)
+    lse[bh * Lq + q0 + lo] = m_run + __logf(l_run);
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    bool causal, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only, got ", q.scalar_type());
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "attn_fwd: (B,H,L,64) expected");
+  const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
+  TORCH_CHECK(k.size(0) == B && k.size(1) == H && v.size(2) == Lk);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
+  const dim3 grid((Lq + QBLK - 1) / QBLK, (unsigned)((int64_t)B * H));
+  const size_t shmem = (KVBLK * LDS_PITCH + D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (causal) {
+    hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, dim3(256), shmem, stream,
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(), Lq, Lk,
+                       (float)scale);
+  } else {
+    hipLaunchKernelGGL((attn_fwd_kernel<false>), grid, dim3(256), shmem, stream,
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(), Lq, Lk,
+                       (float)scale);
+  }
+  return {o, lse};
+}

@@ -1,0 +1,37 @@
+// Python bindings for the jimm_amd HIP/CDNA4 extension.
+
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor mean, torch::Tensor rstd);
+torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, std::string act,
+                           c10::optional<torch::Tensor> residual);
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act);
+torch::Tensor im2col_patch(torch::Tensor img, int64_t patch);
+torch::Tensor col2im_patch(torch::Tensor cols, std::vector<int64_t> img_shape, int64_t patch);
+void adam_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
+               std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+               std::vector<c10::optional<torch::Tensor>> masters, double lr, double b1,
+               double b2, double eps, double wd, int64_t step);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    bool causal, double scale);
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype);
+std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
+                                      c10::optional<torch::Tensor> bias, std::string act,
+                                      c10::optional<torch::Tensor> residual, bool save_z);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (K3)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (K3/K15)");
+  m.def("bias_act_fwd", &bias_act_fwd, "fused bias+activation(+residual) forward");
+  m.def("act_bwd", &act_bwd, "activation backward: dy * act'(z)");
+  m.def("im2col_patch", &im2col_patch, "patch-embed unfold (K1)");
+  m.def("col2im_patch", &col2im_patch, "patch-embed fold backward (K1/K15)");
+  m.def("adam_step", &adam_step, "fused multi-tensor Adam (K14)");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward, head_dim 64 (K5)");
+  m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
+  m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
+  m.def("linear_fwd", &linear_fwd, "MFMA GEMM + fused epilogue (K4/K6/K7/K8)");
+}

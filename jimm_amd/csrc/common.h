@@ -1,0 +1,99 @@
+// Shared helpers for jimm_amd CDNA4 (gfx950) kernels.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#define WAVE 64  // CDNA wavefront width (not 32!)
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess) {                                                 \
+      TORCH_CHECK(false, "HIP error: ", hipGetErrorString(_e), " at ",      \
+                  __FILE__, ":", __LINE__);                                 \
+    }                                                                       \
+  } while (0)
+
+// ---- vector types ----------------------------------------------------------
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef short bf16x4 __attribute__((ext_vector_type(4)));
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+typedef __hip_bfloat16 bf16;
+
+__device__ __forceinline__ float bf2f(bf16 x) { return __bfloat162float(x); }
+__device__ __forceinline__ bf16 f2bf(float x) { return __float2bfloat16(x); }
+
+// raw-bits helpers for packed bf16 handled as short
+__device__ __forceinline__ float bfs2f(short s) {
+  unsigned int u = ((unsigned int)(unsigned short)s) << 16;
+  return __uint_as_float(u);
+}
+__device__ __forceinline__ short f2bfs(float f) {
+  bf16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
+// ---- wave reductions -------------------------------------------------------
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
+// ---- activations (shared fwd/bwd definitions) ------------------------------
+// act codes: 0 = none, 1 = exact erf gelu, 2 = tanh gelu, 3 = quickgelu
+#define ACT_NONE 0
+#define ACT_GELU 1
+#define ACT_GELU_TANH 2
+#define ACT_QUICKGELU 3
+
+__device__ __forceinline__ float act_fwd(float x, int act) {
+  switch (act) {
+    case ACT_GELU:
+      return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f));
+    case ACT_GELU_TANH: {
+      float x3 = x * x * x;
+      float t = tanhf(0.7978845608028654f * (x + 0.044715f * x3));
+      return 0.5f * x * (1.0f + t);
+    }
+    case ACT_QUICKGELU: {
+      float s = 1.0f / (1.0f + __expf(-1.702f * x));
+      return x * s;
+    }
+    default:
+      return x;
+  }
+}
+
+__device__ __forceinline__ float act_grad(float x, int act) {
+  switch (act) {
+    case ACT_GELU: {
+      // d/dx [x * Phi(x)] = Phi(x) + x * phi(x)
+      float cdf = 0.5f * (1.0f + erff(x * 0.70710678118654752440f));
+      float pdf = 0.3989422804014327f * __expf(-0.5f * x * x);
+      return cdf + x * pdf;
+    }
+    case ACT_GELU_TANH: {
+      float x2 = x * x;
+      float inner = 0.7978845608028654f * (x + 0.044715f * x * x2);
+      float t = tanhf(inner);
+      float dinner = 0.7978845608028654f * (1.0f + 3.0f * 0.044715f * x2);
+      return 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * dinner;
+    }
+    case ACT_QUICKGELU: {
+      float s = 1.0f / (1.0f + __expf(-1.702f * x));
+      return s + 1.702f * x * s * (1.0f - s);
+    }
+    default:
+      return 1.0f;
+  }
+}

@@ -1,0 +1,202 @@
+// Fused elementwise kernels:
+//   * bias_act_fwd  — z += bias (in place, kept as pre-activation for bwd),
+//                     y = act(z) [+ residual]; acts: gelu / gelu_tanh /
+//                     quickgelu (K7 epilogue; transformer.py:92-103)
+//   * act_bwd       — dz = dy * act'(z)
+//   * im2col/col2im — K1 patch-embed gather for kernel==stride convs
+//                     (common/vit.py:153-165,228-230): pure permute-gather,
+//                     coalesced on the cols side.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int kMaxGrid = 2048;
+
+template <typename T, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE>
+__global__ void bias_act_kernel(T* __restrict__ z, const T* __restrict__ bias,
+                                const T* __restrict__ res, T* __restrict__ y, int64_t n,
+                                int M, int act) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = (float)z[i];
+    if (HAS_BIAS) v += (float)bias[i % M];
+    if (SAVE_PRE) z[i] = (T)v;  // pre-activation saved for act_bwd
+    float o = act_fwd(v, act);
+    if (HAS_RES) o += (float)res[i];
+    y[i] = (T)o;
+  }
+}
+
+template <typename T>
+__global__ void act_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ z,
+                               T* __restrict__ dz, int64_t n, int act) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    dz[i] = (T)((float)dy[i] * act_grad((float)z[i], act));
+  }
+}
+
+// cols[b*h*w + ph*w + pw][c*P*P + i*P + j] = img[b][c][ph*P+i][pw*P+j]
+template <typename T>
+__global__ void im2col_kernel(const T* __restrict__ img, T* __restrict__ cols, int B,
+                              int C, int HH, int WW, int P) {
+  const int h = HH / P, w = WW / P;
+  const int64_t ncols = (int64_t)C * P * P;
+  const int64_t total = (int64_t)B * h * w * ncols;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = idx / ncols;
+    const int col = (int)(idx - row * ncols);
+    const int pw = (int)(row % w);
+    const int ph = (int)((row / w) % h);
+    const int b = (int)(row / ((int64_t)h * w));
+    const int j = col % P;
+    const int i = (col / P) % P;
+    const int c = col / (P * P);
+    cols[idx] = img[(((int64_t)b * C + c) * HH + ph * P + i) * WW + pw * P + j];
+  }
+}
+
+template <typename T>
+__global__ void col2im_kernel(const T* __restrict__ cols, T* __restrict__ img, int B,
+                              int C, int HH, int WW, int P) {
+  const int h = HH / P, w = WW / P;
+  const int64_t total = (int64_t)B * C * HH * WW;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int x = (int)(idx % WW);
+    const int y = (int)((idx / WW) % HH);
+    const int c = (int)((idx / ((int64_t)WW * HH)) % C);
+    const int b = (int)(idx / ((int64_t)WW * HH * C));
+    const int ph = y / P, i = y % P;
+    const int pw = x / P, j = x % P;
+    const int64_t row = ((int64_t)b * h + ph) * w + pw;
+    const int64_t col = ((int64_t)c * P + i) * P + j;
+    img[idx] = cols[row * (int64_t)C * P * P + col];
+  }
+}
+
+int act_code(const std::string& act) {
+  if (act.empty()) return ACT_NONE;
+  if (act == "gelu") return ACT_GELU;
+  if (act == "gelu_tanh") return ACT_GELU_TANH;
+  if (act == "quickgelu") return ACT_QUICKGELU;
+  TORCH_CHECK(false, "unknown activation ", act);
+}
+
+template <typename T>
+void launch_bias_act(torch::Tensor& z, const c10::optional<torch::Tensor>& bias,
+                     const c10::optional<torch::Tensor>& res, torch::Tensor& y, int M,
+                     int act, bool save_pre, hipStream_t stream) {
+  const int64_t n = z.numel();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n + block - 1) / block, kMaxGrid);
+  const T* bp = bias ? reinterpret_cast<const T*>(bias->data_ptr()) : nullptr;
+  const T* rp = res ? reinterpret_cast<const T*>(res->data_ptr()) : nullptr;
+  T* zp = reinterpret_cast<T*>(z.data_ptr());
+  T* yp = reinterpret_cast<T*>(y.data_ptr());
+#define DISPATCH(HB, HR, SP)                                                              \
+  hipLaunchKernelGGL((bias_act_kernel<T, HB, HR, SP>), dim3(grid), dim3(block), 0,        \
+                     stream, zp, bp, rp, yp, n, M, act)
+  const bool hb = bias.has_value(), hr = res.has_value();
+  if (hb && hr && save_pre) DISPATCH(true, true, true);
+  else if (hb && hr && !save_pre) DISPATCH(true, true, false);
+  else if (hb && !hr && save_pre) DISPATCH(true, false, true);
+  else if (hb && !hr && !save_pre) DISPATCH(true, false, false);
+  else if (!hb && hr && save_pre) DISPATCH(false, true, true);
+  else if (!hb && hr && !save_pre) DISPATCH(false, true, false);
+  else if (!hb && !hr && save_pre) DISPATCH(false, false, true);
+  else DISPATCH(false, false, false);
+#undef DISPATCH
+}
+
+}  // namespace
+
+torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias,
+                           std::string act, c10::optional<torch::Tensor> residual) {
+  TORCH_CHECK(z.is_cuda() && z.is_contiguous());
+  const int M = z.size(-1);
+  const int code = act_code(act);
+  auto stream = at::hip::getCurrentHIPStream();
+  c10::optional<torch::Tensor> b;
+  if (bias) b = bias->contiguous().to(z.scalar_type());
+  c10::optional<torch::Tensor> r;
+  if (residual) {
+    TORCH_CHECK(residual->is_contiguous() && residual->scalar_type() == z.scalar_type());
+    r = *residual;
+  }
+  const bool save_pre = code != ACT_NONE;
+  torch::Tensor y = save_pre || r ? torch::empty_like(z) : z;
+  if (!save_pre && !r && !b) return z;  // nothing to do
+  if (!save_pre && !r) y = z;  // in-place bias add only
+  if (z.scalar_type() == torch::kBFloat16) {
+    launch_bias_act<bf16>(z, b, r, y, M, code, save_pre, stream);
+  } else if (z.scalar_type() == torch::kFloat32) {
+    launch_bias_act<float>(z, b, r, y, M, code, save_pre, stream);
+  } else {
+    TORCH_CHECK(false, "bias_act: unsupported dtype ", z.scalar_type());
+  }
+  return y;
+}
+
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && z.is_contiguous());
+  TORCH_CHECK(dy.scalar_type() == z.scalar_type());
+  const int code = act_code(act);
+  auto dz = torch::empty_like(dy);
+  const int64_t n = dy.numel();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n + block - 1) / block, kMaxGrid);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (dy.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((act_bwd_kernel<bf16>), dim3(grid), dim3(block), 0, stream,
+                       reinterpret_cast<const bf16*>(dy.data_ptr()),
+                       reinterpret_cast<const bf16*>(z.data_ptr()),
+                       reinterpret_cast<bf16*>(dz.data_ptr()), n, code);
+  } else if (dy.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((act_bwd_kernel<float>), dim3(grid), dim3(block), 0, stream,
+                       dy.data_ptr<float>(), z.data_ptr<float>(), dz.data_ptr<float>(), n,
+                       code);
+  } else {
+    TORCH_CHECK(false, "act_bwd: unsupported dtype");
+  }
+  return dz;
+}
+
+torch::Tensor im2col_patch(torch::Tensor img, int64_t patch) {
+  TORCH_CHECK(img.is_cuda() && img.is_contiguous() && img.dim() == 4);
+  const int B = img.size(0), C = img.size(1), HH = img.size(2), WW = img.size(3);
+  const int P = (int)patch;
+  TORCH_CHECK(HH % P == 0 && WW % P == 0);
+  const int h = HH / P, w = WW / P;
+  auto cols = torch::empty({(int64_t)B * h * w, (int64_t)C * P * P}, img.options());
+  const int64_t n = cols.numel();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n + block - 1) / block, kMaxGrid);
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, img.scalar_type(), "im2col", [&] {
+    hipLaunchKernelGGL((im2col_kernel<scalar_t>), dim3(grid), dim3(block), 0, stream,
+                       img.data_ptr<scalar_t>(), cols.data_ptr<scalar_t>(), B, C, HH, WW, P);
+  });
+  return cols;
+}
+
+torch::Tensor col2im_patch(torch::Tensor cols, std::vector<int64_t> img_shape, int64_t patch) {
+  TORCH_CHECK(cols.is_cuda() && cols.is_contiguous());
+  const int B = img_shape[0], C = img_shape[1], HH = img_shape[2], WW = img_shape[3];
+  const int P = (int)patch;
+  auto img = torch::empty(img_shape, cols.options());
+  const int64_t n = img.numel();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n + block - 1) / block, kMaxGrid);
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, cols.scalar_type(), "col2im", [&] {
+    hipLaunchKernelGGL((col2im_kernel<scalar_t>), dim3(grid), dim3(block), 0, stream,
+                       cols.data_ptr<scalar_t>(), img.data_ptr<scalar_t>(), B, C, HH, WW, P);
+  });
+  return img;
+}

@@ -1,0 +1,218 @@
+// K3 — LayerNorm forward/backward for CDNA4 (gfx950).
+//
+// Reference op semantics: flax nnx.LayerNorm as used at
+// /root/reference/src/jimm/common/transformer.py:58-66,80-88 (per-row
+// mean/var over the last dim, eps per model family: 1e-12/1e-5/1e-6).
+//
+// Design: memory-bound streaming op (Appendix B of the CDNA guide):
+//   * one wave per row (H <= a few K), 4 waves (4 rows) per block,
+//     grid-stride over rows, grid capped so blocks stay resident;
+//   * bf16 loads vectorized as short2/short4 (G13: hipcc does not
+//     auto-vectorize bf16), fp32 math, fp32 mean/rstd saved for backward;
+//   * backward dweight/dbias: per-block fp32 partials in LDS, one
+//     global atomicAdd per column per block (G12).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+template <typename T, int VEC>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                              const float* __restrict__ b, T* __restrict__ y,
+                              float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                              int64_t nrows, int H, float eps) {
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  for (int64_t row = (int64_t)blockIdx.x * waves_per_block + wave; row < nrows;
+       row += (int64_t)gridDim.x * waves_per_block) {
+    const T* xr = x + row * H;
+    T* yr = y + row * H;
+    float sum = 0.f, sumsq = 0.f;
+    // per-lane strided vector loads
+    for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float v = (float)xr[i + k];
+        sum += v;
+        sumsq += v * v;
+      }
+    }
+    sum = wave_reduce_sum(sum);
+    sumsq = wave_reduce_sum(sumsq);
+    const float mean = sum / H;
+    const float var = sumsq / H - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    if (lane == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+    for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float v = (float)xr[i + k];
+        yr[i + k] = (T)((v - mean) * rstd * w[i + k] + b[i + k]);
+      }
+    }
+  }
+}
+
+template <typename T, int VEC>
+__global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                              const float* __restrict__ w, const float* __restrict__ mean,
+                              const float* __restrict__ rstd, T* __restrict__ dx,
+                              float* __restrict__ dw, float* __restrict__ db,
+                              int64_t nrows, int H) {
+  // per-wave fp32 partial slabs: [waves][H] for dw and db — no atomics in
+  // the row loop (each wave owns its slab; each lane its columns)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  float* s_dw = reinterpret_cast<float*>(smem) + (size_t)wave * H;
+  float* s_db = reinterpret_cast<float*>(smem) + (size_t)waves_per_block * H + (size_t)wave * H;
+  for (int i = lane; i < H; i += WAVE) {
+    s_dw[i] = 0.f;
+    s_db[i] = 0.f;
+  }
+  __syncthreads();
+  for (int64_t row = (int64_t)blockIdx.x * waves_per_block + wave; row < nrows;
+       row += (int64_t)gridDim.x * waves_per_block) {
+    const T* dyr = dy + row * H;
+    const T* xr = x + row * H;
+    T* dxr = dx + row * H;
+    const float mu = mean[row], rs = rstd[row];
+    // c1 = mean(dy*w), c2 = mean(dy*w*xhat)
+    float c1 = 0.f, c2 = 0.f;
+    for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float g = (float)dyr[i + k] * w[i + k];
+        float xhat = ((float)xr[i + k] - mu) * rs;
+        c1 += g;
+        c2 += g * xhat;
+      }
+    }
+    c1 = wave_reduce_sum(c1) / H;
+    c2 = wave_reduce_sum(c2) / H;
+    for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float gy = (float)dyr[i + k];
+        float g = gy * w[i + k];
+        float xhat = ((float)xr[i + k] - mu) * rs;
+        dxr[i + k] = (T)(rs * (g - c1 - xhat * c2));
+        s_dw[i + k] += gy * xhat;
+        s_db[i + k] += gy;
+      }
+    }
+  }
+  __syncthreads();
+  // merge the per-wave slabs, one global atomic per column per block
+  float* base_dw = reinterpret_cast<float*>(smem);
+  float* base_db = base_dw + (size_t)waves_per_block * H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    float adw = 0.f, adb = 0.f;
+    for (int wv = 0; wv < waves_per_block; ++wv) {
+      adw += base_dw[(size_t)wv * H + i];
+      adb += base_db[(size_t)wv * H + i];
+    }
+    atomicAdd(&dw[i], adw);
+    atomicAdd(&db[i], adb);
+  }
+}
+
+template <typename T>
+void launch_ln_fwd(const T* x, const float* w, const float* b, T* y, float* mean,
+                   float* rstd, int64_t nrows, int H, float eps, hipStream_t stream) {
+  const int block = 256;
+  const int waves_per_block = block / WAVE;
+  int grid = (int)std::min<int64_t>((nrows + waves_per_block - 1) / waves_per_block, 2048);
+  auto pick = [&](auto vec_tag) {
+    constexpr int V = decltype(vec_tag)::value;
+    hipLaunchKernelGGL((ln_fwd_kernel<T, V>), dim3(grid), dim3(block), 0, stream, x, w, b,
+                       y, mean, rstd, nrows, H, eps);
+  };
+  if (H % (WAVE * 8) == 0) pick(std::integral_constant<int, 8>{});
+  else if (H % (WAVE * 4) == 0) pick(std::integral_constant<int, 4>{});
+  else if (H % (WAVE * 2) == 0) pick(std::integral_constant<int, 2>{});
+  else if (H % WAVE == 0) pick(std::integral_constant<int, 1>{});
+  else TORCH_CHECK(false, "layernorm: H must be a multiple of 64, got ", H);
+}
+
+template <typename T>
+void launch_ln_bwd(const T* dy, const T* x, const float* w, const float* mean,
+                   const float* rstd, T* dx, float* dw, float* db, int64_t nrows, int H,
+                   hipStream_t stream) {
+  const int block = 256;
+  const int waves_per_block = block / WAVE;
+  int grid = (int)std::min<int64_t>((nrows + waves_per_block - 1) / waves_per_block, 1024);
+  size_t shmem = 2 * (size_t)waves_per_block * H * sizeof(float);
+  auto pick = [&](auto vec_tag) {
+    constexpr int V = decltype(vec_tag)::value;
+    hipLaunchKernelGGL((ln_bwd_kernel<T, V>), dim3(grid), dim3(block), shmem, stream, dy,
+                       x, w, mean, rstd, dx, dw, db, nrows, H);
+  };
+  if (H % (WAVE * 4) == 0) pick(std::integral_constant<int, 4>{});
+  else if (H % (WAVE * 2) == 0) pick(std::integral_constant<int, 2>{});
+  else if (H % WAVE == 0) pick(std::integral_constant<int, 1>{});
+  else TORCH_CHECK(false, "layernorm: H must be a multiple of 64, got ", H);
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                                         double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int H = x.size(-1);
+  const int64_t nrows = x.numel() / H;
+  auto wf = w.contiguous().to(torch::kFloat32);
+  auto bf = b.contiguous().to(torch::kFloat32);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({nrows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({nrows}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  if (x.scalar_type() == torch::kBFloat16) {
+    launch_ln_fwd(reinterpret_cast<const bf16*>(x.data_ptr()), wf.data_ptr<float>(),
+                  bf.data_ptr<float>(), reinterpret_cast<bf16*>(y.data_ptr()),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(), nrows, H, (float)eps,
+                  stream);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    launch_ln_fwd(x.data_ptr<float>(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+                  y.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  nrows, H, (float)eps, stream);
+  } else {
+    TORCH_CHECK(false, "layernorm: unsupported dtype ", x.scalar_type());
+  }
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor mean, torch::Tensor rstd) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  const int H = x.size(-1);
+  const int64_t nrows = x.numel() / H;
+  auto wf = w.contiguous().to(torch::kFloat32);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto db = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  if (x.scalar_type() == torch::kBFloat16) {
+    launch_ln_bwd(reinterpret_cast<const bf16*>(dy.data_ptr()),
+                  reinterpret_cast<const bf16*>(x.data_ptr()), wf.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  reinterpret_cast<bf16*>(dx.data_ptr()), dw.data_ptr<float>(),
+                  db.data_ptr<float>(), nrows, H, stream);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    launch_ln_bwd(dy.data_ptr<float>(), x.data_ptr<float>(), wf.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr<float>(),
+                  dw.data_ptr<float>(), db.data_ptr<float>(), nrows, H, stream);
+  } else {
+    TORCH_CHECK(false, "layernorm: unsupported dtype ", x.scalar_type());
+  }
+  // cast param grads to param dtype at the python layer if needed
+  return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
+}

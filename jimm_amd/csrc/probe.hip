@@ -1,0 +1,50 @@
+// MFMA fragment-layout probe (gfx950).
+//
+// The attention / GEMM kernels assume these v_mfma_f32_16x16x32_bf16
+// layouts (CDNA guide §3 gives C/D; A/B are the natural extension of the
+// CDNA3 16x16x16 mapping and are VERIFIED ON HARDWARE by this probe —
+// tests/test_kernels_gpu.py::test_mfma_probe runs it with asymmetric
+// random matrices, which catches any transposed assumption):
+//   A[i][k]: lane l holds A[l&15][(l>>4)*8 + j]   j=0..7
+//   B[k][j]: lane l holds B[(l>>4)*8 + j][l&15]
+//   C[i][j]: lane l holds C[(l>>4)*4 + r][l&15]   r=0..3
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+__global__ void mfma_probe_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                                  float* __restrict__ C) {
+  const int lane = threadIdx.x % WAVE;
+  const int lo = lane & 15, hi = lane >> 4;
+  bf16x8_t a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = *reinterpret_cast<const short*>(&A[lo * 32 + hi * 8 + j]);      // A[16][32] row-major
+    b[j] = *reinterpret_cast<const short*>(&B[(hi * 8 + j) * 16 + lo]);    // B[32][16] row-major
+  }
+  f32x4_t c = {};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) C[(hi * 4 + r) * 16 + lo] = c[r];
+}
+
+}  // namespace
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) && B.sizes() == torch::IntArrayRef({32, 16}));
+  auto C = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     reinterpret_cast<const bf16*>(A.contiguous().data_ptr()),
+                     reinterpret_cast<const bf16*>(B.contiguous().data_ptr()),
+                     C.data_ptr<float>());
+  return C;
+}

@@ -1,0 +1,69 @@
+"""Text transformer tower shared by CLIP and SigLIP.
+
+Reference semantics:
+  * CLIP text (/root/reference/src/jimm/models/clip.py:92-123,148-167):
+    token embed + learned pos-emb, CAUSAL transformer, ln_final (eps 1e-5),
+    EOT pooling ``x[arange(B), argmax(ids, -1)]`` (EOT=49407 is the max BPE id);
+  * SigLIP text (/root/reference/src/jimm/models/siglip.py:79-119,135-153):
+    NON-causal transformer, eps 1e-6, last-token pooling ``x[:, -1]``
+    (requires padding="max_length").
+
+The projection (bias-free for CLIP, biased for SigLIP) is owned by the model,
+not this tower.
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+from jimm_amd import ops
+from jimm_amd.models.common.transformer import Encoder
+
+
+class TextTransformer(nn.Module):
+    def __init__(
+        self,
+        vocab_size: int,
+        context_length: int,
+        hidden_size: int,
+        num_layers: int,
+        num_heads: int,
+        mlp_dim: int,
+        *,
+        causal: bool,
+        pooling: str,  # "EOT" | "LAST"
+        hidden_act: str = "gelu",
+        layernorm_epsilon: float = 1e-5,
+        dropout_rate: float = 0.0,
+    ) -> None:
+        super().__init__()
+        if pooling not in ("EOT", "LAST"):
+            raise ValueError(f"pooling must be EOT or LAST, got {pooling!r}")
+        self.pooling = pooling
+        self.eps = layernorm_epsilon
+        self.token_embedding = nn.Embedding(vocab_size, hidden_size)
+        self.pos_embedding = nn.Parameter(torch.empty(1, context_length, hidden_size))
+        nn.init.normal_(self.token_embedding.weight, std=0.02)
+        nn.init.normal_(self.pos_embedding, std=0.01)
+        self.encoder = Encoder(
+            num_layers,
+            hidden_size,
+            num_heads,
+            mlp_dim,
+            dropout_rate=dropout_rate,
+            hidden_act=hidden_act,
+            layernorm_epsilon=layernorm_epsilon,
+            causal=causal,
+        )
+        self.ln_final = nn.LayerNorm(hidden_size, eps=layernorm_epsilon)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        """input_ids (B, L) -> pooled (B, H)."""
+        x = self.token_embedding(input_ids) + self.pos_embedding[:, : input_ids.shape[1]]  # K10
+        x = self.encoder(x)
+        x = ops.layer_norm(x, self.ln_final.weight, self.ln_final.bias, self.eps)
+        if self.pooling == "EOT":
+            # clip.py:164-166 — EOT token has the highest BPE id
+            return x[torch.arange(x.shape[0], device=x.device), input_ids.argmax(dim=-1)]
+        return x[:, -1]

@@ -1,0 +1,118 @@
+"""GPU linear / patch-embed path: GEMM + fused bias/activation/residual epilogue.
+
+GEMM engine selection (env ``JIMM_AMD_GEMM``):
+  * ``hip``  — in-house MFMA bf16 GEMM kernel (csrc/gemm.hip), epilogue fused
+               into the GEMM itself (bias + gelu/quickgelu + residual).
+  * ``blas`` — rocBLAS via ``torch.matmul`` (plain library GEMM) followed by
+               the fused HIP bias+act(+residual) elementwise kernel.
+
+Backward GEMMs (dx = dz@W, dW = dz^T@x) always run through rocBLAS — they are
+plain GEMMs with no epilogue to fuse; dz itself comes from the fused HIP
+``act_bwd`` kernel.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+from jimm_amd.ops import _backend
+
+
+def _gemm_mode() -> str:
+    return os.environ.get("JIMM_AMD_GEMM", "hip")
+
+
+def _gemm_nt(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """z = x2 @ w.T  — (N,K)@(M,K)^T -> (N,M), no epilogue."""
+    return torch.matmul(x2, w.t())
+
+
+class _LinearActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, act, residual):
+        in_f, out_f = w.shape[1], w.shape[0]
+        x2 = x.contiguous().reshape(-1, in_f)
+        res2 = residual.contiguous().reshape(-1, out_f) if residual is not None else None
+        ext = _backend.ext()
+        if _gemm_mode() == "hip" and ext.gemm_supported(x2.shape[0], out_f, in_f, str(x2.dtype)):
+            y, z = ext.linear_fwd(x2, w, b, act or "", res2, act is not None)
+        else:
+            z = _gemm_nt(x2, w)
+            # fused bias+act(+residual): z is updated in place to hold the
+            # pre-activation (z+bias); y aliases z when there is no act.
+            y = ext.bias_act_fwd(z, b, act or "", res2)
+        if act is not None:
+            ctx.save_for_backward(x2, w, z)
+        else:
+            ctx.save_for_backward(x2, w)
+        ctx.act = act
+        ctx.has_bias = b is not None
+        ctx.has_res = residual is not None
+        ctx.res_shape = residual.shape if residual is not None else None
+        ctx.x_shape = x.shape
+        return y.view(*x.shape[:-1], out_f)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.act is not None:
+            x2, w, z = ctx.saved_tensors
+        else:
+            x2, w = ctx.saved_tensors
+            z = None
+        dy2 = dy.contiguous().reshape(-1, dy.shape[-1])
+        if ctx.act is not None:
+            dz = _backend.ext().act_bwd(dy2, z, ctx.act)
+        else:
+            dz = dy2
+        dx = torch.matmul(dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
+        dw = torch.matmul(dz.t(), x2) if ctx.needs_input_grad[1] else None
+        db = dz.sum(dim=0) if (ctx.has_bias and ctx.needs_input_grad[2]) else None
+        dres = dy.view(ctx.res_shape) if (ctx.has_res and ctx.needs_input_grad[4]) else None
+        return dx, dw, db, None, dres
+
+
+def linear_act(x, w, b, act, residual):
+    return _LinearActFn.apply(x, w, b, act, residual)
+
+
+class _PatchEmbedFn(torch.autograd.Function):
+    """K1: stride=kernel conv == im2col (pure gather) + NT GEMM + bias.
+
+    im2col for kernel==stride is a permute-gather:
+      img (B,C,h*P,w*P) -> cols (B*h*w, C*P*P), patch-major rows.
+    The HIP kernel does the gather with coalesced writes; the GEMM against
+    weight (hidden, C*P*P) reuses the linear path.
+    """
+
+    @staticmethod
+    def forward(ctx, img, w, b, patch):
+        B, C, H, W = img.shape
+        h, wn = H // patch, W // patch
+        ext = _backend.ext()
+        cols = ext.im2col_patch(img.contiguous(), patch)  # (B*h*w, C*P*P)
+        w2 = w.reshape(w.shape[0], -1)  # (hidden, C*P*P)
+        z = _gemm_nt(cols, w2)
+        y = ext.bias_act_fwd(z, b, "", None)
+        ctx.save_for_backward(cols, w2)
+        ctx.img_shape = img.shape
+        ctx.patch = patch
+        ctx.has_bias = b is not None
+        ctx.w_shape = w.shape
+        return y.view(B, h * wn, w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        cols, w2 = ctx.saved_tensors
+        patch = ctx.patch
+        dy2 = dy.contiguous().reshape(-1, dy.shape[-1])
+        dcols = torch.matmul(dy2, w2)
+        dimg = _backend.ext().col2im_patch(dcols, list(ctx.img_shape), patch) if ctx.needs_input_grad[0] else None
+        dw = torch.matmul(dy2.t(), cols).view(ctx.w_shape) if ctx.needs_input_grad[1] else None
+        db = dy2.sum(dim=0) if (ctx.has_bias and ctx.needs_input_grad[2]) else None
+        return dimg, dw, db, None
+
+
+def patch_embed(img, w, b, patch):
+    return _PatchEmbedFn.apply(img, w, b, patch)

@@ -1,0 +1,174 @@
+"""Numerical-parity tests against HuggingFace transformers (in-process, no
+network): tiny random-init HF models are saved with save_pretrained and
+loaded through jimm_amd's from_pretrained, then outputs are compared.
+
+This reproduces the reference's test strategy (SURVEY.md §4 —
+tests/test_vit.py, test_clip.py, test_siglip.py) with offline fixtures and
+much tighter tolerances (fp32, same-process): reference tolerances were
+ViT max-abs < 0.05, CLIP atol 1e-1, SigLIP atol 1e-2.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+import jimm_amd
+
+transformers = pytest.importorskip("transformers")
+
+
+def _tiny_vit(tmp_path, old_keys: bool):
+    from transformers import ViTConfig, ViTForImageClassification
+
+    cfg = ViTConfig(
+        hidden_size=32,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        intermediate_size=64,
+        image_size=32,
+        patch_size=16,
+        num_labels=3,
+    )
+    hf = ViTForImageClassification(cfg).eval()
+    for p in hf.parameters():
+        p.data.normal_(0, 0.02)
+    d = tmp_path / "vit"
+    hf.save_pretrained(d, safe_serialization=True)
+    if old_keys:
+        # rewrite to the classic hub key scheme (transformers <=4.x)
+        from safetensors.torch import load_file, save_file
+
+        sd = load_file(d / "model.safetensors")
+        out = {}
+        for k, t in sd.items():
+            k = k.replace("vit.layers.", "vit.encoder.layer.")
+            k = k.replace(".attention.q_proj.", ".attention.attention.query.")
+            k = k.replace(".attention.k_proj.", ".attention.attention.key.")
+            k = k.replace(".attention.v_proj.", ".attention.attention.value.")
+            k = k.replace(".attention.o_proj.", ".attention.output.dense.")
+            k = k.replace(".mlp.fc1.", ".intermediate.dense.")
+            k = k.replace(".mlp.fc2.", ".output.dense.")
+            out[k] = t
+        save_file(out, str(d / "model.safetensors"))
+    return hf, d
+
+
+@pytest.mark.parametrize("old_keys", [False, True])
+def test_vit_parity(tmp_path, old_keys):
+    hf, d = _tiny_vit(tmp_path, old_keys)
+    model = jimm_amd.VisionTransformer.from_pretrained(str(d)).eval()
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        ref = hf(x).logits
+        out = model(x)
+    assert out.shape == ref.shape
+    diff = (out - ref).abs().max().item()
+    assert diff < 1e-4, f"max abs diff {diff}"  # reference bar: <0.05 (test_vit.py:49-52)
+
+
+def _tiny_clip(tmp_path):
+    from transformers import CLIPConfig, CLIPModel
+
+    cfg = CLIPConfig(
+        text_config=dict(
+            hidden_size=32,
+            intermediate_size=64,
+            num_hidden_layers=2,
+            num_attention_heads=2,
+            max_position_embeddings=12,
+            vocab_size=100,
+            bos_token_id=98,
+            eos_token_id=99,
+        ),
+        vision_config=dict(hidden_size=64, intermediate_size=128, num_hidden_layers=2, num_attention_heads=2, image_size=32, patch_size=16),
+        projection_dim=16,
+    )
+    hf = CLIPModel(cfg).eval()
+    for p in hf.parameters():
+        p.data.normal_(0, 0.02)
+    with torch.no_grad():
+        hf.logit_scale.fill_(float(np.log(1 / 0.07)))
+    d = tmp_path / "clip"
+    hf.save_pretrained(d, safe_serialization=True)
+    return hf, d
+
+
+def test_clip_parity(tmp_path):
+    hf, d = _tiny_clip(tmp_path)
+    model = jimm_amd.CLIP.from_pretrained(str(d)).eval()
+    img = torch.randn(2, 3, 32, 32)
+    # ids: eos (=99, also the max id) at a known position per row
+    ids = torch.randint(0, 98, (2, 12))
+    ids[0, 5] = 99
+    ids[1, 9] = 99
+    with torch.no_grad():
+        ref = hf(input_ids=ids, pixel_values=img)
+        li, lt = model(img, ids)
+    d_img = (li - ref.logits_per_image).abs().max().item()
+    d_txt = (lt - ref.logits_per_text).abs().max().item()
+    assert d_img < 1e-3 and d_txt < 1e-3, (d_img, d_txt)  # reference bar: atol 1e-1
+
+    with torch.no_grad():
+        out_i = model.encode_image(img)
+        out_t = model.encode_text(ids)
+    out_i = out_i / out_i.norm(dim=-1, keepdim=True)
+    out_t = out_t / out_t.norm(dim=-1, keepdim=True)
+    ref_i = ref.image_embeds  # HF returns the L2-normalized embeddings
+    ref_t = ref.text_embeds
+    assert (out_i - ref_i).abs().max().item() < 1e-5
+    assert (out_t - ref_t).abs().max().item() < 1e-5
+
+
+def _tiny_siglip(tmp_path):
+    from transformers import SiglipConfig, SiglipModel
+
+    cfg = SiglipConfig(
+        text_config=dict(hidden_size=64, intermediate_size=96, num_hidden_layers=2, num_attention_heads=2, max_position_embeddings=12, vocab_size=100),
+        vision_config=dict(hidden_size=64, intermediate_size=96, num_hidden_layers=2, num_attention_heads=2, image_size=32, patch_size=16),
+    )
+    hf = SiglipModel(cfg).eval()
+    for p in hf.parameters():
+        p.data.normal_(0, 0.02)
+    d = tmp_path / "siglip"
+    hf.save_pretrained(d, safe_serialization=True)
+    return hf, d
+
+
+def test_siglip_parity(tmp_path):
+    hf, d = _tiny_siglip(tmp_path)
+    model = jimm_amd.SigLIP.from_pretrained(str(d)).eval()
+    img = torch.randn(2, 3, 32, 32)
+    ids = torch.randint(0, 100, (2, 12))
+    with torch.no_grad():
+        ref = hf(input_ids=ids, pixel_values=img)
+        out_i = model.encode_image(img)
+        out_t = model.encode_text(ids)
+        li, lt = model(img, ids)
+    out_i = out_i / out_i.norm(dim=-1, keepdim=True)
+    out_t = out_t / out_t.norm(dim=-1, keepdim=True)
+    assert (out_i - ref.image_embeds).abs().max().item() < 1e-5  # reference bar: atol 1e-2
+    assert (out_t - ref.text_embeds).abs().max().item() < 1e-5
+    assert (li - ref.logits_per_image).abs().max().item() < 1e-3
+    assert (lt - ref.logits_per_text).abs().max().item() < 1e-3
+
+
+def test_vit_shape_inference(tmp_path):
+    """Bare safetensors file (no config.json) -> shape-inferred config
+    (reference models/vit.py:144-164)."""
+    hf, d = _tiny_vit(tmp_path, old_keys=True)
+    import os
+
+    os.remove(d / "config.json")
+    # heads inference assumes head_dim 64 (hidden//64) which fails for tiny
+    # hidden sizes; check the structural fields instead by loading weights file
+    weights = str(d / "model.safetensors")
+    from jimm_amd.interop.vit_hf import _infer_config
+    from safetensors.torch import load_file
+
+    cfg = _infer_config(load_file(weights))
+    assert cfg["hidden_size"] == 32
+    assert cfg["num_layers"] == 2
+    assert cfg["patch_size"] == 16
+    assert cfg["img_size"] == 32
+    assert cfg["mlp_dim"] == 64
+    assert cfg["num_classes"] == 3

@@ -1,0 +1,90 @@
+"""CPU reference-path op tests (these ops are the oracle the HIP kernels
+are tested against in tests/test_kernels_gpu.py)."""
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from jimm_amd import ops
+
+
+def test_quickgelu():
+    x = torch.randn(64)
+    assert torch.allclose(ops.quickgelu(x), x * torch.sigmoid(1.702 * x))
+
+
+def test_layer_norm_matches_torch():
+    x = torch.randn(4, 7, 32)
+    w, b = torch.randn(32), torch.randn(32)
+    for eps in (1e-12, 1e-6, 1e-5):
+        assert torch.allclose(ops.layer_norm(x, w, b, eps), F.layer_norm(x, (32,), w, b, eps), atol=1e-6)
+
+
+def test_attention_matches_sdpa():
+    q = torch.randn(2, 3, 11, 16)
+    k = torch.randn(2, 3, 13, 16)
+    v = torch.randn(2, 3, 13, 16)
+    out = ops.attention(q, k, v)
+    ref = F.scaled_dot_product_attention(q, k, v)
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_attention_causal():
+    q = k = v = torch.randn(1, 2, 9, 8)
+    out = ops.attention(q, k, v, causal=True)
+    ref = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+    assert torch.allclose(out, ref, atol=1e-5)
+    # causality: changing a future key must not change earlier outputs
+    k2 = k.clone()
+    k2[:, :, -1] += 10.0
+    out2 = ops.attention(q, k2, v, causal=True)
+    assert torch.allclose(out[:, :, :-1], out2[:, :, :-1], atol=1e-6)
+
+
+def test_linear_epilogues():
+    x = torch.randn(5, 12)
+    w = torch.randn(8, 12)
+    b = torch.randn(8)
+    res = torch.randn(5, 8)
+    assert torch.allclose(ops.linear(x, w, b), F.linear(x, w, b), atol=1e-6)
+    assert torch.allclose(ops.linear(x, w, b, act="gelu"), F.gelu(F.linear(x, w, b)), atol=1e-6)
+    assert torch.allclose(ops.linear(x, w, b, act="gelu_tanh"), F.gelu(F.linear(x, w, b), approximate="tanh"), atol=1e-6)
+    z = F.linear(x, w, b)
+    assert torch.allclose(ops.linear(x, w, b, act="quickgelu"), z * torch.sigmoid(1.702 * z), atol=1e-6)
+    assert torch.allclose(ops.linear(x, w, b, residual=res), F.linear(x, w, b) + res, atol=1e-6)
+
+
+def test_patch_embed_matches_conv():
+    img = torch.randn(2, 3, 32, 32)
+    w = torch.randn(16, 3, 8, 8)
+    b = torch.randn(16)
+    y = ops.patch_embed(img, w, b, 8)
+    ref = F.conv2d(img, w, b, stride=8).flatten(2).transpose(1, 2)
+    assert y.shape == (2, 16, 16)
+    assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_add_cls_pos():
+    x = torch.randn(2, 4, 8)
+    cls = torch.randn(1, 1, 8)
+    pos = torch.randn(1, 5, 8)
+    y = ops.add_cls_pos(x, cls, pos)
+    assert y.shape == (2, 5, 8)
+    assert torch.allclose(y[:, 0], cls[0, 0] + pos[0, 0])
+    assert torch.allclose(y[:, 1:], x + pos[:, 1:])
+    # MAP mode: no cls token
+    y2 = ops.add_cls_pos(x, None, pos)
+    assert torch.allclose(y2, x + pos[:, :4])
+
+
+def test_attention_grad():
+    q = torch.randn(1, 2, 5, 4, requires_grad=True)
+    k = torch.randn(1, 2, 5, 4, requires_grad=True)
+    v = torch.randn(1, 2, 5, 4, requires_grad=True)
+    assert torch.autograd.gradcheck(
+        lambda q, k, v: ops.attention(q.double(), k.double(), v.double()),
+        (q.double(), k.double(), v.double()),
+        eps=1e-6,
+        atol=1e-4,
+    )
