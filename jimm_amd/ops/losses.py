@@ -25,6 +25,7 @@ def clip_contrastive_loss(
     logit_scale: torch.Tensor,
     *,
     group=None,
+    gather: bool = True,
 ) -> torch.Tensor:
     """Symmetric InfoNCE over the GLOBAL batch.
 
@@ -36,9 +37,10 @@ def clip_contrastive_loss(
     txt = txt_emb / txt_emb.norm(dim=-1, keepdim=True)
     scale = logit_scale.exp()
     b_local = img.shape[0]
-    rank = dist.get_rank(group) if dist.is_initialized() else 0
-    img_all = all_gather_with_grad(img, group)
-    txt_all = all_gather_with_grad(txt, group)
+    do_gather = gather and dist.is_initialized()
+    rank = dist.get_rank(group) if do_gather else 0
+    img_all = all_gather_with_grad(img, group) if do_gather else img
+    txt_all = all_gather_with_grad(txt, group) if do_gather else txt
     labels = torch.arange(b_local, device=img.device) + rank * b_local
     logits_i = scale * img @ txt_all.t()  # (B_local, B_global)
     logits_t = scale * txt @ img_all.t()
@@ -52,6 +54,7 @@ def siglip_sigmoid_loss(
     logit_bias: torch.Tensor,
     *,
     group=None,
+    gather: bool = True,
     chunk_size: int = 8192,
 ) -> torch.Tensor:
     """SigLIP pairwise sigmoid loss over the global batch.
@@ -66,8 +69,9 @@ def siglip_sigmoid_loss(
     txt = txt_emb / txt_emb.norm(dim=-1, keepdim=True)
     scale = logit_scale.exp()
     b_local = img.shape[0]
-    rank = dist.get_rank(group) if dist.is_initialized() else 0
-    txt_all = all_gather_with_grad(txt, group)  # one tower gathered suffices
+    do_gather = gather and dist.is_initialized()
+    rank = dist.get_rank(group) if do_gather else 0
+    txt_all = all_gather_with_grad(txt, group) if do_gather else txt
     b_global = txt_all.shape[0]
     diag = torch.arange(b_local, device=img.device)
     total = img.new_zeros(())

@@ -1,0 +1,173 @@
+"""Multi-process logic tests on CPU (gloo, world_size 2) — SURVEY §4
+implication (c)/(d): bucketed DP all-reduce, all-gather-with-grad, and the
+DP-invariance property (loss/grads at DP=2 equal single-process on the same
+global batch)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+import jimm_amd
+from jimm_amd.ops import losses as L
+
+pytestmark = pytest.mark.dist
+
+
+def _run(rank, world, fn, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)
+    try:
+        fn(rank, world)
+    finally:
+        dist.destroy_process_group()
+
+
+def spawn(fn, world=2, port=29511):
+    mp.spawn(_run, args=(world, fn, port), nprocs=world, join=True)
+
+
+# ---------------------------------------------------------------------------
+def _check_allgather_grad(rank, world):
+    from jimm_amd.parallel.gather import all_gather_with_grad
+
+    x = torch.randn(3, 4, requires_grad=True)  # same seed -> same on each rank
+    local = x * (rank + 1.0)
+    g = all_gather_with_grad(local)
+    assert g.shape == (3 * world, 4)
+    # loss uses EVERY rank's slice differently
+    w = torch.arange(1.0, 1.0 + g.shape[0]).unsqueeze(1)
+    (g * w).sum().backward()
+    # d/d local_r = sum over ranks of w-slice-r (each rank computes the same
+    # loss here) -> grad = world * w_r ; then * (rank+1) chain to x
+    wr = w.view(world, 3, 1)[rank]
+    expect = world * wr.expand(3, 4) * (rank + 1.0)
+    assert torch.allclose(x.grad, expect, atol=1e-5), (rank, x.grad, expect)
+
+
+def test_allgather_with_grad():
+    spawn(_check_allgather_grad, port=29512)
+
+
+# ---------------------------------------------------------------------------
+def _check_ddp_average(rank, world):
+    from jimm_amd.parallel.ddp import DataParallelGrads
+
+    m = torch.nn.Linear(4, 3)
+    ddp = DataParallelGrads(m, bucket_bytes=8)  # force multiple buckets
+    data = lambda r: torch.arange(20.0).reshape(5, 4) * (r + 1.0) / 10.0
+    ddp.zero_grad()
+    m(data(rank)).sum().backward()
+    ddp.finalize()
+    got = [p.grad.clone() for p in m.parameters()]
+    # reference: average of per-rank grads
+    m2 = torch.nn.Linear(4, 3)
+    m2.load_state_dict(m.state_dict())
+    ref = None
+    for r in range(world):
+        m2.zero_grad()
+        m2(data(r)).sum().backward()
+        g = [p.grad.clone() for p in m2.parameters()]
+        ref = g if ref is None else [a + b for a, b in zip(ref, g)]
+    ref = [g / world for g in ref]
+    for a, b in zip(got, ref):
+        assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_ddp_bucketed_average():
+    spawn(_check_ddp_average, port=29513)
+
+
+# ---------------------------------------------------------------------------
+def _dp_invariance(rank, world):
+    """CLIP loss + grads at DP=2 must equal the single-process global batch."""
+    torch.manual_seed(0)
+    model = jimm_amd.CLIP(embed_dim=16, image_resolution=32, vision_layers=1, vision_width=64,
+                          vision_patch_size=16, context_length=8, vocab_size=100,
+                          transformer_width=64, transformer_heads=1, transformer_layers=1)
+    from jimm_amd.parallel.ddp import DataParallelGrads
+
+    ddp = DataParallelGrads(model)  # also broadcasts params
+    torch.manual_seed(42)
+    imgs = torch.randn(4, 3, 32, 32)
+    ids = torch.randint(0, 99, (4, 8))
+    ids[:, -1] = 99
+    # shard the global batch
+    b = 4 // world
+    my_imgs, my_ids = imgs[rank * b : (rank + 1) * b], ids[rank * b : (rank + 1) * b]
+    ddp.zero_grad()
+    img_e = model.encode_image(my_imgs)
+    txt_e = model.encode_text(my_ids)
+    loss = L.clip_contrastive_loss(img_e, txt_e, model.logit_scale)
+    loss.backward()
+    ddp.finalize()
+
+    # single-process reference on the full batch
+    model_ref = jimm_amd.CLIP(embed_dim=16, image_resolution=32, vision_layers=1, vision_width=64,
+                              vision_patch_size=16, context_length=8, vocab_size=100,
+                              transformer_width=64, transformer_heads=1, transformer_layers=1)
+    model_ref.load_state_dict(model.state_dict())
+    img_r = model_ref.encode_image(imgs)
+    txt_r = model_ref.encode_text(ids)
+    loss_r = L.clip_contrastive_loss(img_r, txt_r, model_ref.logit_scale, gather=False)
+    loss_r.backward()
+
+    # per-rank loss differs (local rows) but the DP-averaged loss matches
+    lt = loss.detach().clone()
+    dist.all_reduce(lt)
+    assert torch.allclose(lt / world, loss_r.detach(), atol=1e-5), (lt / world, loss_r)
+    for (n, p), (nr, pr) in zip(model.named_parameters(), model_ref.named_parameters()):
+        assert n == nr
+        if pr.grad is None:
+            continue
+        assert torch.allclose(p.grad, pr.grad, atol=1e-4), f"{n}: {(p.grad - pr.grad).abs().max()}"
+
+
+def test_dp_invariance_clip():
+    spawn(_dp_invariance, port=29514)
+
+
+# ---------------------------------------------------------------------------
+def _siglip_invariance(rank, world):
+    torch.manual_seed(0)
+    img = torch.randn(6, 16)
+    txt = torch.randn(6, 16)
+    scale = torch.tensor(0.7, requires_grad=True)
+    bias = torch.tensor(-1.0, requires_grad=True)
+    b = 6 // world
+    my_img = img[rank * b : (rank + 1) * b].clone().requires_grad_(True)
+    my_txt = txt[rank * b : (rank + 1) * b].clone().requires_grad_(True)
+    loss = L.siglip_sigmoid_loss(my_img, my_txt, scale, bias, chunk_size=4)
+    loss.backward()
+    # reference single-process
+    img_r = img.clone().requires_grad_(True)
+    txt_r = txt.clone().requires_grad_(True)
+    scale_r = torch.tensor(0.7, requires_grad=True)
+    bias_r = torch.tensor(-1.0, requires_grad=True)
+    loss_r = L.siglip_sigmoid_loss(img_r, txt_r, scale_r, bias_r, gather=False)
+    loss_r.backward()
+    lt = loss.detach().clone()
+    dist.all_reduce(lt)
+    assert torch.allclose(lt / world, loss_r.detach(), atol=1e-5)
+    # DP grad average must equal the reference grads for this rank's shard
+    gi = my_img.grad.clone()
+    gt = my_txt.grad.clone()
+    dist.all_reduce(gi)  # emb grads are local (no param sharing) -> no averaging;
+    # instead compare directly: ref grad on shard r vs (local grad / world)?
+    # The DP convention averages PARAM grads; embedding inputs here stand in
+    # for activations: expected relation is my_img.grad == world * ref-shard
+    # after loss-mean-normalization (loss normalized by B_local, ref by B_global)
+    assert torch.allclose(my_img.grad / world, img_r.grad[rank * b : (rank + 1) * b], atol=1e-5)
+    assert torch.allclose(my_txt.grad / world, txt_r.grad[rank * b : (rank + 1) * b], atol=1e-5)
+    # scale/bias: DP-average of per-rank grads == reference grad
+    st = scale.grad.clone()
+    dist.all_reduce(st)
+    assert torch.allclose(st / world, scale_r.grad, atol=1e-5)
+
+
+def test_dp_invariance_siglip():
+    spawn(_siglip_invariance, port=29515)

@@ -1,0 +1,250 @@
+"""HIP kernel numerics tests — each kernel vs the plain-PyTorch fp32
+reference (SURVEY §4 implication (a)). All require an MI355X (gfx950)."""
+
+import math
+
+import pytest
+import torch
+
+import jimm_amd  # noqa: F401
+from jimm_amd.ops import _backend
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    EXT = _backend.ext()  # fail loudly — never silently skip to eager on a GPU box
+else:
+    pytest.skip("no GPU", allow_module_level=True)
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+def rel_err(out, ref):
+    ref = ref.float()
+    return (out.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1e-6)
+
+
+# ---------------------------------------------------------------------------
+def test_mfma_probe():
+    torch.manual_seed(0)
+    # asymmetric operands — catches transposed layout assumptions (guide §3)
+    A = torch.randn(16, 32, device=dev()).bfloat16()
+    B = torch.randn(32, 16, device=dev()).bfloat16()
+    C = EXT.mfma_probe(A, B)
+    ref = A.float() @ B.float()
+    assert (C - ref).abs().max().item() < 1e-2, (C - ref).abs().max().item()
+
+
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape,eps", [((8, 197, 768), 1e-12), ((4, 77, 512), 1e-5), ((2, 256, 1152), 1e-6)])
+def test_layernorm_fwd(dtype, shape, eps):
+    torch.manual_seed(0)
+    x = torch.randn(shape, device=dev(), dtype=dtype)
+    w = torch.randn(shape[-1], device=dev(), dtype=dtype)
+    b = torch.randn(shape[-1], device=dev(), dtype=dtype)
+    y, mean, rstd = EXT.layernorm_fwd(x, w, b, eps)
+    ref = torch.nn.functional.layer_norm(x.float(), (shape[-1],), w.float(), b.float(), eps)
+    tol = 5e-6 if dtype == torch.float32 else 3e-2
+    assert rel_err(y, ref) < tol
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_layernorm_bwd(dtype):
+    torch.manual_seed(0)
+    shape, eps = (4, 197, 768), 1e-6
+    x = torch.randn(shape, device=dev(), dtype=dtype)
+    w = torch.randn(shape[-1], device=dev(), dtype=dtype)
+    b = torch.randn(shape[-1], device=dev(), dtype=dtype)
+    dy = torch.randn(shape, device=dev(), dtype=dtype)
+    y, mean, rstd = EXT.layernorm_fwd(x, w, b, eps)
+    dx, dw, db = EXT.layernorm_bwd(dy, x, w, mean, rstd)
+    xr = x.float().clone().requires_grad_(True)
+    wr = w.float().clone().requires_grad_(True)
+    br = b.float().clone().requires_grad_(True)
+    torch.nn.functional.layer_norm(xr, (shape[-1],), wr, br, eps).backward(dy.float())
+    tol = 1e-4 if dtype == torch.float32 else 5e-2
+    assert rel_err(dx, xr.grad) < tol
+    assert rel_err(dw, wr.grad) < tol
+    assert rel_err(db, br.grad) < tol
+
+
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("act", ["", "gelu", "gelu_tanh", "quickgelu"])
+def test_bias_act(act):
+    torch.manual_seed(0)
+    z = torch.randn(128, 768, device=dev(), dtype=torch.bfloat16)
+    z0 = z.clone()
+    bias = torch.randn(768, device=dev(), dtype=torch.bfloat16)
+    res = torch.randn(128, 768, device=dev(), dtype=torch.bfloat16)
+    y = EXT.bias_act_fwd(z, bias, act, res)
+    import jimm_amd.ops.functional as Fn
+
+    pre = z0.float() + bias.float()
+    ref = Fn._act(pre, act or None) + res.float()
+    assert rel_err(y, ref) < 2e-2
+    if act:
+        # z must now hold the pre-activation
+        assert rel_err(z, pre) < 2e-2
+        dy = torch.randn_like(z)
+        dz = EXT.act_bwd(dy, z, act)
+        pre_r = pre.clone().requires_grad_(True)
+        Fn._act(pre_r, act).backward(dy.float())
+        assert rel_err(dz, pre_r.grad) < 3e-2
+
+
+# ---------------------------------------------------------------------------
+def test_im2col_roundtrip():
+    torch.manual_seed(0)
+    img = torch.randn(2, 3, 64, 64, device=dev(), dtype=torch.bfloat16)
+    cols = EXT.im2col_patch(img, 16)
+    # reference unfold: (B, C*P*P, L) -> (B*L, C*P*P)
+    ref = torch.nn.functional.unfold(img.float(), 16, stride=16).transpose(1, 2).reshape(-1, 3 * 256)
+    assert (cols.float() - ref).abs().max().item() == 0.0
+    back = EXT.col2im_patch(cols, [2, 3, 64, 64], 16)
+    assert (back - img).abs().max().item() == 0.0
+
+
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("pdtype", [torch.float32, torch.bfloat16])
+def test_adam_step(pdtype):
+    torch.manual_seed(0)
+    from jimm_amd.train.adam import Adam
+
+    shapes = [(768,), (768, 768), (3072, 768), (197, 768)]
+    ps = [torch.randn(s, device=dev(), dtype=pdtype) for s in shapes]
+    gs = [torch.randn(s, device=dev(), dtype=pdtype) for s in shapes]
+    ps_ref = [p.detach().float().clone() for p in ps]
+    ms = [torch.zeros(s, device=dev()) for s in shapes]
+    vs = [torch.zeros(s, device=dev()) for s in shapes]
+    masters = [p.detach().float().clone() if pdtype != torch.float32 else None for p in ps]
+    ms_ref = [m.clone() for m in ms]
+    vs_ref = [v.clone() for v in vs]
+    for step in (1, 2, 3):
+        EXT.adam_step(ps, gs, ms, vs, masters, 1e-3, 0.9, 0.999, 1e-8, 0.01, step)
+        Adam._ref_step(ps_ref, [g.float() for g in gs], ms_ref, vs_ref, [None] * 4, 1e-3, 0.9, 0.999, 1e-8, 0.01, step)
+    for p, pr in zip(ps, ps_ref):
+        tol = 1e-6 if pdtype == torch.float32 else 1e-2
+        assert rel_err(p, pr) < tol
+
+
+# ---------------------------------------------------------------------------
+def _attn_ref(q, k, v, causal, scale):
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        Lq, Lk = s.shape[-2], s.shape[-1]
+        mask = torch.ones(Lq, Lk, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    return torch.matmul(torch.softmax(s, dim=-1), v.float())
+
+
+@pytest.mark.parametrize(
+    "B,H,Lq,Lk,causal",
+    [
+        (2, 3, 197, 197, False),   # ViT-B/16
+        (1, 2, 77, 77, True),      # CLIP text (causal)
+        (2, 2, 1, 256, False),     # MAP head cross-attn (K9)
+        (1, 2, 1024, 1024, False), # SigLIP2-large/16-512 bound
+        (2, 1, 50, 50, False),     # CLIP-B/32 vision
+        (1, 1, 130, 130, True),    # ragged, causal
+        (1, 1, 64, 64, False),     # single tile
+    ],
+)
+def test_attn_fwd(B, H, Lq, Lk, causal):
+    torch.manual_seed(0)
+    q = torch.randn(B, H, Lq, 64, device=dev()).bfloat16()
+    k = torch.randn(B, H, Lk, 64, device=dev()).bfloat16()
+    v = torch.randn(B, H, Lk, 64, device=dev()).bfloat16()
+    scale = 1.0 / math.sqrt(64)
+    o, lse = EXT.attn_fwd(q, k, v, causal, scale)
+    ref = _attn_ref(q, k, v, causal, scale)
+    assert rel_err(o, ref) < 3e-2, rel_err(o, ref)
+    # lse check
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.ones(Lq, Lk, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = torch.logsumexp(s, dim=-1)
+    assert (lse - lse_ref).abs().max().item() < 2e-2
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_attn_bwd(causal):
+    torch.manual_seed(0)
+    from jimm_amd import ops
+
+    B, H, L = 2, 2, 197
+    q = torch.randn(B, H, L, 64, device=dev()).bfloat16().requires_grad_(True)
+    k = torch.randn(B, H, L, 64, device=dev()).bfloat16().requires_grad_(True)
+    v = torch.randn(B, H, L, 64, device=dev()).bfloat16().requires_grad_(True)
+    do = torch.randn(B, H, L, 64, device=dev()).bfloat16()
+    out = ops.attention(q, k, v, causal=causal)
+    out.backward(do)
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    _attn_ref(qr, kr, vr, causal, 1.0 / math.sqrt(64)).backward(do.float())
+    assert rel_err(q.grad, qr.grad) < 5e-2
+    assert rel_err(k.grad, kr.grad) < 5e-2
+    assert rel_err(v.grad, vr.grad) < 5e-2
+
+
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize(
+    "M,N,K,act,has_res",
+    [
+        (256, 768, 768, "", False),
+        (197 * 4, 2304, 768, "", False),      # fused QKV (K4)
+        (512, 3072, 768, "gelu", False),       # fc1+gelu (K7)
+        (512, 768, 3072, "", True),            # fc2+residual (K8)
+        (100, 512, 2048, "quickgelu", False),  # CLIP text fc1, ragged M
+        (1000, 1000, 768, "", False),          # ragged N (classifier-ish)
+    ],
+)
+def test_linear_fwd_mfma(M, N, K, act, has_res):
+    torch.manual_seed(0)
+    import jimm_amd.ops.functional as Fn
+
+    x = torch.randn(M, K, device=dev()).bfloat16()
+    w = (torch.randn(N, K, device=dev()) / math.sqrt(K)).bfloat16()
+    bias = torch.randn(N, device=dev()).bfloat16()
+    res = torch.randn(M, N, device=dev()).bfloat16() if has_res else None
+    assert EXT.gemm_supported(M, N, K, str(x.dtype))
+    y, z = EXT.linear_fwd(x, w, bias, act, res, bool(act))
+    ref_pre = x.float() @ w.float().t() + bias.float()
+    ref = Fn._act(ref_pre, act or None)
+    if has_res:
+        ref = ref + res.float()
+    assert rel_err(y, ref) < 2e-2, rel_err(y, ref)
+    if act:
+        assert rel_err(z, ref_pre) < 2e-2
+
+
+# ---------------------------------------------------------------------------
+def test_model_train_step_gpu():
+    """End-to-end: one ViT train step in bf16 on the HIP path stays finite."""
+    torch.manual_seed(0)
+    from jimm_amd.train import SyntheticImages, TrainConfig, Trainer
+
+    model = jimm_amd.VisionTransformer(num_classes=10, img_size=64, patch_size=16, num_layers=2, num_heads=2, mlp_dim=256, hidden_size=128).to(dev(), torch.bfloat16)
+    tr = Trainer(model, TrainConfig(task="vit", lr=1e-3))
+    data = SyntheticImages(8, 64, 10, dev(), dtype=torch.bfloat16)
+    losses = []
+    for _ in range(5):
+        out = tr.train_step(next(iter(data)))
+        losses.append(out["loss"].item())
+    assert all(math.isfinite(l) for l in losses), losses
+
+
+def test_gpu_vs_cpu_model_parity():
+    """ViT-Tiny logits: GPU bf16 HIP path vs CPU fp32 reference path."""
+    torch.manual_seed(0)
+    model = jimm_amd.VisionTransformer(num_classes=10, img_size=64, patch_size=16, num_layers=2, num_heads=2, mlp_dim=256, hidden_size=128).eval()
+    x = torch.randn(4, 3, 64, 64)
+    with torch.no_grad():
+        ref = model(x)
+        gm = model.to(dev(), torch.bfloat16)
+        out = gm(x.to(dev(), torch.bfloat16))
+    assert (out.float().cpu() - ref).abs().max().item() < 0.15, (out.float().cpu() - ref).abs().max().item()
