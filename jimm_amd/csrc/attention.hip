@@ -42,7 +42,12 @@ constexpr int LDS_PITCH = D + 8;  // +8 shorts: bank-conflict pad for b128 reads
 template <bool CAUSAL>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
-    bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale) {
+    bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale, int H,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t o_sb, int64_t o_sh, int64_t o_sl) {
+  // (sb, sh, sl) = element strides of the (B,H,L,64) view; innermost dim is
+  // contiguous. Covers contiguous BHLD, (B,L,H,64) and fused-qkv (B,L,3,H,64)
+  // layouts with no permute copies.
   // LDS carve: K [64][72], V^T [64][72], P per-wave 4x[16][72] (shorts)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);                       // 64*72
@@ -55,12 +60,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int lo = lane & 15;   // column index inside a 16-wide fragment
   const int hi = lane >> 4;   // 0..3
   const int64_t bh = blockIdx.y;
+  const int64_t b = bh / H, h = bh % H;
   const int q0 = blockIdx.x * QBLK + wave * 16;  // this wave's first q row
 
-  const bf16* qp = q + bh * Lq * D;
-  const bf16* kp = k + bh * Lk * D;
-  const bf16* vp = v + bh * Lk * D;
-  bf16* op = o + bh * Lq * D;
+  const bf16* qp = q + b * q_sb + h * q_sh;
+  const bf16* kp = k + b * k_sb + h * k_sh;
+  const bf16* vp = v + b * v_sb + h * v_sh;
+  bf16* op = o + b * o_sb + h * o_sh;
 
   // ---- load Q fragments (B-operand of the swapped QK^T) -------------------
   // B[d][q] = Q[q0+lo][d = 32*s + hi*8 + j]
@@ -69,7 +75,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const int qrow = min(q0 + lo, Lq - 1);
 #pragma unroll
     for (int s = 0; s < 2; ++s)
-      qb[s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * D + 32 * s + hi * 8);
+      qb[s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
   }
 
   short* my_p = p_lds + wave * 16 * LDS_PITCH;
@@ -87,12 +93,12 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const int c0 = (tid % 4) * 16;    // 16 shorts per thread
       const int key = kv0 + row;
       if (key < Lk) {
-        const bf16x8_t k0 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * D + c0);
-        const bf16x8_t k1 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * D + c0 + 8);
+        const bf16x8_t k0 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0);
+        const bf16x8_t k1 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0 + 8);
         *reinterpret_cast<bf16x8_t*>(k_lds + row * LDS_PITCH + c0) = k0;
         *reinterpret_cast<bf16x8_t*>(k_lds + row * LDS_PITCH + c0 + 8) = k1;
-        const bf16x8_t v0 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * D + c0);
-        const bf16x8_t v1 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * D + c0 + 8);
+        const bf16x8_t v0 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + c0);
+        const bf16x8_t v1 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + c0 + 8);
 #pragma unroll
         for (int i = 0; i < 8; ++i) vt_lds[(c0 + i) * LDS_PITCH + row] = v0[i];
 #pragma unroll
@@ -196,7 +202,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     if (qrow >= Lq) continue;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      op[(int64_t)qrow * D + 16 * dt + lo] = f2bf(acc_o[dt][r] * invl_r[r]);
+      op[(int64_t)qrow * o_sl + 16 * dt + lo] = f2bf(acc_o[dt][r] * invl_r[r]);
   }
   if (hi == 0 && q0 + lo < Lq && wave * 16 + lo < QBLK)
     lse[bh * Lq + q0 + lo] = m_run + __logf(l_run);
@@ -206,30 +212,35 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     bool causal, double scale) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  // q,k,v: (B,H,L,64) views whose innermost dim is contiguous, head stride
+  // is 64 and L-row stride is uniform — covers both contiguous BHLD and the
+  // un-copied (B,L,H,64) / fused-qkv (B,L,3,H,64) layouts.
+  TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only, got ", q.scalar_type());
   TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "attn_fwd: (B,H,L,64) expected");
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
   TORCH_CHECK(k.size(0) == B && k.size(1) == H && v.size(2) == Lk);
-  auto o = torch::empty_like(q);
+  for (auto* t : {&q, &k, &v})
+    TORCH_CHECK(t->stride(3) == 1, "attn_fwd: innermost dim must be contiguous");
+  // O written in (B,L,H,64) memory order so the model's (B,L,H*64) reshape is
+  // a free view (no permute copy)
+  auto o_storage = torch::empty({B, Lq, H, 64}, q.options());
+  auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
   const dim3 grid((Lq + QBLK - 1) / QBLK, (unsigned)((int64_t)B * H));
   const size_t shmem = (KVBLK * LDS_PITCH + D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
-  if (causal) {
-    hipLaunchKernelGGL((attn_fwd_kernel<true>), grid, dim3(256), shmem, stream,
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(), Lq, Lk,
-                       (float)scale);
-  } else {
-    hipLaunchKernelGGL((attn_fwd_kernel<false>), grid, dim3(256), shmem, stream,
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(), Lq, Lk,
-                       (float)scale);
-  }
+#define ATTN_LAUNCH(C)                                                                     \
+  hipLaunchKernelGGL((attn_fwd_kernel<C>), grid, dim3(256), shmem, stream,                 \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                          \
+                     reinterpret_cast<bf16*>(o_storage.data_ptr()), lse.data_ptr<float>(), \
+                     Lq, Lk, (float)scale, H, q.stride(0), q.stride(1), q.stride(2),       \
+                     k.stride(0), k.stride(1), k.stride(2),                                \
+                     v.stride(0), v.stride(1), v.stride(2),                                \
+                     o.stride(0), o.stride(1), o.stride(2))
+  if (causal) ATTN_LAUNCH(true); else ATTN_LAUNCH(false);
+#undef ATTN_LAUNCH
   return {o, lse};
 }

@@ -16,6 +16,9 @@ void adam_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                double b2, double eps, double wd, int64_t step);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     bool causal, double scale);
+void attn_bwd_p(torch::Tensor s, torch::Tensor lse, bool causal, double scale);
+torch::Tensor attn_d(torch::Tensor dO, torch::Tensor O);
+void attn_ds(torch::Tensor dp, torch::Tensor p, torch::Tensor D, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
@@ -31,6 +34,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("col2im_patch", &col2im_patch, "patch-embed fold backward (K1/K15)");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam (K14)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward, head_dim 64 (K5)");
+  m.def("attn_bwd_p", &attn_bwd_p, "attention bwd: S -> P in place (K15)");
+  m.def("attn_d", &attn_d, "attention bwd: rowsum(dO*O) (K15)");
+  m.def("attn_ds", &attn_ds, "attention bwd: dP -> dS in place (K15)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
   m.def("linear_fwd", &linear_fwd, "MFMA GEMM + fused epilogue (K4/K6/K7/K8)");

@@ -69,10 +69,12 @@ class EncoderBlock(nn.Module):
 
         h = ops.layer_norm(x, self.norm2.weight, self.norm2.bias, self.eps)
         h = ops.linear(h, self.fc1.weight, self.fc1.bias, act=self.act)
-        h = self.dropout(h)
-        h = ops.linear(h, self.fc2.weight, self.fc2.bias)
-        h = self.dropout(h)
-        return x + h
+        if self.dropout.p > 0:
+            h = self.dropout(h)
+            h = ops.linear(h, self.fc2.weight, self.fc2.bias)
+            return x + self.dropout(h)
+        # dropout inactive: residual add fused into the fc2 epilogue (K8)
+        return ops.linear(h, self.fc2.weight, self.fc2.bias, residual=x)
 
 
 class Encoder(nn.Module):

@@ -95,7 +95,8 @@ class _AttentionFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        # kernel consumes strided (B,H,L,64) views directly (e.g. slices of
+        # the fused qkv projection) — no permute copies on the forward path
         o, lse = _backend.ext().attn_fwd(q, k, v, causal, scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
@@ -105,21 +106,18 @@ class _AttentionFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, do = o.contiguous(), do.contiguous()
         scale = ctx.scale
-        do = do.contiguous()
+        ext = _backend.ext()
         s = torch.matmul(q, k.transpose(-1, -2))  # (B,H,Lq,Lk) bf16
-        p = torch.exp(scale * s.float() - lse.unsqueeze(-1))
-        if ctx.causal:
-            Lq, Lk = p.shape[-2], p.shape[-1]
-            mask = torch.ones(Lq, Lk, dtype=torch.bool, device=p.device).tril()
-            p = p * mask
-        pb = p.to(q.dtype)
-        dcoef = (do.float() * o.float()).sum(-1, keepdim=True)  # rowsum(dO*O)
-        dp = torch.matmul(do, v.transpose(-1, -2)).float()
-        ds = (p * (dp - dcoef) * scale).to(q.dtype)
-        dq = torch.matmul(ds, k)
-        dk = torch.matmul(ds.transpose(-1, -2), q)
-        dv = torch.matmul(pb.transpose(-1, -2), do)
+        ext.attn_bwd_p(s, lse, ctx.causal, scale)  # in place: s -> P (bf16)
+        dcoef = ext.attn_d(do, o)                  # (rows,) fp32 rowsum(dO*O)
+        dp = torch.matmul(do, v.transpose(-1, -2))
+        ext.attn_ds(dp, s, dcoef, scale)           # in place: dp -> dS (bf16)
+        dq = torch.matmul(dp, k)
+        dk = torch.matmul(dp.transpose(-1, -2), q)
+        dv = torch.matmul(s.transpose(-1, -2), do)
         return dq, dk, dv, None, None
 
 
