@@ -2,16 +2,23 @@
 //
 // linear_fwd computes y = act(x @ w^T + bias) [+ residual] for x (M,K)
 // row-major and w (N,K) row-major (torch Linear convention) — an "NT" GEMM
-// mapped onto v_mfma_f32_16x16x32_bf16 tiles.
+// on v_mfma_f32_16x16x32_bf16 tiles.
 //
-// v1 structure (CDNA-guide §5 ladder step ~2): 128x128 macro-tile, 4 waves,
-// each wave a 64x64 sub-tile of 4x4 16x16 fragments; K staged in LDS
-// double-buffered; epilogue applies bias/act/residual in-register before the
-// bf16 store. The 256²/8-phase schedule (guide §5 template) is the planned
-// upgrade once the baseline is profiled.
+// Fast path (full 128x128 tiles, K%64==0 — every hot shape in the model zoo:
+// ViT/CLIP/SigLIP widths are multiples of 128 and M = B*L*? lands on
+// multiples of 128 for the bench batch sizes):
+//   * 128x128 macro-tile, BK=64, 4 waves, 64x64 per wave (4x4 fragments);
+//   * async global->LDS staging via __builtin_amdgcn_global_load_lds
+//     width 16 (CDNA guide §5 ladder step 3: +69% over register staging);
+//   * LDS image XOR-swizzled (chunk ^= row&7, 16B chunks) with the inverse
+//     swizzle applied to the per-lane SOURCE address (guide rule 21) so the
+//     ds_read_b128 fragment reads are bank-conflict-free;
+//   * double-buffered LDS; stage(t+1) issued BEFORE compute(t) (minimum
+//     2-phase recipe); __syncthreads() drains the in-flight DMA (vmcnt(0)
+//     is emitted by the compiler with a glds outstanding);
+//   * XCD-aware bijective blockIdx remap (guide T1) for L2 tile locality.
 //
-// gemm_supported() gates dispatch: the python layer falls back to
-// rocBLAS (torch.matmul) + the fused bias_act kernel when unsupported.
+// Ragged shapes fall back to the generic guarded kernel below.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -26,9 +33,125 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
 constexpr int BM = 128, BN = 128, BK = 64;
-constexpr int PITCH = BK + 8;  // shorts; pad keeps b128 row reads conflict-light
+constexpr int PITCH = BK + 8;  // generic path only
 
-// act codes shared with elementwise.hip via common.h
+// ---------------------------------------------------------------------------
+// fast path: 128x128, glds staging, swizzled LDS
+// ---------------------------------------------------------------------------
+
+// LDS linear image: [128 rows][64 shorts] = 128 B/row, 16 KiB per operand
+// tile; chunk c (16 B) of row r lives at byte r*128 + (c ^ (r&7))*16.
+
+__device__ __forceinline__ void stage_tile_glds(const bf16* __restrict__ gsrc, int64_t ldg,
+                                                short* lds_base, int tid) {
+  // 256 threads stage 128x64 shorts (16 KiB): 4 rounds of 4 KiB; each wave
+  // writes 1 KiB linearly at (round*4K + wave*1K); lane l covers offset
+  // l*16. Source address carries the inverse swizzle.
+  const int wave = tid / WAVE;
+  const int lane = tid % WAVE;
+#pragma unroll
+  for (int round = 0; round < 4; ++round) {
+    const int off = round * 4096 + wave * 1024 + lane * 16;  // byte offset in image
+    const int row = off >> 7;              // /128
+    const int chunk = (off >> 4) & 7;      // 16B chunk within row
+    const int src_chunk = chunk ^ (row & 7);
+    const bf16* g = gsrc + (int64_t)row * ldg + src_chunk * 8;
+    // C-style casts perform the addrspace conversion (generic->AS1/AS3);
+    // reinterpret_cast refuses (same idiom as CK's c_style_pointer_cast)
+    typedef const __attribute__((address_space(1))) unsigned int* gp_t;
+    typedef __attribute__((address_space(3))) unsigned int* lp_t;
+    __builtin_amdgcn_global_load_lds((gp_t)(const void*)g,
+                                     (lp_t)(void*)(reinterpret_cast<char*>(lds_base) + round * 4096 + wave * 1024),
+                                     16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ bf16x8_t lds_read_frag(const short* base, int row, int chunk) {
+  const int byte = (row << 7) + ((chunk ^ (row & 7)) << 4);
+  return *reinterpret_cast<const bf16x8_t*>(reinterpret_cast<const char*>(base) + byte);
+}
+
+template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE>
+__global__ __launch_bounds__(256) void gemm_nt_fast_kernel(
+    const bf16* __restrict__ X, const bf16* __restrict__ W, const float* __restrict__ bias,
+    const bf16* __restrict__ res, bf16* __restrict__ Y, bf16* __restrict__ Z,
+    int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* smem_s = reinterpret_cast<short*>(smem);
+  // buffers: x tiles at [0,8192) and [8192,16384); w tiles at +16384
+  auto xs = [&](int buf) { return smem_s + buf * 8192; };
+  auto ws = [&](int buf) { return smem_s + 16384 + buf * 8192; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid % WAVE;
+  const int wave = tid / WAVE;
+  const int lo = lane & 15, hi = lane >> 4;
+
+  // XCD-aware bijective remap of the linear tile id (guide T1)
+  const int mt = M / BM, nt = N / BN;
+  const int nwg = mt * nt;
+  int wg = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int m0 = (wg / nt) * BM;
+  const int n0 = (wg % nt) * BN;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+
+  f32x4_t acc[4][4] = {};
+
+  stage_tile_glds(X + (int64_t)m0 * K, K, xs(0), tid);
+  stage_tile_glds(W + (int64_t)n0 * K, K, ws(0), tid);
+  __syncthreads();
+
+  const int nk = K / BK;
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < nk) {
+      stage_tile_glds(X + (int64_t)m0 * K + (kt + 1) * BK, K, xs(buf ^ 1), tid);
+      stage_tile_glds(W + (int64_t)n0 * K + (kt + 1) * BK, K, ws(buf ^ 1), tid);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_t xa[4], wb[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) xa[mi] = lds_read_frag(xs(buf), wm + 16 * mi + lo, 4 * ks + hi);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) wb[ni] = lds_read_frag(ws(buf), wn + 16 * ni + lo, 4 * ks + hi);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = MFMA16(xa[mi], wb[ni], acc[mi][ni]);
+    }
+    __syncthreads();  // also drains the in-flight global_load_lds (vmcnt 0)
+  }
+
+  // epilogue: C rows = m (hi*4+r of each 16-fragment), col = n (lo)
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = m0 + wm + 16 * mi + hi * 4 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int n = n0 + wn + 16 * ni + lo;
+        float vpre = acc[mi][ni][r];
+        if (HAS_BIAS) vpre += bias[n];
+        if (SAVE_PRE) Z[(int64_t)m * N + n] = f2bf(vpre);
+        float vy = act_fwd(vpre, ACT);
+        if (HAS_RES) vy += bf2f(res[(int64_t)m * N + n]);
+        Y[(int64_t)m * N + n] = f2bf(vy);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// generic (ragged) path — guarded staging, padded LDS, register staging
+// ---------------------------------------------------------------------------
+
 template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE>
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ W, const float* __restrict__ bias,
@@ -43,25 +166,21 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   const int lane = tid % WAVE;
   const int lo = lane & 15, hi = lane >> 4;
 
-  const int m0 = blockIdx.x * BM;  // block row
-  const int n0 = blockIdx.y * BN;  // block col
-  // wave sub-tile: 2x2 wave grid, each wave 64x64
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
   const int wm = (wave >> 1) * 64;
   const int wn = (wave & 1) * 64;
 
-  f32x4_t acc[4][4] = {};  // [mi][ni] 16x16 fragments
+  f32x4_t acc[4][4] = {};
 
   auto stage = [&](int buf, int k0) {
-    // 256 threads stage BM x BK of X and BN x BK of W (bf16, row-major K-major)
-    // each thread: 128*64/256 = 32 shorts of each = 2 x bf16x8 x 2
     short* xd = xs + buf * BM * PITCH;
     short* wd = ws + buf * BN * PITCH;
-    const int row = tid / 2;           // 0..127
-    const int c0 = (tid & 1) * 32;     // two 32-short halves
+    const int row = tid / 2;
+    const int c0 = (tid & 1) * 32;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int cc = c0 + h * 16;
-      // X rows are guarded (M may be ragged); W rows assumed N%? guarded too
       if (m0 + row < M) {
         *reinterpret_cast<bf16x8_t*>(xd + row * PITCH + cc) =
             *reinterpret_cast<const bf16x8_t*>(X + (int64_t)(m0 + row) * K + k0 + cc);
@@ -91,7 +210,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     short* xd = xs + buf * BM * PITCH;
     short* wd = ws + buf * BN * PITCH;
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {  // two 32-deep steps per BK
+    for (int ks = 0; ks < 2; ++ks) {
       bf16x8_t xa[4], wb[4];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
@@ -108,9 +227,6 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     __syncthreads();
   }
 
-  // epilogue: C fragment rows = x rows (m), cols = w rows (n)
-  // A=x fragment rows are A[l&15] -> m index; B=w fragment cols l&15 -> n.
-  // C[i][j]: lane holds rows (hi*4+r) of m-tile, col lo of n-tile.
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -160,18 +276,29 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
   else if (act == "quickgelu") act_code = ACT_QUICKGELU;
   else TORCH_CHECK(act.empty(), "unknown act ", act);
 
-  const dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-  const size_t shmem = 2 * (BM + BN) * PITCH * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
   const bf16* resp = residual ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr;
   const float* biasp = bf ? bf->data_ptr<float>() : nullptr;
   bf16* zp = save_z ? reinterpret_cast<bf16*>(z.data_ptr()) : nullptr;
+  const bool fast = (M % BM == 0) && (N % BN == 0);
 
 #define LAUNCH(ACTC, HB, HR, SP)                                                          \
-  hipLaunchKernelGGL((gemm_nt_kernel<ACTC, HB, HR, SP>), grid, dim3(256), shmem, stream,  \
-                     reinterpret_cast<const bf16*>(x.data_ptr()),                         \
-                     reinterpret_cast<const bf16*>(w.data_ptr()), biasp, resp,            \
-                     reinterpret_cast<bf16*>(y.data_ptr()), zp, M, N, K)
+  do {                                                                                    \
+    if (fast) {                                                                           \
+      hipLaunchKernelGGL((gemm_nt_fast_kernel<ACTC, HB, HR, SP>),                         \
+                         dim3((M / BM) * (N / BN)), dim3(256), 4 * 8192 * sizeof(short),  \
+                         stream, reinterpret_cast<const bf16*>(x.data_ptr()),             \
+                         reinterpret_cast<const bf16*>(w.data_ptr()), biasp, resp,        \
+                         reinterpret_cast<bf16*>(y.data_ptr()), zp, M, N, K);             \
+    } else {                                                                              \
+      hipLaunchKernelGGL((gemm_nt_kernel<ACTC, HB, HR, SP>),                              \
+                         dim3((M + BM - 1) / BM, (N + BN - 1) / BN), dim3(256),           \
+                         2 * (BM + BN) * PITCH * sizeof(short), stream,                   \
+                         reinterpret_cast<const bf16*>(x.data_ptr()),                     \
+                         reinterpret_cast<const bf16*>(w.data_ptr()), biasp, resp,        \
+                         reinterpret_cast<bf16*>(y.data_ptr()), zp, M, N, K);             \
+    }                                                                                     \
+  } while (0)
 #define DISPATCH_ACT(HB, HR, SP)                                                          \
   switch (act_code) {                                                                     \
     case ACT_NONE: LAUNCH(ACT_NONE, HB, HR, SP); break;                                   \
