@@ -1,7 +1,8 @@
 """Microbench: in-house MFMA GEMM (csrc/gemm.hip) vs rocBLAS (torch.matmul)
 on the model-zoo hot shapes. Run on an MI355X via gpurun."""
 
-import time
+import os, sys, time
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
