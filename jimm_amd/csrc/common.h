@@ -97,3 +97,37 @@ __device__ __forceinline__ float act_grad(float x, int act) {
       return 1.0f;
   }
 }
+
+// ---- vectorized load/store helpers (G13: hipcc does not auto-vectorize
+// bf16 loads; 8-16 B per lane is the coalescing sweet spot) ----------------
+template <int V, typename T>
+__device__ __forceinline__ void vload_f32(const T* p, float (&o)[V]) {
+  if constexpr (sizeof(T) == 2) {
+    typedef short vt __attribute__((ext_vector_type(V)));
+    vt v = *reinterpret_cast<const vt*>(p);
+#pragma unroll
+    for (int j = 0; j < V; ++j) o[j] = bfs2f((short)v[j]);
+  } else {
+    typedef float vt __attribute__((ext_vector_type(V)));
+    vt v = *reinterpret_cast<const vt*>(p);
+#pragma unroll
+    for (int j = 0; j < V; ++j) o[j] = v[j];
+  }
+}
+
+template <int V, typename T>
+__device__ __forceinline__ void vstore_f32(T* p, const float (&in)[V]) {
+  if constexpr (sizeof(T) == 2) {
+    typedef short vt __attribute__((ext_vector_type(V)));
+    vt v;
+#pragma unroll
+    for (int j = 0; j < V; ++j) v[j] = f2bfs(in[j]);
+    *reinterpret_cast<vt*>(p) = v;
+  } else {
+    typedef float vt __attribute__((ext_vector_type(V)));
+    vt v;
+#pragma unroll
+    for (int j = 0; j < V; ++j) v[j] = in[j];
+    *reinterpret_cast<vt*>(p) = v;
+  }
+}

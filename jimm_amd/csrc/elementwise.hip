@@ -31,13 +31,53 @@ __global__ void bias_act_kernel(T* __restrict__ z, const T* __restrict__ bias,
   }
 }
 
+template <typename T, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE>
+__global__ void bias_act_vec_kernel(T* __restrict__ z, const T* __restrict__ bias,
+                                    const T* __restrict__ res, T* __restrict__ y,
+                                    int64_t nv, int Mv, int act) {
+  // vectorized x8 variant; requires M % 8 == 0 (Mv = M/8, nv = n/8)
+  constexpr int V = 8;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float zv[V], o[V];
+    vload_f32<V>(z + i * V, zv);
+    if (HAS_BIAS) {
+      float bv[V];
+      vload_f32<V>(bias + (i % Mv) * V, bv);
+#pragma unroll
+      for (int j = 0; j < V; ++j) zv[j] += bv[j];
+    }
+    if (SAVE_PRE) vstore_f32<V>(z + i * V, zv);
+#pragma unroll
+    for (int j = 0; j < V; ++j) o[j] = act_fwd(zv[j], act);
+    if (HAS_RES) {
+      float rv[V];
+      vload_f32<V>(res + i * V, rv);
+#pragma unroll
+      for (int j = 0; j < V; ++j) o[j] += rv[j];
+    }
+    vstore_f32<V>(y + i * V, o);
+  }
+}
+
 template <typename T>
 __global__ void act_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ z,
                                T* __restrict__ dz, int64_t n, int act) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+  constexpr int V = 8;
+  const int64_t nv = n / V;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
        i += (int64_t)gridDim.x * blockDim.x) {
-    dz[i] = (T)((float)dy[i] * act_grad((float)z[i], act));
+    float dyv[V], zv[V], o[V];
+    vload_f32<V>(dy + i * V, dyv);
+    vload_f32<V>(z + i * V, zv);
+#pragma unroll
+    for (int j = 0; j < V; ++j) o[j] = dyv[j] * act_grad(zv[j], act);
+    vstore_f32<V>(dz + i * V, o);
   }
+  // scalar tail
+  for (int64_t i = nv * V + (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dz[i] = (T)((float)dy[i] * act_grad((float)z[i], act));
 }
 
 // cols[b*h*w + ph*w + pw][c*P*P + i*P + j] = img[b][c][ph*P+i][pw*P+j]
@@ -99,9 +139,16 @@ void launch_bias_act(torch::Tensor& z, const c10::optional<torch::Tensor>& bias,
   const T* rp = res ? reinterpret_cast<const T*>(res->data_ptr()) : nullptr;
   T* zp = reinterpret_cast<T*>(z.data_ptr());
   T* yp = reinterpret_cast<T*>(y.data_ptr());
+  const bool vec = (M % 8) == 0;
 #define DISPATCH(HB, HR, SP)                                                              \
-  hipLaunchKernelGGL((bias_act_kernel<T, HB, HR, SP>), dim3(grid), dim3(block), 0,        \
-                     stream, zp, bp, rp, yp, n, M, act)
+  do {                                                                                    \
+    if (vec)                                                                              \
+      hipLaunchKernelGGL((bias_act_vec_kernel<T, HB, HR, SP>), dim3(grid), dim3(block),   \
+                         0, stream, zp, bp, rp, yp, n / 8, M / 8, act);                   \
+    else                                                                                  \
+      hipLaunchKernelGGL((bias_act_kernel<T, HB, HR, SP>), dim3(grid), dim3(block), 0,    \
+                         stream, zp, bp, rp, yp, n, M, act);                              \
+  } while (0)
   const bool hb = bias.has_value(), hr = res.has_value();
   if (hb && hr && save_pre) DISPATCH(true, true, true);
   else if (hb && hr && !save_pre) DISPATCH(true, true, false);

@@ -32,13 +32,13 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
     const T* xr = x + row * H;
     T* yr = y + row * H;
     float sum = 0.f, sumsq = 0.f;
-    // per-lane strided vector loads
     for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+      float xv[VEC];
+      vload_f32<VEC>(xr + i, xv);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float v = (float)xr[i + k];
-        sum += v;
-        sumsq += v * v;
+        sum += xv[k];
+        sumsq += xv[k] * xv[k];
       }
     }
     sum = wave_reduce_sum(sum);
@@ -51,11 +51,13 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
       rstd_out[row] = rstd;
     }
     for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+      float xv[VEC], wv[VEC], bv[VEC], yv[VEC];
+      vload_f32<VEC>(xr + i, xv);
+      vload_f32<VEC>(w + i, wv);
+      vload_f32<VEC>(b + i, bv);
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        float v = (float)xr[i + k];
-        yr[i + k] = (T)((v - mean) * rstd * w[i + k] + b[i + k]);
-      }
+      for (int k = 0; k < VEC; ++k) yv[k] = (xv[k] - mean) * rstd * wv[k] + bv[k];
+      vstore_f32<VEC>(yr + i, yv);
     }
   }
 }
@@ -88,10 +90,14 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     // c1 = mean(dy*w), c2 = mean(dy*w*xhat)
     float c1 = 0.f, c2 = 0.f;
     for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+      float dyv[VEC], xv[VEC], wv[VEC];
+      vload_f32<VEC>(dyr + i, dyv);
+      vload_f32<VEC>(xr + i, xv);
+      vload_f32<VEC>(w + i, wv);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float g = (float)dyr[i + k] * w[i + k];
-        float xhat = ((float)xr[i + k] - mu) * rs;
+        float g = dyv[k] * wv[k];
+        float xhat = (xv[k] - mu) * rs;
         c1 += g;
         c2 += g * xhat;
       }
@@ -99,15 +105,18 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     c1 = wave_reduce_sum(c1) / H;
     c2 = wave_reduce_sum(c2) / H;
     for (int i = lane * VEC; i < H; i += WAVE * VEC) {
+      float dyv[VEC], xv[VEC], wv[VEC], dxv[VEC];
+      vload_f32<VEC>(dyr + i, dyv);
+      vload_f32<VEC>(xr + i, xv);
+      vload_f32<VEC>(w + i, wv);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        float gy = (float)dyr[i + k];
-        float g = gy * w[i + k];
-        float xhat = ((float)xr[i + k] - mu) * rs;
-        dxr[i + k] = (T)(rs * (g - c1 - xhat * c2));
-        s_dw[i + k] += gy * xhat;
-        s_db[i + k] += gy;
+        float xhat = (xv[k] - mu) * rs;
+        dxv[k] = rs * (dyv[k] * wv[k] - c1 - xhat * c2);
+        s_dw[i + k] += dyv[k] * xhat;
+        s_db[i + k] += dyv[k];
       }
+      vstore_f32<VEC>(dxr + i, dxv);
     }
   }
   __syncthreads();

@@ -61,9 +61,8 @@ class EncoderBlock(nn.Module):
         B, L, H = x.shape
         h = ops.layer_norm(x, self.norm1.weight, self.norm1.bias, self.eps)
         qkv = ops.linear(h, self.qkv.weight, self.qkv.bias)  # (B, L, 3H)
-        qkv = qkv.view(B, L, 3, self.num_heads, self.head_dim).permute(2, 0, 3, 1, 4)
-        q, k, v = qkv[0], qkv[1], qkv[2]  # (B, heads, L, d)
-        o = ops.attention(q, k, v, causal=self.causal)
+        qkv = qkv.view(B, L, 3, self.num_heads, self.head_dim)
+        o = ops.attention_qkv(qkv, causal=self.causal)  # (B, heads, L, d)
         o = o.transpose(1, 2).reshape(B, L, H)
         x = ops.linear(o, self.proj.weight, self.proj.bias, residual=x)
 

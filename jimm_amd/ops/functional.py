@@ -121,6 +121,58 @@ class _AttentionFn(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
+class _AttentionQKVFn(torch.autograd.Function):
+    """Self-attention taking the fused QKV projection output directly.
+
+    Input qkv: (B, L, 3, H, D) (a view of the (B,L,3H*D) GEMM output).
+    Avoids autograd's per-slice backward (which materializes three
+    full-size zero tensors + scatter + add per layer): backward assembles
+    dqkv with three strided copies into one buffer.
+    Output: (B, H, L, D) view of (B,L,H,D) storage (free reshape to (B,L,HD)).
+    """
+
+    @staticmethod
+    def forward(ctx, qkv, causal, scale):
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        o, lse = _backend.ext().attn_fwd(q, k, v, causal, scale)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o, lse = ctx.saved_tensors
+        scale = ctx.scale
+        ext = _backend.ext()
+        q = qkv[:, :, 0].transpose(1, 2).contiguous()
+        k = qkv[:, :, 1].transpose(1, 2).contiguous()
+        v = qkv[:, :, 2].transpose(1, 2).contiguous()
+        o, do = o.contiguous(), do.contiguous()
+        s = torch.matmul(q, k.transpose(-1, -2))
+        ext.attn_bwd_p(s, lse, ctx.causal, scale)
+        dcoef = ext.attn_d(do, o)
+        dp = torch.matmul(do, v.transpose(-1, -2))
+        ext.attn_ds(dp, s, dcoef, scale)
+        dqkv = torch.empty_like(qkv)
+        dqkv[:, :, 0].transpose(1, 2).copy_(torch.matmul(dp, k))
+        dqkv[:, :, 1].transpose(1, 2).copy_(torch.matmul(dp.transpose(-1, -2), q))
+        dqkv[:, :, 2].transpose(1, 2).copy_(torch.matmul(s.transpose(-1, -2), do))
+        return dqkv, None, None
+
+
+def attention_qkv(qkv: torch.Tensor, *, causal: bool = False, scale: float | None = None) -> torch.Tensor:
+    """qkv (B, L, 3, H, D) -> (B, H, L, D). GPU fast path for self-attention."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(qkv.shape[-1])
+    if _backend.use_hip(qkv):
+        return _AttentionQKVFn.apply(qkv, causal, scale)
+    q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+    return attention(q, k, v, causal=causal, scale=scale)
+
+
 def attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -218,6 +270,7 @@ __all__ = [
     "quickgelu",
     "layer_norm",
     "attention",
+    "attention_qkv",
     "linear",
     "patch_embed",
     "add_cls_pos",
