@@ -1,0 +1,463 @@
+// K15 — fused flash-attention backward for CDNA4 (gfx950), bf16, head_dim 64.
+//
+// Replaces the recompute-P composite (attention_bwd.hip + 5 rocBLAS batched
+// GEMMs + .contiguous() copies) that the round-1 profile measured at
+// ~15 ms/step for ViT-B/16 train (profiles/r01_NOTES.md). Two kernels, both
+// recomputing P = exp(scale*S - lse) from the saved forward LSE:
+//
+//   attn_bwd_dkv_kernel — grid over (kv_tile, B*H): each workgroup owns 64
+//     keys, loops over q tiles, accumulates dK/dV in MFMA registers, writes
+//     them STRIDED straight into the fused dqkv buffer (no assembly copies).
+//   attn_bwd_dq_kernel  — grid over (q_tile, B*H): each workgroup owns 64
+//     q rows, loops over kv tiles, accumulates dQ. No atomics anywhere.
+//
+//   attn_d2_kernel      — D[b,h,l] = rowsum(dO * O), stride-aware fp32
+//     (consumed by both kernels for dS = P*(dP - D)*scale).
+//
+// All tensor arguments are (B,H,L,64) *views* with arbitrary (b,h,l) strides
+// and a contiguous innermost dim — q/k/v slices of the fused (B,L,3,H,64)
+// QKV projection and (B,L,H,64)-storage dO/O are consumed with zero permute
+// copies (the round-1 profile showed 3.9 ms/step of pure copy kernels).
+//
+// MFMA fragment conventions identical to attention.hip (verified on hardware
+// by csrc/probe.hip + tests/test_kernels_gpu.py::test_mfma_probe):
+//   A[i][k]: lane l holds A[l&15][(l>>4)*8 + j], j=0..7
+//   B[k][j]: lane l holds B[(l>>4)*8 + j][l&15]
+//   C[i][j]: lane l holds rows (l>>4)*4+r (r=0..3), col l&15
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+#define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
+
+constexpr int BLK = 64;            // q rows / keys per workgroup tile
+constexpr int D = 64;              // head_dim (checked host-side)
+constexpr int PITCH = D + 8;       // bank-conflict pad for b128 reads
+
+// ---------------------------------------------------------------------------
+// D = rowsum(dO * O), stride-aware
+// ---------------------------------------------------------------------------
+
+__global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
+                               float* __restrict__ Dv, int H, int L, int64_t nrows,
+                               int64_t do_sb, int64_t do_sh, int64_t do_sl,
+                               int64_t o_sb, int64_t o_sh, int64_t o_sl) {
+  // one wave per (b,h,l) row; 8-wide vector loads
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int wpb = blockDim.x / WAVE;
+  for (int64_t row = (int64_t)blockIdx.x * wpb + wave; row < nrows;
+       row += (int64_t)gridDim.x * wpb) {
+    const int64_t b = row / ((int64_t)H * L);
+    const int h = (int)((row / L) % H);
+    const int l = (int)(row % L);
+    const bf16* a = dO + b * do_sb + h * do_sh + (int64_t)l * do_sl;
+    const bf16* o = O + b * o_sb + h * o_sh + (int64_t)l * o_sl;
+    float acc = 0.f;
+    if (lane < 8) {  // 8 lanes x 8 elements = 64
+      float av[8], ov[8];
+      vload_f32<8>(a + lane * 8, av);
+      vload_f32<8>(o + lane * 8, ov);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += av[j] * ov[j];
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) Dv[row] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dK/dV kernel: workgroup owns keys [kv0, kv0+64); loops q tiles
+// ---------------------------------------------------------------------------
+
+template <bool CAUSAL>
+__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
+    bf16* __restrict__ dk, bf16* __restrict__ dv, int Lq, int Lk, float scale, int H,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
+    int64_t dk_sb, int64_t dk_sh, int64_t dk_sl, int64_t dv_sb, int64_t dv_sh, int64_t dv_sl) {
+  // LDS: K [64][72], V [64][72] staged once (row-major); Q^T [64][72] and
+  // dO^T [64][72] re-staged per q tile (scatter-transpose, like the forward
+  // kernel's V^T); per-wave P/dS tile 4x[16][72]. Total ~46 KiB.
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = reinterpret_cast<short*>(smem);
+  short* v_lds = k_lds + BLK * PITCH;
+  short* qt_lds = v_lds + BLK * PITCH;
+  short* dot_lds = qt_lds + D * PITCH;
+  short* p_lds = dot_lds + D * PITCH;
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lo = lane & 15, hi = lane >> 4;
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / H, h = bh % H;
+  const int kv0 = blockIdx.x * BLK;
+
+  const bf16* qp = q + b * q_sb + h * q_sh;
+  const bf16* kp = k + b * k_sb + h * k_sh;
+  const bf16* vp = v + b * v_sb + h * v_sh;
+  const bf16* dop = dO + b * do_sb + h * do_sh;
+  const float* lsep = lse + bh * Lq;
+  const float* dvp_row = Dv + bh * Lq;
+
+  // ---- stage K and V tiles once (row-major, zero-padded tail) -------------
+  {
+    const int row = tid / 4;
+    const int c0 = (tid % 4) * 16;
+    const int key = kv0 + row;
+    if (key < Lk) {
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+        *reinterpret_cast<bf16x8_t*>(k_lds + row * PITCH + c0 + hh * 8) =
+            *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0 + hh * 8);
+        *reinterpret_cast<bf16x8_t*>(v_lds + row * PITCH + c0 + hh * 8) =
+            *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + c0 + hh * 8);
+      }
+    } else {
+      for (int i = 0; i < 16; ++i) k_lds[row * PITCH + c0 + i] = 0;
+      for (int i = 0; i < 16; ++i) v_lds[row * PITCH + c0 + i] = 0;
+    }
+  }
+  __syncthreads();
+
+  // A-fragments of K and V for this wave's 16 keys (fixed for whole kernel)
+  bf16x8_t ka[2], va[2];
+#pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    ka[s] = *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * wave + lo) * PITCH + 32 * s + hi * 8);
+    va[s] = *reinterpret_cast<const bf16x8_t*>(v_lds + (16 * wave + lo) * PITCH + 32 * s + hi * 8);
+  }
+  __syncthreads();  // qt/dot staging below reuses nothing, but keep tiles safe
+
+  short* my_p = p_lds + wave * 16 * PITCH;
+
+  f32x4_t acc_dk[4] = {};  // rows key = 16*wave + hi*4+r, cols d = 16*dt + lo
+  f32x4_t acc_dv[4] = {};
+
+  const int q_start = CAUSAL ? (kv0 / BLK) * BLK : 0;
+
+  for (int q0 = q_start; q0 < Lq; q0 += BLK) {
+    // ---- stage Q^T and dO^T for this q tile (scatter-transpose) ----------
+    {
+      const int row = tid / 4;          // q index within tile
+      const int c0 = (tid % 4) * 16;
+      const int qi = q0 + row;
+      if (qi < Lq) {
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) {
+          const bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qi * q_sl + c0 + hh * 8);
+          const bf16x8_t dv8 = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qi * do_sl + c0 + hh * 8);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) qt_lds[(c0 + hh * 8 + i) * PITCH + row] = qv[i];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) dot_lds[(c0 + hh * 8 + i) * PITCH + row] = dv8[i];
+        }
+      } else {
+        for (int i = 0; i < 16; ++i) qt_lds[(c0 + i) * PITCH + row] = 0;
+        for (int i = 0; i < 16; ++i) dot_lds[(c0 + i) * PITCH + row] = 0;
+      }
+    }
+
+    // ---- B-fragments of Q^T and dO^T from global (b128, L2-resident) ------
+    // B[k=d][j=q] element = Q[q0 + 16*qt + lo][32*s + hi*8 + jj]
+    bf16x8_t qb[4][2], dob[4][2];
+#pragma unroll
+    for (int qt = 0; qt < 4; ++qt) {
+      const int qi = min(q0 + 16 * qt + lo, Lq - 1);  // clamped; masked via P=0
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qi * q_sl + 32 * s + hi * 8);
+        dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qi * do_sl + 32 * s + hi * 8);
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K . Q^T ; P^T = exp(scale*S^T - lse[q]) -------------------
+    // C rows = key = 16*wave + hi*4+r, cols = q = 16*qt + lo
+    bf16x4 ds_stash[4];
+#pragma unroll
+    for (int qt = 0; qt < 4; ++qt) {
+      f32x4_t sc = {};
+#pragma unroll
+      for (int s = 0; s < 2; ++s) sc = MFMA16(ka[s], qb[qt][s], sc);
+      const int qi = q0 + 16 * qt + lo;
+      const float l = lsep[min(qi, Lq - 1)];
+      f32x4_t dpc = {};
+#pragma unroll
+      for (int s = 0; s < 2; ++s) dpc = MFMA16(va[s], dob[qt][s], dpc);
+      const float dcoef = dvp_row[min(qi, Lq - 1)];
+      bf16x4 pk, dsk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = kv0 + 16 * wave + hi * 4 + r;
+        float p = __expf(sc[r] * scale - l);
+        if (qi >= Lq || (CAUSAL && key > qi)) p = 0.f;
+        pk[r] = f2bfs(p);
+        dsk[r] = f2bfs(p * (dpc[r] - dcoef) * scale);
+      }
+      // route P^T through the per-wave LDS tile now (A-operand of the dV
+      // mfma); dS^T is stashed in registers and reuses the same tile after
+      // the dV mfma has consumed P^T.
+#pragma unroll
+      for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = pk[r];
+      ds_stash[qt] = dsk;
+    }
+
+    // ---- dV += P^T . dO  (A = P^T via LDS, B = dO^T rows) ----------------
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const bf16x8_t bfrag =
+            *reinterpret_cast<const bf16x8_t*>(dot_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
+        acc_dv[dt] = MFMA16(pa, bfrag, acc_dv[dt]);
+      }
+    }
+
+    // ---- overwrite the wave tile with dS^T, then dK += dS^T . Q ----------
+#pragma unroll
+    for (int qt = 0; qt < 4; ++qt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = ds_stash[qt][r];
+    }
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const bf16x8_t bfrag =
+            *reinterpret_cast<const bf16x8_t*>(qt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
+        acc_dk[dt] = MFMA16(dsa, bfrag, acc_dk[dt]);
+      }
+    }
+    __syncthreads();  // qt_lds/dot_lds re-staged next iteration
+  }
+
+  // ---- store dK, dV (strided, bf16) ---------------------------------------
+  bf16* dkp = dk + b * dk_sb + h * dk_sh;
+  bf16* dvp = dv + b * dv_sb + h * dv_sh;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int key = kv0 + 16 * wave + hi * 4 + r;
+    if (key >= Lk) continue;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      dkp[(int64_t)key * dk_sl + 16 * dt + lo] = f2bf(acc_dk[dt][r]);
+      dvp[(int64_t)key * dv_sl + 16 * dt + lo] = f2bf(acc_dv[dt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dQ kernel: workgroup owns q rows [q0, q0+64); loops kv tiles
+// ---------------------------------------------------------------------------
+
+template <bool CAUSAL>
+__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
+    bf16* __restrict__ dq, int Lq, int Lk, float scale, int H,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
+    int64_t dq_sb, int64_t dq_sh, int64_t dq_sl) {
+  // LDS: K^T [64][72] re-staged per kv tile; per-wave dS tile 4x[16][72].
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* kt_lds = reinterpret_cast<short*>(smem);
+  short* ds_lds = kt_lds + D * PITCH;
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lo = lane & 15, hi = lane >> 4;
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * BLK + wave * 16;  // this wave's first q row
+
+  const bf16* qp = q + b * q_sb + h * q_sh;
+  const bf16* kp = k + b * k_sb + h * k_sh;
+  const bf16* vp = v + b * v_sb + h * v_sh;
+  const bf16* dop = dO + b * do_sb + h * do_sh;
+  const float* lsep = lse + bh * Lq;
+  const float* dvp_row = Dv + bh * Lq;
+
+  // A-fragments of Q and dO for this wave's 16 q rows (fixed)
+  bf16x8_t qa[2], doa[2];
+  {
+    const int qrow = min(q0 + lo, Lq - 1);
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      qa[s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
+      doa[s] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qrow * do_sl + 32 * s + hi * 8);
+    }
+  }
+  // per-row lse and D for rows q = q0 + hi*4 + r
+  float lse_r[4], d_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qi = min(q0 + hi * 4 + r, Lq - 1);
+    lse_r[r] = lsep[qi];
+    d_r[r] = dvp_row[qi];
+  }
+
+  short* my_ds = ds_lds + wave * 16 * PITCH;
+  f32x4_t acc_dq[4] = {};  // rows q = hi*4+r, cols d = 16*dt + lo
+
+  const int kv_end = CAUSAL ? min(Lk, blockIdx.x * BLK + BLK) : Lk;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += BLK) {
+    // ---- stage K^T for this kv tile (scatter-transpose) ------------------
+    {
+      const int row = tid / 4;          // key within tile
+      const int c0 = (tid % 4) * 16;
+      const int key = kv0 + row;
+      if (key < Lk) {
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) {
+          const bf16x8_t kv8 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0 + hh * 8);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kt_lds[(c0 + hh * 8 + i) * PITCH + row] = kv8[i];
+        }
+      } else {
+        for (int i = 0; i < 16; ++i) kt_lds[(c0 + i) * PITCH + row] = 0;
+      }
+    }
+
+    // ---- B-fragments of K^T and V^T from global --------------------------
+    // B[k=d][j=key] element = K[kv0 + 16*kt + lo][32*s + hi*8 + jj]
+    bf16x8_t kb[4][2], vb[4][2];
+#pragma unroll
+    for (int kt = 0; kt < 4; ++kt) {
+      const int key = min(kv0 + 16 * kt + lo, Lk - 1);  // clamped; masked below
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + 32 * s + hi * 8);
+        vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + 32 * s + hi * 8);
+      }
+    }
+    __syncthreads();
+
+    // ---- S, P, dP, dS per 16-key tile ------------------------------------
+    // C rows = q = hi*4+r, cols = key = 16*kt + lo
+#pragma unroll
+    for (int kt = 0; kt < 4; ++kt) {
+      f32x4_t sc = {};
+      f32x4_t dpc = {};
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        sc = MFMA16(qa[s], kb[kt][s], sc);
+        dpc = MFMA16(doa[s], vb[kt][s], dpc);
+      }
+      const int key = kv0 + 16 * kt + lo;
+      bf16x4 dsk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qi = q0 + hi * 4 + r;
+        float p = __expf(sc[r] * scale - lse_r[r]);
+        if (key >= Lk || (CAUSAL && key > qi)) p = 0.f;
+        dsk[r] = f2bfs(p * (dpc[r] - d_r[r]) * scale);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) my_ds[(hi * 4 + r) * PITCH + 16 * kt + lo] = dsk[r];
+    }
+
+    // ---- dQ += dS . K (A = dS via LDS, B = K from K^T rows) --------------
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_ds + lo * PITCH + 32 * s + hi * 8);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const bf16x8_t bfrag =
+            *reinterpret_cast<const bf16x8_t*>(kt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
+        acc_dq[dt] = MFMA16(dsa, bfrag, acc_dq[dt]);
+      }
+    }
+    __syncthreads();  // kt_lds re-staged next iteration
+  }
+
+  // ---- store dQ (strided, bf16) -------------------------------------------
+  bf16* dqp = dq + b * dq_sb + h * dq_sh;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qi = q0 + hi * 4 + r;
+    if (qi >= Lq) continue;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      dqp[(int64_t)qi * dq_sl + 16 * dt + lo] = f2bf(acc_dq[dt][r]);
+  }
+}
+
+}  // namespace
+
+void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
+                    torch::Tensor dO, torch::Tensor lse, torch::Tensor dq, torch::Tensor dk,
+                    torch::Tensor dv, bool causal, double scale) {
+  // all tensors (B,H,L,64) views, innermost contiguous, arbitrary b/h/l strides
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "attn_bwd_fused: (B,H,L,64) expected");
+  const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
+  for (auto* t : {&q, &k, &v, &o, &dO, &dq, &dk, &dv})
+    TORCH_CHECK(t->stride(3) == 1, "attn_bwd_fused: innermost dim must be contiguous");
+  TORCH_CHECK(lse.is_contiguous() && lse.scalar_type() == torch::kFloat32);
+  auto stream = at::hip::getCurrentHIPStream();
+
+  // D = rowsum(dO * O)
+  auto Dv = torch::empty({(int64_t)B * H * Lq}, q.options().dtype(torch::kFloat32));
+  {
+    const int64_t nrows = (int64_t)B * H * Lq;
+    const dim3 grid((unsigned)std::min<int64_t>((nrows + 3) / 4, 4096));
+    hipLaunchKernelGGL(attn_d2_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const bf16*>(dO.data_ptr()),
+                       reinterpret_cast<const bf16*>(o.data_ptr()), Dv.data_ptr<float>(), H, Lq,
+                       nrows, dO.stride(0), dO.stride(1), dO.stride(2),
+                       o.stride(0), o.stride(1), o.stride(2));
+  }
+
+  const size_t shmem_dkv = (4 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const size_t shmem_dq = (D * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const dim3 grid_dkv((Lk + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
+  const dim3 grid_dq((Lq + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
+
+#define DKV_LAUNCH(C)                                                                        \
+  hipLaunchKernelGGL((attn_bwd_dkv_kernel<C>), grid_dkv, dim3(256), shmem_dkv, stream,       \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
+                     Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dk.data_ptr()),           \
+                     reinterpret_cast<bf16*>(dv.data_ptr()), Lq, Lk, (float)scale, H,        \
+                     q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),        \
+                     k.stride(2), v.stride(0), v.stride(1), v.stride(2), dO.stride(0),       \
+                     dO.stride(1), dO.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),   \
+                     dv.stride(0), dv.stride(1), dv.stride(2))
+#define DQ_LAUNCH(C)                                                                         \
+  hipLaunchKernelGGL((attn_bwd_dq_kernel<C>), grid_dq, dim3(256), shmem_dq, stream,          \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
+                     Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dq.data_ptr()),           \
+                     Lq, Lk, (float)scale, H, q.stride(0), q.stride(1), q.stride(2),         \
+                     k.stride(0), k.stride(1), k.stride(2), v.stride(0), v.stride(1),        \
+                     v.stride(2), dO.stride(0), dO.stride(1), dO.stride(2),                  \
+                     dq.stride(0), dq.stride(1), dq.stride(2))
+  if (causal) {
+    DKV_LAUNCH(true);
+    DQ_LAUNCH(true);
+  } else {
+    DKV_LAUNCH(false);
+    DQ_LAUNCH(false);
+  }
+#undef DKV_LAUNCH
+#undef DQ_LAUNCH
+}

@@ -8,6 +8,7 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
 torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, std::string act,
                            c10::optional<torch::Tensor> residual);
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act);
+torch::Tensor colsum(torch::Tensor dz);
 torch::Tensor im2col_patch(torch::Tensor img, int64_t patch);
 torch::Tensor col2im_patch(torch::Tensor cols, std::vector<int64_t> img_shape, int64_t patch);
 void adam_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
@@ -19,6 +20,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
 void attn_bwd_p(torch::Tensor s, torch::Tensor lse, bool causal, double scale);
 torch::Tensor attn_d(torch::Tensor dO, torch::Tensor O);
 void attn_ds(torch::Tensor dp, torch::Tensor p, torch::Tensor D, double scale);
+void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
+                    torch::Tensor dO, torch::Tensor lse, torch::Tensor dq, torch::Tensor dk,
+                    torch::Tensor dv, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
@@ -30,6 +34,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (K3/K15)");
   m.def("bias_act_fwd", &bias_act_fwd, "fused bias+activation(+residual) forward");
   m.def("act_bwd", &act_bwd, "activation backward: dy * act'(z)");
+  m.def("colsum", &colsum, "column sum -> fp32 (bias gradient, K15)");
   m.def("im2col_patch", &im2col_patch, "patch-embed unfold (K1)");
   m.def("col2im_patch", &col2im_patch, "patch-embed fold backward (K1/K15)");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam (K14)");
@@ -37,6 +42,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_p", &attn_bwd_p, "attention bwd: S -> P in place (K15)");
   m.def("attn_d", &attn_d, "attention bwd: rowsum(dO*O) (K15)");
   m.def("attn_ds", &attn_ds, "attention bwd: dP -> dS in place (K15)");
+  m.def("attn_bwd_fused", &attn_bwd_fused, "fused flash attention backward (K15)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
   m.def("linear_fwd", &linear_fwd, "MFMA GEMM + fused epilogue (K4/K6/K7/K8)");

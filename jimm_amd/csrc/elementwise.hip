@@ -80,6 +80,30 @@ __global__ void act_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ z
     dz[i] = (T)((float)dy[i] * act_grad((float)z[i], act));
 }
 
+// db[col] = sum_rows dz[row][col] — bias gradient (K15). Thread owns 8
+// consecutive columns (b128 loads), loops a row chunk with local fp32
+// accumulation, one atomicAdd per column at the end. Replaces torch's
+// bf16 reduce (2.5 ms/step in profiles/r01_NOTES.md).
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ dz, float* __restrict__ db, int64_t M,
+                              int N, int rows_per_wg) {
+  constexpr int V = 8;
+  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * V;
+  if (c0 >= N) return;
+  const int64_t r0 = (int64_t)blockIdx.y * rows_per_wg;
+  int64_t r1 = r0 + rows_per_wg;
+  if (r1 > M) r1 = M;
+  float acc[V] = {};
+  for (int64_t r = r0; r < r1; ++r) {
+    float v[V];
+    vload_f32<V>(dz + r * N + c0, v);
+#pragma unroll
+    for (int j = 0; j < V; ++j) acc[j] += v[j];
+  }
+#pragma unroll
+  for (int j = 0; j < V; ++j) atomicAdd(db + c0 + j, acc[j]);
+}
+
 // cols[b*h*w + ph*w + pw][c*P*P + i*P + j] = img[b][c][ph*P+i][pw*P+j]
 template <typename T>
 __global__ void im2col_kernel(const T* __restrict__ img, T* __restrict__ cols, int B,
@@ -212,6 +236,30 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act) {
     TORCH_CHECK(false, "act_bwd: unsupported dtype");
   }
   return dz;
+}
+
+torch::Tensor colsum(torch::Tensor dz) {
+  TORCH_CHECK(dz.is_cuda() && dz.is_contiguous() && dz.dim() == 2);
+  const int64_t M = dz.size(0);
+  const int N = dz.size(1);
+  TORCH_CHECK(N % 8 == 0, "colsum: N % 8 != 0");
+  auto db = torch::zeros({N}, dz.options().dtype(torch::kFloat32));
+  const int block = 256;
+  const int gx = (N / 8 + block - 1) / block;
+  const int rows_per_wg = 256;
+  const int gy = (int)std::min<int64_t>((M + rows_per_wg - 1) / rows_per_wg, 65535);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (dz.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((colsum_kernel<bf16>), dim3(gx, gy), dim3(block), 0, stream,
+                       reinterpret_cast<const bf16*>(dz.data_ptr()), db.data_ptr<float>(),
+                       M, N, rows_per_wg);
+  } else if (dz.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((colsum_kernel<float>), dim3(gx, gy), dim3(block), 0, stream,
+                       dz.data_ptr<float>(), db.data_ptr<float>(), M, N, rows_per_wg);
+  } else {
+    TORCH_CHECK(false, "colsum: unsupported dtype");
+  }
+  return db;
 }
 
 torch::Tensor im2col_patch(torch::Tensor img, int64_t patch) {

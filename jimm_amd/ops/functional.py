@@ -83,14 +83,40 @@ def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor, eps: f
 # ---------------------------------------------------------------------------
 
 
+def _strided_ok(t: torch.Tensor) -> bool:
+    return t.stride(3) == 1
+
+
+def _attn_bwd_composite(ext, q, k, v, o, do, lse, causal, scale):
+    """Recompute-P composite via rocBLAS batched GEMMs (fallback path,
+    JIMM_AMD_ATTN_BWD=composite); materializes (B,H,Lq,Lk)."""
+    q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+    o, do = o.contiguous(), do.contiguous()
+    s = torch.matmul(q, k.transpose(-1, -2))  # (B,H,Lq,Lk) bf16
+    ext.attn_bwd_p(s, lse, causal, scale)     # in place: s -> P (bf16)
+    dcoef = ext.attn_d(do, o)                 # (rows,) fp32 rowsum(dO*O)
+    dp = torch.matmul(do, v.transpose(-1, -2))
+    ext.attn_ds(dp, s, dcoef, scale)          # in place: dp -> dS (bf16)
+    dq = torch.matmul(dp, k)
+    dk = torch.matmul(dp.transpose(-1, -2), q)
+    dv = torch.matmul(s.transpose(-1, -2), do)
+    return dq, dk, dv
+
+
+def _use_fused_bwd() -> bool:
+    import os
+
+    return os.environ.get("JIMM_AMD_ATTN_BWD", "fused") == "fused"
+
+
 class _AttentionFn(torch.autograd.Function):
     """Forward: fused flash HIP kernel (csrc/attention.hip), saves lse.
 
-    Backward: recompute-P composite — rocBLAS batched GEMMs (plain library
-    GEMMs) + elementwise, an MI355X-first choice: at the in-scope L (50-1024,
-    SURVEY §5 long-context note) the S/P tiles are L2/LLC-resident and HBM
-    capacity (288 GB) makes the materialized (B,H,Lq,Lk) workspace free,
-    so a fused bwd kernel buys little; revisit with rocprof evidence.
+    Backward: fused flash backward (csrc/attention_bwd_fused.hip) — two MFMA
+    kernels (dK/dV over kv tiles, dQ over q tiles) recomputing P from lse,
+    reading/writing strided views with no permute copies. The round-1 profile
+    measured the old recompute-P rocBLAS composite at ~15 ms/step for ViT-B
+    (profiles/r01_NOTES.md), ~4 ms of it pure `.contiguous()` copies.
     """
 
     @staticmethod
@@ -106,18 +132,12 @@ class _AttentionFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        o, do = o.contiguous(), do.contiguous()
-        scale = ctx.scale
         ext = _backend.ext()
-        s = torch.matmul(q, k.transpose(-1, -2))  # (B,H,Lq,Lk) bf16
-        ext.attn_bwd_p(s, lse, ctx.causal, scale)  # in place: s -> P (bf16)
-        dcoef = ext.attn_d(do, o)                  # (rows,) fp32 rowsum(dO*O)
-        dp = torch.matmul(do, v.transpose(-1, -2))
-        ext.attn_ds(dp, s, dcoef, scale)           # in place: dp -> dS (bf16)
-        dq = torch.matmul(dp, k)
-        dk = torch.matmul(dp.transpose(-1, -2), q)
-        dv = torch.matmul(s.transpose(-1, -2), do)
+        if _use_fused_bwd() and all(_strided_ok(t) for t in (q, k, v, o, do)):
+            dq, dk, dv = torch.empty_like(q), torch.empty_like(k), torch.empty_like(v)
+            ext.attn_bwd_fused(q, k, v, o, do, lse, dq, dk, dv, ctx.causal, ctx.scale)
+        else:
+            dq, dk, dv = _attn_bwd_composite(ext, q, k, v, o, do, lse, ctx.causal, ctx.scale)
         return dq, dk, dv, None, None
 
 
@@ -147,19 +167,23 @@ class _AttentionQKVFn(torch.autograd.Function):
         qkv, o, lse = ctx.saved_tensors
         scale = ctx.scale
         ext = _backend.ext()
-        q = qkv[:, :, 0].transpose(1, 2).contiguous()
-        k = qkv[:, :, 1].transpose(1, 2).contiguous()
-        v = qkv[:, :, 2].transpose(1, 2).contiguous()
-        o, do = o.contiguous(), do.contiguous()
-        s = torch.matmul(q, k.transpose(-1, -2))
-        ext.attn_bwd_p(s, lse, ctx.causal, scale)
-        dcoef = ext.attn_d(do, o)
-        dp = torch.matmul(do, v.transpose(-1, -2))
-        ext.attn_ds(dp, s, dcoef, scale)
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
         dqkv = torch.empty_like(qkv)
-        dqkv[:, :, 0].transpose(1, 2).copy_(torch.matmul(dp, k))
-        dqkv[:, :, 1].transpose(1, 2).copy_(torch.matmul(dp.transpose(-1, -2), q))
-        dqkv[:, :, 2].transpose(1, 2).copy_(torch.matmul(s.transpose(-1, -2), do))
+        if _use_fused_bwd() and all(_strided_ok(t) for t in (o, do)):
+            # fused kernels write dq/dk/dv straight into the strided dqkv
+            # buffer — zero assembly copies
+            ext.attn_bwd_fused(
+                q, k, v, o, do, lse,
+                dqkv[:, :, 0].transpose(1, 2), dqkv[:, :, 1].transpose(1, 2),
+                dqkv[:, :, 2].transpose(1, 2), ctx.causal, scale,
+            )
+        else:
+            dq, dk, dv = _attn_bwd_composite(ext, q, k, v, o, do, lse, ctx.causal, scale)
+            dqkv[:, :, 0].transpose(1, 2).copy_(dq)
+            dqkv[:, :, 1].transpose(1, 2).copy_(dk)
+            dqkv[:, :, 2].transpose(1, 2).copy_(dv)
         return dqkv, None, None
 
 

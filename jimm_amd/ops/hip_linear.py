@@ -68,7 +68,13 @@ class _LinearActFn(torch.autograd.Function):
             dz = dy2
         dx = torch.matmul(dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
         dw = torch.matmul(dz.t(), x2) if ctx.needs_input_grad[1] else None
-        db = dz.sum(dim=0) if (ctx.has_bias and ctx.needs_input_grad[2]) else None
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            if dz.shape[-1] % 8 == 0:
+                db = _backend.ext().colsum(dz).to(dz.dtype)
+            else:
+                db = dz.sum(dim=0)
+        else:
+            db = None
         dres = dy.view(ctx.res_shape) if (ctx.has_res and ctx.needs_input_grad[4]) else None
         return dx, dw, db, None, dres
 
@@ -110,7 +116,13 @@ class _PatchEmbedFn(torch.autograd.Function):
         dcols = torch.matmul(dy2, w2)
         dimg = _backend.ext().col2im_patch(dcols, list(ctx.img_shape), patch) if ctx.needs_input_grad[0] else None
         dw = torch.matmul(dy2.t(), cols).view(ctx.w_shape) if ctx.needs_input_grad[1] else None
-        db = dy2.sum(dim=0) if (ctx.has_bias and ctx.needs_input_grad[2]) else None
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            if dy2.shape[-1] % 8 == 0:
+                db = _backend.ext().colsum(dy2).to(dy2.dtype)
+            else:
+                db = dy2.sum(dim=0)
+        else:
+            db = None
         return dimg, dw, db, None
 
 

@@ -28,6 +28,7 @@ setup(
                 "jimm_amd/csrc/adam.hip",
                 "jimm_amd/csrc/attention.hip",
                 "jimm_amd/csrc/attention_bwd.hip",
+                "jimm_amd/csrc/attention_bwd_fused.hip",
                 "jimm_amd/csrc/gemm.hip",
                 "jimm_amd/csrc/probe.hip",
             ],
