@@ -72,8 +72,11 @@ def main() -> None:
     p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
     args = p.parse_args()
 
+    from jimm_amd.ops._backend import maybe_enable_tunableop
     from jimm_amd.train import SyntheticImages, SyntheticImageText
     from jimm_amd.train.trainer import init_distributed
+
+    maybe_enable_tunableop()  # JIMM_AMD_TUNABLE=<csv>: committed hipBLASLt algo table
 
     rank, world, local_rank, device = init_distributed()
     on_gpu = device.type == "cuda"

@@ -1,0 +1,82 @@
+"""Microbench of the training-step GEMM family (fwd NT, dX NN, dW TN) via
+rocBLAS/hipBLASLt, optionally under PyTorch TunableOp (hipBLASLt algo search,
+including split-K — the round-1 profile showed Tensile picking a 27-workgroup
+kernel for the dW shapes, ~0.4 PF/s on a 256-CU chip).
+
+Usage (on an MI355X via gpurun):
+    python benchmarks/bwd_gemm_bench.py             # baseline
+    python benchmarks/bwd_gemm_bench.py --tune out.csv   # tune + save results
+    python benchmarks/bwd_gemm_bench.py --load out.csv   # use saved tunings
+"""
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+import torch
+
+SHAPES = [
+    # (M, N, K) of the forward x(M,K) @ w(N,K)^T — ViT-B/16 @ bs256
+    (50432, 2304, 768),   # QKV
+    (50432, 768, 768),    # attn proj
+    (50432, 3072, 768),   # fc1
+    (50432, 768, 3072),   # fc2
+    (50176, 768, 768),    # patch embed
+]
+
+
+def bench(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--tune", metavar="CSV", default=None)
+    p.add_argument("--load", metavar="CSV", default=None)
+    args = p.parse_args()
+
+    if args.tune or args.load:
+        tun = torch.cuda.tunable
+        tun.enable(True)
+        if args.tune:
+            tun.tuning_enable(True)
+            tun.set_max_tuning_duration(50)
+        else:
+            tun.tuning_enable(False)
+            assert tun.read_file(args.load), f"failed to read {args.load}"
+
+    dev = torch.device("cuda:0")
+    for M, N, K in SHAPES:
+        x = torch.randn(M, K, device=dev).bfloat16()
+        w = torch.randn(N, K, device=dev).bfloat16() / K**0.5
+        dz = torch.randn(M, N, device=dev).bfloat16()
+        tf = 2 * M * N * K / 1e12
+
+        t_fwd = bench(lambda: torch.matmul(x, w.t()))      # NT fwd
+        t_dx = bench(lambda: torch.matmul(dz, w))          # NN dX
+        t_dw = bench(lambda: torch.matmul(dz.t(), x))      # TN dW
+        print(
+            f"M={M:6d} N={N:5d} K={K:5d}  "
+            f"fwd {tf/t_fwd:7.1f} TF/s ({t_fwd*1e3:6.2f} ms)  "
+            f"dX {tf/t_dx:7.1f} TF/s ({t_dx*1e3:6.2f} ms)  "
+            f"dW {tf/t_dw:7.1f} TF/s ({t_dw*1e3:6.2f} ms)",
+            flush=True,
+        )
+
+    if args.tune:
+        torch.cuda.tunable.write_file(args.tune)
+        print("wrote", args.tune)
+
+
+if __name__ == "__main__":
+    main()

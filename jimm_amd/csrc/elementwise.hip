@@ -88,13 +88,28 @@ template <typename T>
 __global__ void colsum_kernel(const T* __restrict__ dz, float* __restrict__ db, int64_t M,
                               int N, int rows_per_wg) {
   constexpr int V = 8;
-  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * V;
-  if (c0 >= N) return;
+  const int strips = N / V;                       // column strips of 8
+  const int strip = (blockIdx.x * blockDim.x + threadIdx.x) % strips;
+  const int rlane = (blockIdx.x * blockDim.x + threadIdx.x) / strips;
+  const int rstep = (gridDim.x * blockDim.x) / strips;  // row-parallel threads per strip
+  if (rlane >= rstep) return;  // leftover threads would double-count rows
+  const int c0 = strip * V;
   const int64_t r0 = (int64_t)blockIdx.y * rows_per_wg;
   int64_t r1 = r0 + rows_per_wg;
   if (r1 > M) r1 = M;
   float acc[V] = {};
-  for (int64_t r = r0; r < r1; ++r) {
+  // 4-deep unrolled row loop keeps >=4 b128 loads in flight per thread
+  int64_t r = r0 + rlane;
+  for (; r + 3 * rstep < r1; r += 4 * rstep) {
+    float v0[V], v1[V], v2[V], v3[V];
+    vload_f32<V>(dz + r * N + c0, v0);
+    vload_f32<V>(dz + (r + rstep) * N + c0, v1);
+    vload_f32<V>(dz + (r + 2 * rstep) * N + c0, v2);
+    vload_f32<V>(dz + (r + 3 * rstep) * N + c0, v3);
+#pragma unroll
+    for (int j = 0; j < V; ++j) acc[j] += (v0[j] + v1[j]) + (v2[j] + v3[j]);
+  }
+  for (; r < r1; r += rstep) {
     float v[V];
     vload_f32<V>(dz + r * N + c0, v);
 #pragma unroll
@@ -245,8 +260,11 @@ torch::Tensor colsum(torch::Tensor dz) {
   TORCH_CHECK(N % 8 == 0, "colsum: N % 8 != 0");
   auto db = torch::zeros({N}, dz.options().dtype(torch::kFloat32));
   const int block = 256;
-  const int gx = (N / 8 + block - 1) / block;
-  const int rows_per_wg = 256;
+  // grid sized so ~1024 workgroups cover the matrix: each thread owns one
+  // 8-wide column strip within a row chunk; surplus threads parallelize rows
+  const int strips = N / 8;
+  const int gx = std::max(1, (strips + block - 1) / block);
+  const int rows_per_wg = 64;
   const int gy = (int)std::min<int64_t>((M + rows_per_wg - 1) / rows_per_wg, 65535);
   auto stream = at::hip::getCurrentHIPStream();
   if (dz.scalar_type() == torch::kBFloat16) {

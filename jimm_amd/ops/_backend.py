@@ -16,6 +16,25 @@ and A/B numerics only; never the default).
 
 from __future__ import annotations
 
+
+def maybe_enable_tunableop() -> bool:
+    """Load a committed TunableOp result table (hipBLASLt algo selections)
+    when JIMM_AMD_TUNABLE points at a CSV produced by
+    benchmarks/bwd_gemm_bench.py --tune. Returns True if enabled."""
+    import os
+
+    path = os.environ.get("JIMM_AMD_TUNABLE", "")
+    if not path:
+        return False
+    import torch
+
+    tun = torch.cuda.tunable
+    tun.enable(True)
+    tun.tuning_enable(False)
+    if os.path.exists(path):
+        tun.read_file(path)
+    return True
+
 import importlib
 import os
 
