@@ -25,6 +25,8 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
                     torch::Tensor dv, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype);
+bool gemm_dw_supported(int64_t M, int64_t N, int64_t K);
+torch::Tensor gemm_tn_splitk(torch::Tensor dz, torch::Tensor x);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias, std::string act,
                                       c10::optional<torch::Tensor> residual, bool save_z);
@@ -45,5 +47,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_fused", &attn_bwd_fused, "fused flash attention backward (K15)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
+  m.def("gemm_dw_supported", &gemm_dw_supported, "split-K dW GEMM shape check");
+  m.def("gemm_tn_splitk", &gemm_tn_splitk, "split-K TN GEMM for weight grads (K15)");
   m.def("linear_fwd", &linear_fwd, "MFMA GEMM + fused epilogue (K4/K6/K7/K8)");
 }

@@ -250,6 +250,12 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 
 }  // namespace
 
+// 256x256-tile kernel (gemm256.hip) — used when the shape tiles evenly
+bool gemm256_supported(int64_t M, int64_t N, int64_t K);
+void gemm_nt_256(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias_f32,
+                 std::string act, c10::optional<torch::Tensor> residual, torch::Tensor y,
+                 c10::optional<torch::Tensor> z);
+
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype) {
   if (dtype != "torch.bfloat16") return false;
   if (K % BK != 0) return false;
@@ -275,6 +281,14 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
   else if (act == "gelu_tanh") act_code = ACT_GELU_TANH;
   else if (act == "quickgelu") act_code = ACT_QUICKGELU;
   else TORCH_CHECK(act.empty(), "unknown act ", act);
+
+  if (gemm256_supported(M, N, K)) {
+    c10::optional<torch::Tensor> zopt;
+    if (save_z) zopt = z;
+    gemm_nt_256(x, w, bf, act, residual, y, zopt);
+    if (save_z) return {y, z};
+    return {y, torch::Tensor()};
+  }
 
   auto stream = at::hip::getCurrentHIPStream();
   const bf16* resp = residual ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr;

@@ -66,8 +66,18 @@ class _LinearActFn(torch.autograd.Function):
             dz = _backend.ext().act_bwd(dy2, z, ctx.act)
         else:
             dz = dy2
+        ext = _backend.ext()
         dx = torch.matmul(dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
-        dw = torch.matmul(dz.t(), x2) if ctx.needs_input_grad[1] else None
+        if ctx.needs_input_grad[1]:
+            # dW = dz^T @ x: rocBLAS picks non-split kernels for these huge-K
+            # skinny-output shapes (0.2-0.6 PF/s); the in-house split-K TN
+            # kernel (csrc/gemm_dw.hip) covers the model-zoo shapes
+            if dz.dtype == torch.bfloat16 and ext.gemm_dw_supported(dz.shape[0], dz.shape[1], x2.shape[1]):
+                dw = ext.gemm_tn_splitk(dz, x2)
+            else:
+                dw = torch.matmul(dz.t(), x2)
+        else:
+            dw = None
         if ctx.has_bias and ctx.needs_input_grad[2]:
             if dz.shape[-1] % 8 == 0:
                 db = _backend.ext().colsum(dz).to(dz.dtype)

@@ -30,6 +30,8 @@ setup(
                 "jimm_amd/csrc/attention_bwd.hip",
                 "jimm_amd/csrc/attention_bwd_fused.hip",
                 "jimm_amd/csrc/gemm.hip",
+                "jimm_amd/csrc/gemm256.hip",
+                "jimm_amd/csrc/gemm_dw.hip",
                 "jimm_amd/csrc/probe.hip",
             ],
             extra_compile_args={
