@@ -47,11 +47,13 @@ def main():
 
     if args.tune or args.load:
         tun = torch.cuda.tunable
-        tun.enable(True)
         if args.tune:
+            tun.set_filename(args.tune)  # auto-written at process exit
+            tun.enable(True)
             tun.tuning_enable(True)
             tun.set_max_tuning_duration(50)
         else:
+            tun.enable(True)
             tun.tuning_enable(False)
             assert tun.read_file(args.load), f"failed to read {args.load}"
 
@@ -74,8 +76,7 @@ def main():
         )
 
     if args.tune:
-        torch.cuda.tunable.write_file(args.tune)
-        print("wrote", args.tune)
+        print("tuning results will be written at exit to", args.tune)
 
 
 if __name__ == "__main__":

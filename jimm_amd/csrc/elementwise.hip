@@ -81,11 +81,12 @@ __global__ void act_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ z
 }
 
 // db[col] = sum_rows dz[row][col] — bias gradient (K15). Thread owns 8
-// consecutive columns (b128 loads), loops a row chunk with local fp32
-// accumulation, one atomicAdd per column at the end. Replaces torch's
-// bf16 reduce (2.5 ms/step in profiles/r01_NOTES.md).
+// consecutive columns (b128 loads), 4-deep unrolled row loop keeps loads in
+// flight; partial sums land in a (gy*rstep, N) fp32 workspace (NO global
+// atomics — v1/v2 with atomics measured 133/262 us for (50k,768); the
+// per-address serialization dominated). Host reduces the workspace.
 template <typename T>
-__global__ void colsum_kernel(const T* __restrict__ dz, float* __restrict__ db, int64_t M,
+__global__ void colsum_kernel(const T* __restrict__ dz, float* __restrict__ ws, int64_t M,
                               int N, int rows_per_wg) {
   constexpr int V = 8;
   const int strips = N / V;                       // column strips of 8
@@ -98,7 +99,6 @@ __global__ void colsum_kernel(const T* __restrict__ dz, float* __restrict__ db, 
   int64_t r1 = r0 + rows_per_wg;
   if (r1 > M) r1 = M;
   float acc[V] = {};
-  // 4-deep unrolled row loop keeps >=4 b128 loads in flight per thread
   int64_t r = r0 + rlane;
   for (; r + 3 * rstep < r1; r += 4 * rstep) {
     float v0[V], v1[V], v2[V], v3[V];
@@ -115,8 +115,9 @@ __global__ void colsum_kernel(const T* __restrict__ dz, float* __restrict__ db, 
 #pragma unroll
     for (int j = 0; j < V; ++j) acc[j] += v[j];
   }
+  float* out = ws + ((int64_t)blockIdx.y * rstep + rlane) * N + c0;
 #pragma unroll
-  for (int j = 0; j < V; ++j) atomicAdd(db + c0 + j, acc[j]);
+  for (int j = 0; j < V; ++j) out[j] = acc[j];
 }
 
 // cols[b*h*w + ph*w + pw][c*P*P + i*P + j] = img[b][c][ph*P+i][pw*P+j]
@@ -258,26 +259,27 @@ torch::Tensor colsum(torch::Tensor dz) {
   const int64_t M = dz.size(0);
   const int N = dz.size(1);
   TORCH_CHECK(N % 8 == 0, "colsum: N % 8 != 0");
-  auto db = torch::zeros({N}, dz.options().dtype(torch::kFloat32));
   const int block = 256;
-  // grid sized so ~1024 workgroups cover the matrix: each thread owns one
+  // grid sized so ~1000 workgroups cover the matrix: each thread owns one
   // 8-wide column strip within a row chunk; surplus threads parallelize rows
   const int strips = N / 8;
   const int gx = std::max(1, (strips + block - 1) / block);
+  const int rstep = gx * block / strips;
   const int rows_per_wg = 64;
   const int gy = (int)std::min<int64_t>((M + rows_per_wg - 1) / rows_per_wg, 65535);
+  auto ws = torch::empty({(int64_t)gy * rstep, N}, dz.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   if (dz.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL((colsum_kernel<bf16>), dim3(gx, gy), dim3(block), 0, stream,
-                       reinterpret_cast<const bf16*>(dz.data_ptr()), db.data_ptr<float>(),
+                       reinterpret_cast<const bf16*>(dz.data_ptr()), ws.data_ptr<float>(),
                        M, N, rows_per_wg);
   } else if (dz.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((colsum_kernel<float>), dim3(gx, gy), dim3(block), 0, stream,
-                       dz.data_ptr<float>(), db.data_ptr<float>(), M, N, rows_per_wg);
+                       dz.data_ptr<float>(), ws.data_ptr<float>(), M, N, rows_per_wg);
   } else {
     TORCH_CHECK(false, "colsum: unsupported dtype");
   }
-  return db;
+  return ws.sum(0);  // (gy*rstep, N) fp32 reduce — tiny
 }
 
 torch::Tensor im2col_patch(torch::Tensor img, int64_t patch) {

@@ -21,7 +21,12 @@ from jimm_amd.ops import _backend
 
 
 def _gemm_mode() -> str:
-    return os.environ.get("JIMM_AMD_GEMM", "hip")
+    # Default "blas": measured on MI355X (benchmarks/gemm_bench.py, r03): the
+    # in-house 256-tile MFMA GEMM reaches 542-868 TF/s vs rocBLAS 845-1089 on
+    # the model-zoo shapes, so plain GEMMs run through rocBLAS with the fused
+    # HIP bias+act(+residual) epilogue kernel. JIMM_AMD_GEMM=hip re-enables
+    # the in-house path (kernels stay built and tested).
+    return os.environ.get("JIMM_AMD_GEMM", "blas")
 
 
 def _gemm_nt(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
@@ -69,10 +74,15 @@ class _LinearActFn(torch.autograd.Function):
         ext = _backend.ext()
         dx = torch.matmul(dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
         if ctx.needs_input_grad[1]:
-            # dW = dz^T @ x: rocBLAS picks non-split kernels for these huge-K
-            # skinny-output shapes (0.2-0.6 PF/s); the in-house split-K TN
-            # kernel (csrc/gemm_dw.hip) covers the model-zoo shapes
-            if dz.dtype == torch.bfloat16 and ext.gemm_dw_supported(dz.shape[0], dz.shape[1], x2.shape[1]):
+            # dW = dz^T @ x. rocBLAS picks weak non-split kernels here
+            # (0.23-0.6 PF/s) but still beats the in-house split-K TN kernel
+            # (0.25-0.28 PF/s, r03 measurements) — rocBLAS by default, the
+            # in-house kernel opt-in via JIMM_AMD_DW=hip while it is tuned.
+            if (
+                os.environ.get("JIMM_AMD_DW", "blas") == "hip"
+                and dz.dtype == torch.bfloat16
+                and ext.gemm_dw_supported(dz.shape[0], dz.shape[1], x2.shape[1])
+            ):
                 dw = ext.gemm_tn_splitk(dz, x2)
             else:
                 dw = torch.matmul(dz.t(), x2)

@@ -39,8 +39,21 @@ class Trainer:
         self.opt = Adam(model.parameters(), lr=cfg.lr, betas=cfg.betas, weight_decay=cfg.weight_decay)
         self.group = process_group
         self.step_idx = 0
+        # JIMM_AMD_ROCTX=1: wrap each step in a roctx range (shows up in
+        # rocprofv3 --marker-trace; SURVEY §5 tracing)
+        import os
+
+        self._roctx = os.environ.get("JIMM_AMD_ROCTX", "0") == "1"
 
     def train_step(self, batch) -> dict[str, torch.Tensor]:
+        if self._roctx:
+            torch.cuda.nvtx.range_push(f"train_step_{self.cfg.task}")  # roctx range on ROCm
+        out = self._train_step_inner(batch)
+        if self._roctx:
+            torch.cuda.nvtx.range_pop()
+        return out
+
+    def _train_step_inner(self, batch) -> dict[str, torch.Tensor]:
         self.ddp.zero_grad()
         out: dict[str, torch.Tensor] = {}
         if self.cfg.task == "vit":
@@ -66,6 +79,36 @@ class Trainer:
         self.step_idx += 1
         out["loss"] = loss.detach()
         return out
+
+    # -- checkpoint / resume (the reference is load-only; we add
+    # full trainer state so long runs are resumable) ------------------------
+    def save_checkpoint(self, path: str) -> None:
+        """Rank-0 writes model + Adam state (moments, masters, step count)."""
+        import torch.distributed as dist
+
+        if dist.is_initialized() and dist.get_rank(self.group) != 0:
+            return
+        torch.save(
+            {
+                "model": self.model.state_dict(),
+                "optimizer": self.opt.state_dict(),
+                "step_idx": self.step_idx,
+                "task": self.cfg.task,
+            },
+            path,
+        )
+
+    def load_checkpoint(self, path: str) -> None:
+        ckpt = torch.load(path, map_location="cpu", weights_only=False)
+        self.model.load_state_dict(ckpt["model"])
+        self.opt.load_state_dict(ckpt["optimizer"])
+        # moments/masters must live on the parameters' device
+        dev = next(self.model.parameters()).device
+        for st in self.opt.state.values():
+            for k, v in st.items():
+                if torch.is_tensor(v):
+                    st[k] = v.to(dev)
+        self.step_idx = ckpt["step_idx"]
 
     @torch.no_grad()
     def eval_step(self, batch) -> dict[str, torch.Tensor]:
