@@ -25,6 +25,12 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
                     torch::Tensor dv, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype);
+std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x);
+torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor rinv);
+std::vector<torch::Tensor> xent_rows_fwd(torch::Tensor logits, torch::Tensor labels);
+torch::Tensor xent_rows_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+                            double gscale);
+std::vector<torch::Tensor> sigmoid_loss_ew(torch::Tensor logits, int64_t diag0);
 bool gemm_dw_supported(int64_t M, int64_t N, int64_t K);
 torch::Tensor gemm_tn_splitk(torch::Tensor dz, torch::Tensor x);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
@@ -48,6 +54,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
   m.def("gemm_dw_supported", &gemm_dw_supported, "split-K dW GEMM shape check");
+  m.def("l2norm_fwd", &l2norm_fwd, "row L2-normalize forward (K12)");
+  m.def("l2norm_bwd", &l2norm_bwd, "row L2-normalize backward (K12)");
+  m.def("xent_rows_fwd", &xent_rows_fwd, "fused softmax-CE forward (K13)");
+  m.def("xent_rows_bwd", &xent_rows_bwd, "fused softmax-CE backward (K13)");
+  m.def("sigmoid_loss_ew", &sigmoid_loss_ew, "SigLIP sigmoid loss + dLogits (K13)");
   m.def("gemm_tn_splitk", &gemm_tn_splitk, "split-K TN GEMM for weight grads (K15)");
   m.def("linear_fwd", &linear_fwd, "MFMA GEMM + fused epilogue (K4/K6/K7/K8)");
 }
