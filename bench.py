@@ -70,6 +70,8 @@ def main() -> None:
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--task", choices=["vit", "clip", "siglip"], default="vit")
     p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
+    p.add_argument("--graph", choices=["auto", "1", "0"], default="auto",
+                   help="hipGraph-capture the train step (auto: single-GPU only)")
     args = p.parse_args()
 
     from jimm_amd.ops._backend import maybe_enable_tunableop
@@ -105,6 +107,13 @@ def main() -> None:
             torch.cuda.synchronize()
 
     it = iter(data)
+    use_graph = on_gpu and (args.graph == "1" or (args.graph == "auto" and world == 1))
+    if use_graph:
+        try:
+            trainer.enable_graph(next(it))
+        except Exception as e:  # fall back to eager launches
+            if rank == 0:
+                print(f"# graph capture failed ({e!r}); running eager", flush=True)
     for _ in range(args.warmup):
         trainer.train_step(next(it))
     sync()

@@ -52,7 +52,84 @@ __global__ void adam_kernel(const ChunkDesc* __restrict__ chunks, int nchunks, f
   }
 }
 
+__global__ void adam_kernel_dev(const ChunkDesc* __restrict__ chunks, int nchunks,
+                                const float* __restrict__ lr_ptr,
+                                const int* __restrict__ step_ptr, float b1, float b2,
+                                float eps, float wd) {
+  const float lr = *lr_ptr;
+  const float st = (float)*step_ptr;
+  const float bc1 = 1.f - __powf(b1, st);
+  const float bc2 = 1.f - __powf(b2, st);
+  for (int ci = blockIdx.x; ci < nchunks; ci += gridDim.x) {
+    ChunkDesc c = chunks[ci];
+    for (int64_t i = threadIdx.x; i < c.n; i += blockDim.x) {
+      const int64_t idx = c.offset + i;
+      float g = c.is_bf16 ? bf2f(reinterpret_cast<const bf16*>(c.g)[idx])
+                          : reinterpret_cast<const float*>(c.g)[idx];
+      float m = c.m[idx] = b1 * c.m[idx] + (1.f - b1) * g;
+      float v = c.v[idx] = b2 * c.v[idx] + (1.f - b2) * g * g;
+      float upd = (m / bc1) / (sqrtf(v / bc2) + eps);
+      float w0 = c.master ? c.master[idx]
+                          : (c.is_bf16 ? bf2f(reinterpret_cast<bf16*>(c.p)[idx])
+                                       : reinterpret_cast<float*>(c.p)[idx]);
+      if (wd != 0.f) upd += wd * w0;
+      float w1 = w0 - lr * upd;
+      if (c.master) c.master[idx] = w1;
+      if (c.is_bf16)
+        reinterpret_cast<bf16*>(c.p)[idx] = f2bf(w1);
+      else
+        reinterpret_cast<float*>(c.p)[idx] = w1;
+    }
+  }
+}
+
+__global__ void incr_kernel(int* __restrict__ step) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *step += 1;
+}
+
 }  // namespace
+
+// Build the chunk-descriptor buffer ONCE (device uint8 tensor). Pointers are
+// stable across steps, so the graph-capturable adam_apply path needs no
+// per-step host work at all.
+std::vector<torch::Tensor> adam_prepare(std::vector<torch::Tensor> ps,
+                                        std::vector<torch::Tensor> gs,
+                                        std::vector<torch::Tensor> ms,
+                                        std::vector<torch::Tensor> vs,
+                                        std::vector<c10::optional<torch::Tensor>> masters) {
+  std::vector<ChunkDesc> chunks;
+  chunks.reserve(512);
+  for (size_t t = 0; t < ps.size(); ++t) {
+    auto& p = ps[t];
+    TORCH_CHECK(p.is_cuda() && p.is_contiguous());
+    const bool is_bf16 = p.scalar_type() == torch::kBFloat16;
+    float* master = masters[t].has_value() ? masters[t]->data_ptr<float>() : nullptr;
+    const int64_t n = p.numel();
+    for (int64_t off = 0; off < n; off += kChunkSize) {
+      chunks.push_back(ChunkDesc{gs[t].data_ptr(), p.data_ptr(), ms[t].data_ptr<float>(),
+                                 vs[t].data_ptr<float>(), master, off,
+                                 std::min<int64_t>(kChunkSize, n - off), is_bf16 ? 1 : 0});
+    }
+  }
+  auto host = torch::from_blob(chunks.data(), {(int64_t)(chunks.size() * sizeof(ChunkDesc))},
+                               torch::TensorOptions().dtype(torch::kUInt8));
+  auto dev = host.to(ps[0].device());
+  auto nchunks = torch::tensor({(int64_t)chunks.size()});
+  return {dev, nchunks};
+}
+
+// Graph-capturable update: lr and step live in device tensors; step is
+// incremented on-device first (so a replayed graph advances bias correction).
+void adam_apply(torch::Tensor desc, int64_t nchunks, torch::Tensor lr_dev,
+                torch::Tensor step_dev, double b1, double b2, double eps, double wd) {
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(incr_kernel, dim3(1), dim3(1), 0, stream, step_dev.data_ptr<int>());
+  const int grid = std::min<int>((int)nchunks, 2048);
+  hipLaunchKernelGGL(adam_kernel_dev, dim3(grid), dim3(256), 0, stream,
+                     reinterpret_cast<const ChunkDesc*>(desc.data_ptr()), (int)nchunks,
+                     lr_dev.data_ptr<float>(), step_dev.data_ptr<int>(), (float)b1,
+                     (float)b2, (float)eps, (float)wd);
+}
 
 void adam_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,

@@ -15,6 +15,13 @@ void adam_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                std::vector<c10::optional<torch::Tensor>> masters, double lr, double b1,
                double b2, double eps, double wd, int64_t step);
+std::vector<torch::Tensor> adam_prepare(std::vector<torch::Tensor> ps,
+                                        std::vector<torch::Tensor> gs,
+                                        std::vector<torch::Tensor> ms,
+                                        std::vector<torch::Tensor> vs,
+                                        std::vector<c10::optional<torch::Tensor>> masters);
+void adam_apply(torch::Tensor desc, int64_t nchunks, torch::Tensor lr_dev,
+                torch::Tensor step_dev, double b1, double b2, double eps, double wd);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     bool causal, double scale);
 void attn_bwd_p(torch::Tensor s, torch::Tensor lse, bool causal, double scale);
@@ -46,6 +53,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("im2col_patch", &im2col_patch, "patch-embed unfold (K1)");
   m.def("col2im_patch", &col2im_patch, "patch-embed fold backward (K1/K15)");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam (K14)");
+  m.def("adam_prepare", &adam_prepare, "build device chunk descriptors once (K14)");
+  m.def("adam_apply", &adam_apply, "graph-capturable fused Adam update (K14)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward, head_dim 64 (K5)");
   m.def("attn_bwd_p", &attn_bwd_p, "attention bwd: S -> P in place (K15)");
   m.def("attn_d", &attn_d, "attention bwd: rowsum(dO*O) (K15)");

@@ -32,6 +32,9 @@ class Adam(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
         self.master_weights = master_weights
+        # graph-capturable fast path: chunk descriptors uploaded once, lr and
+        # step live on-device (csrc/adam.hip adam_prepare/adam_apply)
+        self._prepared = None  # (desc_dev, nchunks, lr_dev, step_dev)
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -63,10 +66,47 @@ class Adam(torch.optim.Optimizer):
             if not ps:
                 continue
             if ps[0].is_cuda and not _backend.force_eager():
-                _backend.ext().adam_step(ps, gs, ms, vs, masters, lr, b1, b2, eps, wd, step_t)
+                ext = _backend.ext()
+                if self._prepared is None:
+                    if step_t == 1:
+                        # first step initializes state lazily; take it eagerly
+                        # then freeze the descriptor table (pointers stable)
+                        ext.adam_step(ps, gs, ms, vs, masters, lr, b1, b2, eps, wd, step_t)
+                    desc, nchunks = ext.adam_prepare(ps, gs, ms, vs, masters)
+                    lr_dev = torch.full((1,), lr, dtype=torch.float32, device=ps[0].device)
+                    # adam_apply pre-increments on device: seed so the next
+                    # apply computes bias correction for the right step
+                    seed = step_t if step_t == 1 else step_t - 1
+                    step_dev = torch.full((1,), seed, dtype=torch.int32, device=ps[0].device)
+                    self._prepared = (desc, int(nchunks.item()), lr_dev, step_dev)
+                    if step_t == 1:
+                        continue
+                desc, nchunks, lr_dev, step_dev = self._prepared
+                if lr != self._last_lr:
+                    lr_dev.fill_(lr)
+                    self._last_lr = lr
+                ext.adam_apply(desc, nchunks, lr_dev, step_dev, b1, b2, eps, wd)
             else:
                 self._ref_step(ps, gs, ms, vs, masters, lr, b1, b2, eps, wd, step_t)
         return loss
+
+    _last_lr: float = float("nan")
+
+    def load_state_dict(self, state_dict) -> None:
+        super().load_state_dict(state_dict)
+        self._prepared = None  # moment tensors were replaced: descriptors stale
+
+    @torch.no_grad()
+    def sync_step_from_device(self) -> None:
+        """Copy the on-device step counter back into per-param state (needed
+        after graph-replayed training, where host counters do not advance)."""
+        if self._prepared is None:
+            return
+        step = int(self._prepared[3].item())
+        for group in self.param_groups:
+            for p in group["params"]:
+                if self.state[p]:
+                    self.state[p]["step"] = step
 
     @staticmethod
     def _ref_step(ps, gs, ms, vs, masters, lr, b1, b2, eps, wd, t):
