@@ -422,8 +422,8 @@ def test_contrastive_losses_gpu_vs_cpu(task):
     bias = torch.tensor(-2.0)
 
     def run(device):
-        i = img.to(device).requires_grad_(True)
-        t = txt.to(device).requires_grad_(True)
+        i = img.detach().clone().to(device).requires_grad_(True)
+        t = txt.detach().clone().to(device).requires_grad_(True)
         if task == "clip":
             loss = L.clip_contrastive_loss(i, t, scale.to(device), gather=False)
         else:
