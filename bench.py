@@ -28,6 +28,13 @@ def build(task: str, device, dtype):
     if task == "vit":
         model = jimm_amd.VisionTransformer(num_classes=1000, img_size=224, patch_size=16)  # ViT-B/16
         model_name = "ViT-B/16@224"
+    elif task == "vitl384":
+        # BASELINE config 5: ViT-L/16 @ 384 with multihead-attention pooling
+        model = jimm_amd.VisionTransformer(
+            num_classes=1000, img_size=384, patch_size=16, num_layers=24,
+            hidden_size=1024, num_heads=16, mlp_dim=4096, pooling="MAP",
+        )
+        model_name = "ViT-L/16@384-MAP"
     elif task == "clip":
         model = jimm_amd.CLIP(  # CLIP ViT-B/32 (openai/clip-vit-base-patch32 dims)
             embed_dim=512,
@@ -59,7 +66,7 @@ def build(task: str, device, dtype):
         raise ValueError(task)
     model = model.to(device=device, dtype=dtype)
     # LayerNorm/scalars stay bf16-safe; master fp32 weights live in Adam state
-    trainer = Trainer(model, TrainConfig(task=task))
+    trainer = Trainer(model, TrainConfig(task="vit" if task.startswith("vit") else task))
     return model, trainer, model_name
 
 
@@ -68,7 +75,9 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--task", choices=["vit", "clip", "siglip"], default="vit")
+    p.add_argument("--task", choices=["vit", "vitl384", "clip", "siglip"], default="vit")
+    p.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16",
+                   help="fp8: e4m3 forward GEMMs via MFMA fp8 (bf16 backward)")
     p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
     p.add_argument("--graph", choices=["auto", "1", "0"], default="auto",
                    help="hipGraph-capture the train step (auto: single-GPU only)")
@@ -83,13 +92,18 @@ def main() -> None:
     rank, world, local_rank, device = init_distributed()
     on_gpu = device.type == "cuda"
     dtype = torch.bfloat16 if on_gpu else torch.float32
-    batch = args.batch or ({"vit": 256, "clip": 256, "siglip": 256}[args.task] if on_gpu else 4)
+    batch = args.batch or ({"vit": 256, "vitl384": 64, "clip": 256, "siglip": 256}[args.task] if on_gpu else 4)
 
     torch.manual_seed(1234 + rank)
     model, trainer, model_name = build(args.task, device, dtype)
 
-    if args.task == "vit":
-        data = SyntheticImages(batch, 224, 1000, device, dtype=dtype, seed=rank)
+    if args.dtype == "fp8" and on_gpu:
+        from jimm_amd.ops import set_fp8
+
+        set_fp8(True)
+    if args.task in ("vit", "vitl384"):
+        img = 224 if args.task == "vit" else 384
+        data = SyntheticImages(batch, img, 1000, device, dtype=dtype, seed=rank)
         items_per_step = batch * world  # images
         unit = "images/sec"
     else:
@@ -141,12 +155,12 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no throughput numbers
-            "dtype": "bf16" if on_gpu else "fp32",
+            "dtype": (args.dtype if on_gpu else "fp32"),
             "data": "synthetic",
             "config": {
                 "model": model_name,
                 "global_batch": items_per_step,
-                "img_size": {"vit": 224, "clip": 224, "siglip": 256}[args.task],
+                "img_size": {"vit": 224, "vitl384": 384, "clip": 224, "siglip": 256}[args.task],
                 "parallelism": f"dp{world}",
             },
         }

@@ -32,6 +32,7 @@ class VisionTransformer(nn.Module):
         do_classification: bool = True,
         hidden_act: str = "gelu",
         layernorm_epsilon: float = 1e-12,
+        pooling: str = "CLS",  # "CLS" | "MAP" (BASELINE config 5: ViT-L/384 MAP)
     ) -> None:
         super().__init__()
         self.do_classification = do_classification
@@ -47,7 +48,7 @@ class VisionTransformer(nn.Module):
             use_pre_norm=False,
             use_patch_bias=True,
             hidden_act=hidden_act,
-            pooling="CLS",
+            pooling=pooling,
             layernorm_epsilon=layernorm_epsilon,
         )
         if do_classification:
@@ -56,7 +57,7 @@ class VisionTransformer(nn.Module):
             self.classifier = None
 
     def forward(self, images: torch.Tensor) -> torch.Tensor:
-        x = self.vision(images)  # (B, H) CLS-pooled
+        x = self.vision(images)  # (B, H) CLS- or MAP-pooled
         if self.do_classification:
             x = ops.linear(x, self.classifier.weight, self.classifier.bias)
         return x

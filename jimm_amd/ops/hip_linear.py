@@ -20,6 +20,50 @@ import torch
 from jimm_amd.ops import _backend
 
 
+# ---------------------------------------------------------------------------
+# fp8 forward path (BASELINE config 5): per-tensor dynamic-scaled e4m3 GEMM
+# via torch._scaled_mm (hipBLASLt fp8 MFMA — measured 1.4-3.0 PF/s on the
+# model shapes, benchmarks/fp8_probe.py, vs ~1.0 PF/s bf16). Backward stays
+# bf16 (dX/dW GEMMs on the saved bf16 activations); LN/softmax/losses are
+# bf16/fp32 throughout. Enable with jimm_amd.ops.set_fp8(True).
+# ---------------------------------------------------------------------------
+
+_FP8_STATE = {"enabled": False}
+_FP8_MAX = 448.0  # e4m3 (OCP) max normal
+
+
+def set_fp8(on: bool = True) -> None:
+    _FP8_STATE["enabled"] = bool(on)
+
+
+def fp8_enabled() -> bool:
+    return _FP8_STATE["enabled"]
+
+
+def _fp8_ok(x2: torch.Tensor, w: torch.Tensor) -> bool:
+    # _scaled_mm needs K % 16 == 0 and fp8-capable torch; bf16 inputs only
+    return (
+        x2.dtype == torch.bfloat16
+        and x2.shape[1] % 16 == 0
+        and w.shape[0] % 16 == 0
+        and hasattr(torch, "_scaled_mm")
+    )
+
+
+def _quant_e4m3(t: torch.Tensor):
+    scale = (t.abs().amax().float() / _FP8_MAX).clamp(min=1e-12)
+    t8 = (t * scale.reciprocal().to(t.dtype)).to(torch.float8_e4m3fn)
+    return t8, scale
+
+
+def _gemm_nt_fp8(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    x8, sx = _quant_e4m3(x2)
+    w8, sw = _quant_e4m3(w)
+    return torch._scaled_mm(
+        x8, w8.t(), scale_a=sx.view(1, 1), scale_b=sw.view(1, 1), out_dtype=torch.bfloat16
+    )
+
+
 def _gemm_mode() -> str:
     # Default "blas": measured on MI355X (benchmarks/gemm_bench.py, r03): the
     # in-house 256-tile MFMA GEMM reaches 542-868 TF/s vs rocBLAS 845-1089 on
@@ -41,7 +85,10 @@ class _LinearActFn(torch.autograd.Function):
         x2 = x.contiguous().reshape(-1, in_f)
         res2 = residual.contiguous().reshape(-1, out_f) if residual is not None else None
         ext = _backend.ext()
-        if _gemm_mode() == "hip" and ext.gemm_supported(x2.shape[0], out_f, in_f, str(x2.dtype)):
+        if _FP8_STATE["enabled"] and _fp8_ok(x2, w):
+            z = _gemm_nt_fp8(x2, w)
+            y = ext.bias_act_fwd(z, b, act or "", res2)
+        elif _gemm_mode() == "hip" and ext.gemm_supported(x2.shape[0], out_f, in_f, str(x2.dtype)):
             y, z = ext.linear_fwd(x2, w, b, act or "", res2, act is not None)
         else:
             z = _gemm_nt(x2, w)
