@@ -466,3 +466,23 @@ def test_graph_captured_training():
     for p, q in zip(eager, graphed):
         assert torch.isfinite(q).all()
         assert (p - q).abs().max().item() < 0.2, (p - q).abs().max().item()
+
+
+def test_fp8_linear_path():
+    """fp8 e4m3 forward GEMM (scaled_mm) vs bf16 reference, loose tolerance."""
+    from jimm_amd import ops
+    from jimm_amd.ops import set_fp8
+
+    torch.manual_seed(0)
+    x = torch.randn(512, 768, device=dev()).bfloat16().requires_grad_(True)
+    w = (torch.randn(3072, 768, device=dev()) / 28.0).bfloat16()
+    b = torch.randn(3072, device=dev()).bfloat16()
+    set_fp8(True)
+    try:
+        y = ops.linear(x, w, b, act="gelu")
+        y.sum().backward()
+    finally:
+        set_fp8(False)
+    ref = torch.nn.functional.gelu(x.detach().float() @ w.float().t() + b.float())
+    assert rel_err(y, ref) < 0.08, rel_err(y, ref)   # fp8 quantization tolerance
+    assert torch.isfinite(x.grad).all()
