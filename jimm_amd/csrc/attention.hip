@@ -48,11 +48,14 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   // (sb, sh, sl) = element strides of the (B,H,L,64) view; innermost dim is
   // contiguous. Covers contiguous BHLD, (B,L,H,64) and fused-qkv (B,L,3,H,64)
   // layouts with no permute copies.
-  // LDS carve: K [64][72], V^T [64][72], P per-wave 4x[16][72] (shorts)
+  // LDS carve (double-buffered K/V^T, T14 split staging — issue next tile's
+  // global loads before this tile's compute, ds_write after the barrier;
+  // ONE barrier per KV tile): K [2][64][72], V^T [2][64][72],
+  // P per-wave 4x[16][72] (shorts)
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds = reinterpret_cast<short*>(smem);                       // 64*72
-  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // 64*72
-  short* p_lds = vt_lds + D * LDS_PITCH;                               // 4*16*72
+  short* k_lds0 = reinterpret_cast<short*>(smem);                      // 2*64*72
+  short* vt_lds0 = k_lds0 + 2 * KVBLK * LDS_PITCH;                     // 2*64*72
+  short* p_lds = vt_lds0 + 2 * D * LDS_PITCH;                          // 4*16*72
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -86,32 +89,58 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   const int kv_end = CAUSAL ? min(Lk, blockIdx.x * QBLK + QBLK) : Lk;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-    // ---- stage K tile (row-major) and V tile (transposed) ----------------
-    {
-      const int row = tid / 4;          // 0..63 (key within tile)
-      const int c0 = (tid % 4) * 16;    // 16 shorts per thread
-      const int key = kv0 + row;
-      if (key < Lk) {
-        const bf16x8_t k0 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0);
-        const bf16x8_t k1 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0 + 8);
-        *reinterpret_cast<bf16x8_t*>(k_lds + row * LDS_PITCH + c0) = k0;
-        *reinterpret_cast<bf16x8_t*>(k_lds + row * LDS_PITCH + c0 + 8) = k1;
-        const bf16x8_t v0 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + c0);
-        const bf16x8_t v1 = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + c0 + 8);
+  // this thread's staging slot: row = key within tile, 16-short chunk
+  const int st_row = tid / 4;
+  const int st_c0 = (tid % 4) * 16;
+
+  bf16x8_t kreg[2], vreg[2];
+  bool reg_valid;
+  auto load_tile_regs = [&](int kv0) {
+    const int key = kv0 + st_row;
+    reg_valid = key < Lk;
+    const int krow = min(key, Lk - 1);
+    kreg[0] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)krow * k_sl + st_c0);
+    kreg[1] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)krow * k_sl + st_c0 + 8);
+    vreg[0] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)krow * v_sl + st_c0);
+    vreg[1] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)krow * v_sl + st_c0 + 8);
+  };
+  auto write_tile = [&](int buf) {
+    short* kd = k_lds0 + buf * KVBLK * LDS_PITCH;
+    short* vd = vt_lds0 + buf * D * LDS_PITCH;
+    if (reg_valid) {
+      *reinterpret_cast<bf16x8_t*>(kd + st_row * LDS_PITCH + st_c0) = kreg[0];
+      *reinterpret_cast<bf16x8_t*>(kd + st_row * LDS_PITCH + st_c0 + 8) = kreg[1];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) vt_lds[(c0 + i) * LDS_PITCH + row] = v0[i];
+      for (int i = 0; i < 8; ++i) vd[(st_c0 + i) * LDS_PITCH + st_row] = vreg[0][i];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) vt_lds[(c0 + 8 + i) * LDS_PITCH + row] = v1[i];
-      } else {
-        // zero-fill so mfma on the padded tail contributes nothing
-        for (int i = 0; i < 16; ++i) k_lds[row * LDS_PITCH + c0 + i] = 0;
-        for (int i = 0; i < 16; ++i) vt_lds[(c0 + i) * LDS_PITCH + row] = 0;
-      }
+      for (int i = 0; i < 8; ++i) vd[(st_c0 + 8 + i) * LDS_PITCH + st_row] = vreg[1][i];
+    } else {
+      // zero-fill so mfma on the padded tail contributes nothing
+      for (int i = 0; i < 16; ++i) kd[st_row * LDS_PITCH + st_c0 + i] = 0;
+      for (int i = 0; i < 16; ++i) vd[(st_c0 + i) * LDS_PITCH + st_row] = 0;
     }
-    __syncthreads();
+  };
+
+  const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
+  load_tile_regs(0);
+  write_tile(0);
+  if (ntiles > 1) load_tile_regs(KVBLK);
+  __syncthreads();
+
+  for (int it = 0; it < ntiles; ++it) {
+    const int kv0 = it * KVBLK;
+    const int buf = it & 1;
+    short* k_lds = k_lds0 + buf * KVBLK * LDS_PITCH;
+    short* vt_lds = vt_lds0 + buf * D * LDS_PITCH;
+    // T14 write-after-barrier: tile it+1 (regs loaded last iteration) goes
+    // into the other buffer, then tile it+2's loads are issued immediately
+    if (it + 1 < ntiles) {
+      write_tile(buf ^ 1);
+      if (it + 2 < ntiles) load_tile_regs(kv0 + 2 * KVBLK);
+    }
 
     // ---- S^T = K . Q^T : 4 key tiles x 2 d-steps -------------------------
+    __builtin_amdgcn_s_setprio(1);
     f32x4_t sc[4] = {};
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -122,6 +151,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         sc[t] = MFMA16(ka, qb[s], sc[t]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- masked, scaled scores; per-row (q = lo) online softmax ----------
     float sv[16];
@@ -177,6 +207,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int r = 0; r < 4; ++r) acc_o[dt][r] *= alpha_r[r];
 
     // ---- O += P . V : A = P (from LDS), B = V^T reads --------------------
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
       const bf16x8_t pa =
@@ -188,7 +219,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         acc_o[dt] = MFMA16(pa, vb, acc_o[dt]);
       }
     }
-    __syncthreads();  // K/V^T tiles reused next iteration
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();  // next tile's ds_writes (into the other buffer) + reuse
   }
 
   // ---- epilogue: O /= l, store O and lse ----------------------------------
@@ -228,7 +260,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
   const dim3 grid((Lq + QBLK - 1) / QBLK, (unsigned)((int64_t)B * H));
-  const size_t shmem = (KVBLK * LDS_PITCH + D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
+  const size_t shmem = (2 * KVBLK * LDS_PITCH + 2 * D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
 #define ATTN_LAUNCH(C)                                                                     \
   hipLaunchKernelGGL((attn_fwd_kernel<C>), grid, dim3(256), shmem, stream,                 \

@@ -85,15 +85,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dk_sb, int64_t dk_sh, int64_t dk_sl, int64_t dv_sb, int64_t dv_sh, int64_t dv_sl) {
-  // LDS: K [64][72], V [64][72] staged once (row-major); Q^T [64][72] and
-  // dO^T [64][72] re-staged per q tile (scatter-transpose, like the forward
-  // kernel's V^T); per-wave P/dS tile 4x[16][72]. Total ~46 KiB.
+  // LDS: K [64][72], V [64][72] staged once (row-major); Q^T and dO^T
+  // DOUBLE-buffered [2][64][72] each (T14 split staging: next tile's global
+  // loads issue before this tile's MFMA, the scatter-transpose writes land
+  // after the barrier — one barrier per q tile); per-wave P/dS tile
+  // 4x[16][72]. Total ~64 KiB.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);
   short* v_lds = k_lds + BLK * PITCH;
-  short* qt_lds = v_lds + BLK * PITCH;
-  short* dot_lds = qt_lds + D * PITCH;
-  short* p_lds = dot_lds + D * PITCH;
+  short* qt_lds0 = v_lds + BLK * PITCH;
+  short* dot_lds0 = qt_lds0 + 2 * D * PITCH;
+  short* p_lds = dot_lds0 + 2 * D * PITCH;
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -145,27 +147,53 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   f32x4_t acc_dv[4] = {};
 
   const int q_start = CAUSAL ? (kv0 / BLK) * BLK : 0;
+  const int ntiles = (Lq - q_start + BLK - 1) / BLK;
 
-  for (int q0 = q_start; q0 < Lq; q0 += BLK) {
-    // ---- stage Q^T and dO^T for this q tile (scatter-transpose) ----------
-    {
-      const int row = tid / 4;          // q index within tile
-      const int c0 = (tid % 4) * 16;
-      const int qi = q0 + row;
-      if (qi < Lq) {
+  // per-thread staging slot for the scatter-transpose
+  const int st_row = tid / 4;
+  const int st_c0 = (tid % 4) * 16;
+  bf16x8_t qreg[2], doreg[2];
+  bool st_valid;
+  auto load_stage_regs = [&](int q0) {
+    const int qi = q0 + st_row;
+    st_valid = qi < Lq;
+    const int qr = min(qi, Lq - 1);
 #pragma unroll
-        for (int hh = 0; hh < 2; ++hh) {
-          const bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qi * q_sl + c0 + hh * 8);
-          const bf16x8_t dv8 = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qi * do_sl + c0 + hh * 8);
+    for (int hh = 0; hh < 2; ++hh) {
+      qreg[hh] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qr * q_sl + st_c0 + hh * 8);
+      doreg[hh] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qr * do_sl + st_c0 + hh * 8);
+    }
+  };
+  auto write_stage = [&](int buf) {
+    short* qt_lds = qt_lds0 + buf * D * PITCH;
+    short* dot_lds = dot_lds0 + buf * D * PITCH;
+    if (st_valid) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) qt_lds[(c0 + hh * 8 + i) * PITCH + row] = qv[i];
+      for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) dot_lds[(c0 + hh * 8 + i) * PITCH + row] = dv8[i];
-        }
-      } else {
-        for (int i = 0; i < 16; ++i) qt_lds[(c0 + i) * PITCH + row] = 0;
-        for (int i = 0; i < 16; ++i) dot_lds[(c0 + i) * PITCH + row] = 0;
+        for (int i = 0; i < 8; ++i) qt_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = qreg[hh][i];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) dot_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = doreg[hh][i];
       }
+    } else {
+      for (int i = 0; i < 16; ++i) qt_lds[(st_c0 + i) * PITCH + st_row] = 0;
+      for (int i = 0; i < 16; ++i) dot_lds[(st_c0 + i) * PITCH + st_row] = 0;
+    }
+  };
+
+  load_stage_regs(q_start);
+  write_stage(0);
+  if (ntiles > 1) load_stage_regs(q_start + BLK);
+  __syncthreads();
+
+  for (int it = 0; it < ntiles; ++it) {
+    const int q0 = q_start + it * BLK;
+    const int buf = it & 1;
+    short* qt_lds = qt_lds0 + buf * D * PITCH;
+    short* dot_lds = dot_lds0 + buf * D * PITCH;
+    if (it + 1 < ntiles) {
+      write_stage(buf ^ 1);
+      if (it + 2 < ntiles) load_stage_regs(q0 + 2 * BLK);
     }
 
     // ---- B-fragments of Q^T and dO^T from global (b128, L2-resident) ------
@@ -180,7 +208,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
         dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qi * do_sl + 32 * s + hi * 8);
       }
     }
-    __syncthreads();
 
     // ---- S^T = K . Q^T ; P^T = exp(scale*S^T - lse[q]) -------------------
     // C rows = key = 16*wave + hi*4+r, cols = q = 16*qt + lo
@@ -271,10 +298,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dq_sb, int64_t dq_sh, int64_t dq_sl) {
-  // LDS: K^T [64][72] re-staged per kv tile; per-wave dS tile 4x[16][72].
+  // LDS: K^T double-buffered [2][64][72] (T14 split staging, one barrier
+  // per kv tile); per-wave dS tile 4x[16][72].
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* kt_lds = reinterpret_cast<short*>(smem);
-  short* ds_lds = kt_lds + D * PITCH;
+  short* kt_lds0 = reinterpret_cast<short*>(smem);
+  short* ds_lds = kt_lds0 + 2 * D * PITCH;
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -314,23 +342,44 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   f32x4_t acc_dq[4] = {};  // rows q = hi*4+r, cols d = 16*dt + lo
 
   const int kv_end = CAUSAL ? min(Lk, blockIdx.x * BLK + BLK) : Lk;
+  const int ntiles = (kv_end + BLK - 1) / BLK;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += BLK) {
-    // ---- stage K^T for this kv tile (scatter-transpose) ------------------
-    {
-      const int row = tid / 4;          // key within tile
-      const int c0 = (tid % 4) * 16;
-      const int key = kv0 + row;
-      if (key < Lk) {
+  const int st_row = tid / 4;
+  const int st_c0 = (tid % 4) * 16;
+  bf16x8_t kreg[2];
+  bool st_valid;
+  auto load_stage_regs = [&](int kv0) {
+    const int key = kv0 + st_row;
+    st_valid = key < Lk;
+    const int kr = min(key, Lk - 1);
 #pragma unroll
-        for (int hh = 0; hh < 2; ++hh) {
-          const bf16x8_t kv8 = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0 + hh * 8);
+    for (int hh = 0; hh < 2; ++hh)
+      kreg[hh] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)kr * k_sl + st_c0 + hh * 8);
+  };
+  auto write_stage = [&](int buf) {
+    short* kt_lds = kt_lds0 + buf * D * PITCH;
+    if (st_valid) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) kt_lds[(c0 + hh * 8 + i) * PITCH + row] = kv8[i];
-        }
-      } else {
-        for (int i = 0; i < 16; ++i) kt_lds[(c0 + i) * PITCH + row] = 0;
-      }
+      for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+        for (int i = 0; i < 8; ++i) kt_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = kreg[hh][i];
+    } else {
+      for (int i = 0; i < 16; ++i) kt_lds[(st_c0 + i) * PITCH + st_row] = 0;
+    }
+  };
+
+  load_stage_regs(0);
+  write_stage(0);
+  if (ntiles > 1) load_stage_regs(BLK);
+  __syncthreads();
+
+  for (int it = 0; it < ntiles; ++it) {
+    const int kv0 = it * BLK;
+    const int buf = it & 1;
+    short* kt_lds = kt_lds0 + buf * D * PITCH;
+    if (it + 1 < ntiles) {
+      write_stage(buf ^ 1);
+      if (it + 2 < ntiles) load_stage_regs(kv0 + 2 * BLK);
     }
 
     // ---- B-fragments of K^T and V^T from global --------------------------
@@ -345,7 +394,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + 32 * s + hi * 8);
       }
     }
-    __syncthreads();
 
     // ---- S, P, dP, dS per 16-key tile ------------------------------------
     // C rows = q = hi*4+r, cols = key = 16*kt + lo
@@ -423,8 +471,8 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
                        o.stride(0), o.stride(1), o.stride(2));
   }
 
-  const size_t shmem_dkv = (4 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
-  const size_t shmem_dq = (D * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const size_t shmem_dkv = (6 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const size_t shmem_dq = (2 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
   const dim3 grid_dkv((Lk + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
   const dim3 grid_dq((Lq + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
 
