@@ -18,13 +18,26 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
+# (M, N, K) of the forward x(M,K) @ w(N,K)^T — full model-zoo shape set
+# (ViT-B/16@224 bs256, ViT-L/16@384 bs64, CLIP-B/32 bs256 both towers,
+# SigLIP-base/16-256 bs256 both towers, patch embeds, projections)
 SHAPES = [
-    # (M, N, K) of the forward x(M,K) @ w(N,K)^T — ViT-B/16 @ bs256
-    (50432, 2304, 768),   # QKV
-    (50432, 768, 768),    # attn proj
-    (50432, 3072, 768),   # fc1
-    (50432, 768, 3072),   # fc2
-    (50176, 768, 768),    # patch embed
+    # ViT-B/16 @ 224, bs 256 (M = 256*197)
+    (50432, 2304, 768), (50432, 768, 768), (50432, 3072, 768), (50432, 768, 3072),
+    (50176, 768, 768),             # patch embed (M = 256*196)
+    (256, 1000, 768),              # classifier
+    # ViT-L/16 @ 384, bs 64 (M = 64*577)
+    (36928, 3072, 1024), (36928, 1024, 1024), (36928, 4096, 1024), (36928, 1024, 4096),
+    (36864, 1024, 768),            # patch embed
+    # CLIP-B/32 vision bs 256 (M = 256*50) + text (M = 256*77)
+    (12800, 2304, 768), (12800, 768, 768), (12800, 3072, 768), (12800, 768, 3072),
+    (12544, 768, 3072),            # patch embed (K = 3*32*32)
+    (19712, 1536, 512), (19712, 512, 512), (19712, 2048, 512), (19712, 512, 2048),
+    (256, 512, 768), (256, 512, 512),   # visual/text projections
+    # SigLIP-base/16-256 vision bs 256 (M = 256*256) + text (M = 256*64)
+    (65536, 2304, 768), (65536, 768, 768), (65536, 3072, 768), (65536, 768, 3072),
+    (16384, 2304, 768), (16384, 768, 768), (16384, 3072, 768), (16384, 768, 3072),
+    (256, 768, 768),               # text head / MAP-head linears
 ]
 
 
@@ -51,7 +64,7 @@ def main():
             tun.set_filename(args.tune)  # auto-written at process exit
             tun.enable(True)
             tun.tuning_enable(True)
-            tun.set_max_tuning_duration(50)
+            tun.set_max_tuning_duration(100)
         else:
             tun.enable(True)
             tun.tuning_enable(False)

@@ -17,17 +17,29 @@ and A/B numerics only; never the default).
 from __future__ import annotations
 
 
+_DEFAULT_TUNABLE = "data/tunableop_mi355x.csv"  # relative to jimm_amd/
+
+
 def maybe_enable_tunableop() -> bool:
-    """Load a committed TunableOp result table (hipBLASLt algo selections)
-    when JIMM_AMD_TUNABLE points at a CSV produced by
-    benchmarks/bwd_gemm_bench.py --tune. Returns True if enabled."""
+    """Load a TunableOp result table (hipBLASLt algo selections, including
+    the split-K picks for the dW shapes — measured 0.23-0.6 -> 0.6-0.9 PF/s).
+
+    Default: the committed jimm_amd/data/tunableop_mi355x.csv (produced by
+    benchmarks/bwd_gemm_bench.py --tune on an MI355X). Override the path with
+    JIMM_AMD_TUNABLE=<csv>; disable with JIMM_AMD_TUNABLE=0."""
     import os
 
     path = os.environ.get("JIMM_AMD_TUNABLE", "")
-    if not path:
+    if path in ("0", "none", "off"):
         return False
+    if not path:
+        path = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", _DEFAULT_TUNABLE)
+        if not os.path.exists(path):
+            return False
     import torch
 
+    if not torch.cuda.is_available():
+        return False
     tun = torch.cuda.tunable
     tun.enable(True)
     tun.tuning_enable(False)
