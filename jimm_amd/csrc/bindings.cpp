@@ -4,7 +4,8 @@
 
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps);
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
-                                         torch::Tensor mean, torch::Tensor rstd);
+                                         torch::Tensor mean, torch::Tensor rstd,
+                                         c10::optional<torch::Tensor> addend);
 torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, std::string act,
                            c10::optional<torch::Tensor> residual);
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act);
@@ -46,7 +47,9 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (K3)");
-  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (K3/K15)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (K3/K15), optional fused addend",
+        py::arg("dy"), py::arg("x"), py::arg("w"), py::arg("mean"), py::arg("rstd"),
+        py::arg("addend") = py::none());
   m.def("bias_act_fwd", &bias_act_fwd, "fused bias+activation(+residual) forward");
   m.def("act_bwd", &act_bwd, "activation backward: dy * act'(z)");
   m.def("colsum", &colsum, "column sum -> fp32 (bias gradient, K15)");

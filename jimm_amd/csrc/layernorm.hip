@@ -62,12 +62,12 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
   }
 }
 
-template <typename T, int VEC>
+template <typename T, int VEC, bool HAS_ADD>
 __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                               const float* __restrict__ w, const float* __restrict__ mean,
-                              const float* __restrict__ rstd, T* __restrict__ dx,
-                              float* __restrict__ dw, float* __restrict__ db,
-                              int64_t nrows, int H) {
+                              const float* __restrict__ rstd, const T* __restrict__ addend,
+                              T* __restrict__ dx, float* __restrict__ dw,
+                              float* __restrict__ db, int64_t nrows, int H) {
   // per-wave fp32 partial slabs: [waves][H] for dw and db — no atomics in
   // the row loop (each wave owns its slab; each lane its columns)
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -104,6 +104,7 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     }
     c1 = wave_reduce_sum(c1) / H;
     c2 = wave_reduce_sum(c2) / H;
+    const T* addr = HAS_ADD ? addend + row * H : nullptr;
     for (int i = lane * VEC; i < H; i += WAVE * VEC) {
       float dyv[VEC], xv[VEC], wv[VEC], dxv[VEC];
       vload_f32<VEC>(dyr + i, dyv);
@@ -115,6 +116,12 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
         dxv[k] = rs * (dyv[k] * wv[k] - c1 - xhat * c2);
         s_dw[i + k] += dyv[k] * xhat;
         s_db[i + k] += dyv[k];
+      }
+      if (HAS_ADD) {
+        float av[VEC];
+        vload_f32<VEC>(addr + i, av);
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) dxv[k] += av[k];  // fused residual grad
       }
       vstore_f32<VEC>(dxr + i, dxv);
     }
@@ -154,16 +161,20 @@ void launch_ln_fwd(const T* x, const float* w, const float* b, T* y, float* mean
 
 template <typename T>
 void launch_ln_bwd(const T* dy, const T* x, const float* w, const float* mean,
-                   const float* rstd, T* dx, float* dw, float* db, int64_t nrows, int H,
-                   hipStream_t stream) {
+                   const float* rstd, const T* addend, T* dx, float* dw, float* db,
+                   int64_t nrows, int H, hipStream_t stream) {
   const int block = 256;
   const int waves_per_block = block / WAVE;
   int grid = (int)std::min<int64_t>((nrows + waves_per_block - 1) / waves_per_block, 1024);
   size_t shmem = 2 * (size_t)waves_per_block * H * sizeof(float);
   auto pick = [&](auto vec_tag) {
     constexpr int V = decltype(vec_tag)::value;
-    hipLaunchKernelGGL((ln_bwd_kernel<T, V>), dim3(grid), dim3(block), shmem, stream, dy,
-                       x, w, mean, rstd, dx, dw, db, nrows, H);
+    if (addend)
+      hipLaunchKernelGGL((ln_bwd_kernel<T, V, true>), dim3(grid), dim3(block), shmem,
+                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, nrows, H);
+    else
+      hipLaunchKernelGGL((ln_bwd_kernel<T, V, false>), dim3(grid), dim3(block), shmem,
+                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, nrows, H);
   };
   if (H % (WAVE * 4) == 0) pick(std::integral_constant<int, 4>{});
   else if (H % (WAVE * 2) == 0) pick(std::integral_constant<int, 2>{});
@@ -200,7 +211,10 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w, torch
 }
 
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
-                                         torch::Tensor mean, torch::Tensor rstd) {
+                                         torch::Tensor mean, torch::Tensor rstd,
+                                         c10::optional<torch::Tensor> addend) {
+  // addend (optional, same shape as x): fused into dx — the residual-branch
+  // gradient of a pre-LN transformer block lands here for free
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
   const int H = x.size(-1);
   const int64_t nrows = x.numel() / H;
@@ -209,15 +223,23 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
   auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   auto db = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
+  const void* addp = nullptr;
+  if (addend) {
+    TORCH_CHECK(addend->is_contiguous() && addend->scalar_type() == x.scalar_type() &&
+                addend->numel() == x.numel());
+    addp = addend->data_ptr();
+  }
   if (x.scalar_type() == torch::kBFloat16) {
     launch_ln_bwd(reinterpret_cast<const bf16*>(dy.data_ptr()),
                   reinterpret_cast<const bf16*>(x.data_ptr()), wf.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  reinterpret_cast<const bf16*>(addp),
                   reinterpret_cast<bf16*>(dx.data_ptr()), dw.data_ptr<float>(),
                   db.data_ptr<float>(), nrows, H, stream);
   } else if (x.scalar_type() == torch::kFloat32) {
     launch_ln_bwd(dy.data_ptr<float>(), x.data_ptr<float>(), wf.data_ptr<float>(),
-                  mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  reinterpret_cast<const float*>(addp), dx.data_ptr<float>(),
                   dw.data_ptr<float>(), db.data_ptr<float>(), nrows, H, stream);
   } else {
     TORCH_CHECK(false, "layernorm: unsupported dtype ", x.scalar_type());

@@ -57,7 +57,41 @@ class EncoderBlock(nn.Module):
         self.fc2 = nn.Linear(mlp_dim, hidden_size, bias=True)
         self.dropout = nn.Dropout(dropout_rate)
 
+    def _forward_tp(self, x: torch.Tensor) -> torch.Tensor:
+        """Tensor-parallel path (parallel/tp.py): local heads + local MLP
+        shard, partial out-projections all-reduced over the TP group."""
+        from jimm_amd.parallel.tp import copy_to_tp, reduce_from_tp
+
+        B, L, H = x.shape
+        g = self._tp_group
+        h = ops.layer_norm(x, self.norm1.weight, self.norm1.bias, self.eps)
+        h = copy_to_tp(h, g)
+        qkv = ops.linear(h, self.qkv.weight, self.qkv.bias)  # (B, L, 3*H_local)
+        qkv = qkv.view(B, L, 3, self.num_heads, self.head_dim)
+        o = ops.attention_qkv(qkv, causal=self.causal)       # local heads
+        o = o.transpose(1, 2).reshape(B, L, self.num_heads * self.head_dim)
+        part = ops.linear(o, self.proj.weight)               # partial, no bias
+        x = reduce_from_tp(part, g) + self.proj.bias + x
+
+        h = ops.layer_norm(x, self.norm2.weight, self.norm2.bias, self.eps)
+        h = copy_to_tp(h, g)
+        h = ops.linear(h, self.fc1.weight, self.fc1.bias, act=self.act)
+        part = ops.linear(h, self.fc2.weight)
+        return reduce_from_tp(part, g) + self.fc2.bias + x
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if getattr(self, "_tp_group", None) is not None:
+            return self._forward_tp(x)
+        from jimm_amd.ops import block as fused
+
+        if fused.fused_block_enabled(x, self.dropout.p):
+            import math
+
+            return fused.encoder_block(
+                x, self.norm1, self.qkv, self.proj, self.norm2, self.fc1, self.fc2,
+                num_heads=self.num_heads, act=self.act, eps=self.eps,
+                causal=self.causal, scale=1.0 / math.sqrt(self.head_dim),
+            )
         B, L, H = x.shape
         h = ops.layer_norm(x, self.norm1.weight, self.norm1.bias, self.eps)
         qkv = ops.linear(h, self.qkv.weight, self.qkv.bias)  # (B, L, 3H)

@@ -1,0 +1,132 @@
+"""Whole-encoder-block autograd Function (GPU fast path).
+
+The composite EncoderBlock (models/common/transformer.py) is numerically
+identical but lets autograd insert two full-tensor grad-accumulation adds
+per block (the residual branches: x feeds both LN1 and the attention
+residual; a feeds both LN2 and the MLP residual — ~1.1 ms/step for ViT-B
+bs256, profiles/r01_NOTES.md). Here the block is one Function and those
+adds ride the ln_bwd kernel's fused ``addend`` input for free. The forward
+matches the composite blas-mode path kernel-for-kernel.
+
+Enabled on GPU when dropout == 0 (the composite path remains the reference
+and the dropout/CPU path). JIMM_AMD_FUSED_BLOCK=0 disables.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+from jimm_amd.ops import _backend
+
+
+def _colsum(ext, dz: torch.Tensor) -> torch.Tensor:
+    if dz.shape[-1] % 8 == 0:
+        return ext.colsum(dz).to(dz.dtype)
+    return dz.sum(dim=0)
+
+
+def fused_block_enabled(x: torch.Tensor, dropout_p: float) -> bool:
+    return (
+        x.is_cuda
+        and dropout_p == 0.0
+        and x.dtype == torch.bfloat16
+        and _backend.use_hip(x)
+        and os.environ.get("JIMM_AMD_FUSED_BLOCK", "1") == "1"
+    )
+
+
+class EncoderBlockFn(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx, x, ln1w, ln1b, wqkv, bqkv, wproj, bproj, ln2w, ln2b, w1, b1, w2, b2,
+        num_heads: int, act: str, eps: float, causal: bool, scale: float,
+    ):
+        ext = _backend.ext()
+        B, L, H = x.shape
+        x = x.contiguous()
+        x2 = x.view(-1, H)
+        h1, mean1, rstd1 = ext.layernorm_fwd(x, ln1w, ln1b, eps)
+        qkv2 = torch.addmm(bqkv, h1.view(-1, H), wqkv.t())       # (M, 3H) fused bias
+        qkv = qkv2.view(B, L, 3, num_heads, H // num_heads)
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        o, lse = ext.attn_fwd(q, k, v, causal, scale)            # (B,nh,L,d), (B,L,nh,d) storage
+        o2 = o.transpose(1, 2).reshape(-1, H)                    # free view
+        a2 = torch.matmul(o2, wproj.t())
+        a = ext.bias_act_fwd(a2, bproj, "", x2)                  # + bias + residual (one pass)
+        a3 = a.view(B, L, H)
+        h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)
+        z1 = torch.matmul(h2.view(-1, H), w1.t())
+        f = ext.bias_act_fwd(z1, b1, act, None)                  # z1 -> pre-activation in place
+        y2 = torch.matmul(f, w2.t())
+        y = ext.bias_act_fwd(y2, b2, "", a)                      # + bias + residual
+        ctx.save_for_backward(x, ln1w, wqkv, wproj, ln2w, w1, w2,
+                              h1, qkv, o, lse, a, mean1, rstd1, mean2, rstd2, h2, z1, f)
+        ctx.dims = (B, L, H, num_heads)
+        ctx.meta = (act, causal, scale)
+        return y.view(B, L, H)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _backend.ext()
+        (x, ln1w, wqkv, wproj, ln2w, w1, w2,
+         h1, qkv, o, lse, a, mean1, rstd1, mean2, rstd2, h2, z1, f) = ctx.saved_tensors
+        B, L, H, nh = ctx.dims
+        act, causal, scale = ctx.meta
+        dy2 = dy.contiguous().view(-1, H)
+        h1_2 = h1.view(-1, H)
+        h2_2 = h2.view(-1, H)
+
+        # MLP fc2 (+residual into a)
+        df = torch.matmul(dy2, w2)
+        dw2 = torch.matmul(dy2.t(), f)
+        db2 = _colsum(ext, dy2)
+        dz1 = ext.act_bwd(df, z1, act)
+        dh2 = torch.matmul(dz1, w1)
+        dw1 = torch.matmul(dz1.t(), h2_2)
+        db1 = _colsum(ext, dz1)
+        # LN2 backward with the MLP residual grad (dy) fused into dx
+        da3, dln2w, dln2b = ext.layernorm_bwd(
+            dh2.view(B, L, H), a.view(B, L, H), ln2w, mean2, rstd2, dy.contiguous()
+        )
+        da2 = da3.view(-1, H)
+
+        # attention out-projection
+        do2 = torch.matmul(da2, wproj)
+        dwproj = torch.matmul(da2.t(), o.transpose(1, 2).reshape(-1, H))
+        dbproj = _colsum(ext, da2)
+        do_v = do2.view(B, L, nh, H // nh).permute(0, 2, 1, 3)   # (B,nh,L,d) strided view
+
+        # fused flash attention backward straight into the strided dqkv
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        dqkv = torch.empty_like(qkv)
+        ext.attn_bwd_fused(
+            q, k, v, o, do_v, lse,
+            dqkv[:, :, 0].transpose(1, 2), dqkv[:, :, 1].transpose(1, 2),
+            dqkv[:, :, 2].transpose(1, 2), causal, scale,
+        )
+        dqkv2 = dqkv.view(-1, 3 * H)
+
+        # QKV projection
+        dh1 = torch.matmul(dqkv2, wqkv)
+        dwqkv = torch.matmul(dqkv2.t(), h1_2)
+        dbqkv = _colsum(ext, dqkv2)
+        # LN1 backward with the attention residual grad (da) fused into dx
+        dx, dln1w, dln1b = ext.layernorm_bwd(
+            dh1.view(B, L, H), x, ln1w, mean1, rstd1, da3
+        )
+        return (dx, dln1w, dln1b, dwqkv, dbqkv, dwproj, dbproj, dln2w, dln2b,
+                dw1, db1, dw2, db2, None, None, None, None, None)
+
+
+def encoder_block(x, norm1, qkv, proj, norm2, fc1, fc2, *, num_heads, act, eps, causal, scale):
+    return EncoderBlockFn.apply(
+        x, norm1.weight, norm1.bias, qkv.weight, qkv.bias, proj.weight, proj.bias,
+        norm2.weight, norm2.bias, fc1.weight, fc1.bias, fc2.weight, fc2.bias,
+        num_heads, act, eps, causal, scale,
+    )

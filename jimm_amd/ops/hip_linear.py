@@ -90,6 +90,10 @@ class _LinearActFn(torch.autograd.Function):
             y = ext.bias_act_fwd(z, b, act or "", res2)
         elif _gemm_mode() == "hip" and ext.gemm_supported(x2.shape[0], out_f, in_f, str(x2.dtype)):
             y, z = ext.linear_fwd(x2, w, b, act or "", res2, act is not None)
+        elif act is None and res2 is None and b is not None:
+            # bias-only: hipBLASLt fuses the bias epilogue into the GEMM —
+            # no separate full-tensor bias pass
+            y = z = torch.addmm(b, x2, w.t())
         else:
             z = _gemm_nt(x2, w)
             # fused bias+act(+residual): z is updated in place to hold the

@@ -171,3 +171,35 @@ def _siglip_invariance(rank, world):
 
 def test_dp_invariance_siglip():
     spawn(_siglip_invariance, port=29515)
+
+
+# ---------------------------------------------------------------------------
+def _check_tensor_parallel(rank, world):
+    from jimm_amd.models.common.transformer import EncoderBlock
+    from jimm_amd.parallel.tp import shard_encoder_block
+
+    torch.manual_seed(0)
+    blk = EncoderBlock(64, 4, 128, hidden_act="gelu", layernorm_epsilon=1e-6)
+    x = torch.randn(2, 9, 64)
+    dy = torch.randn(2, 9, 64)
+
+    ref_blk = EncoderBlock(64, 4, 128, hidden_act="gelu", layernorm_epsilon=1e-6)
+    ref_blk.load_state_dict(blk.state_dict())
+    xr = x.clone().requires_grad_(True)
+    yr = ref_blk(xr)
+    yr.backward(dy)
+
+    shard_encoder_block(blk, None)
+    xt = x.clone().requires_grad_(True)
+    yt = blk(xt)
+    yt.backward(dy)
+
+    assert torch.allclose(yt, yr, atol=1e-5), (yt - yr).abs().max()
+    assert torch.allclose(xt.grad, xr.grad, atol=1e-5), (xt.grad - xr.grad).abs().max()
+    mrows = slice(rank * (128 // world), (rank + 1) * (128 // world))
+    assert torch.allclose(blk.fc1.weight.grad, ref_blk.fc1.weight.grad[mrows], atol=1e-5)
+    assert torch.allclose(blk.norm1.weight.grad, ref_blk.norm1.weight.grad, atol=1e-5)
+
+
+def test_tensor_parallel_block():
+    spawn(_check_tensor_parallel, port=29515)

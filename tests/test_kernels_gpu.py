@@ -486,3 +486,35 @@ def test_fp8_linear_path():
     ref = torch.nn.functional.gelu(x.detach().float() @ w.float().t() + b.float())
     assert rel_err(y, ref) < 0.08, rel_err(y, ref)   # fp8 quantization tolerance
     assert torch.isfinite(x.grad).all()
+
+
+def test_fused_block_vs_composite():
+    """EncoderBlockFn (single-Function block) vs the composite autograd path:
+    same forward, same grads."""
+    import os
+
+    from jimm_amd.models.common.transformer import EncoderBlock
+
+    torch.manual_seed(3)
+    blk = EncoderBlock(128, 2, 256, hidden_act="gelu", layernorm_epsilon=1e-6).to(dev(), torch.bfloat16)
+    x = torch.randn(2, 197, 128, device=dev()).bfloat16()
+    dy = torch.randn_like(x)
+
+    def run():
+        for p in blk.parameters():
+            p.grad = None
+        xi = x.detach().clone().requires_grad_(True)
+        y = blk(xi)
+        y.backward(dy)
+        return y.detach(), xi.grad.clone(), {n: p.grad.clone() for n, p in blk.named_parameters()}
+
+    y_f, dx_f, g_f = run()
+    os.environ["JIMM_AMD_FUSED_BLOCK"] = "0"
+    try:
+        y_c, dx_c, g_c = run()
+    finally:
+        os.environ["JIMM_AMD_FUSED_BLOCK"] = "1"
+    assert rel_err(y_f, y_c) < 1e-2, rel_err(y_f, y_c)
+    assert rel_err(dx_f, dx_c) < 2e-2, rel_err(dx_f, dx_c)
+    for n in g_f:
+        assert rel_err(g_f[n], g_c[n]) < 3e-2, (n, rel_err(g_f[n], g_c[n]))
