@@ -49,27 +49,31 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
                                float* __restrict__ Dv, int H, int L, int64_t nrows,
                                int64_t do_sb, int64_t do_sh, int64_t do_sl,
                                int64_t o_sb, int64_t o_sh, int64_t o_sl) {
-  // one wave per (b,h,l) row; 8-wide vector loads
+  // 8 rows per wave (all 64 lanes loading): lane -> (sub-row l>>3, 8 cols)
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int wpb = blockDim.x / WAVE;
-  for (int64_t row = (int64_t)blockIdx.x * wpb + wave; row < nrows;
-       row += (int64_t)gridDim.x * wpb) {
-    const int64_t b = row / ((int64_t)H * L);
-    const int h = (int)((row / L) % H);
-    const int l = (int)(row % L);
+  const int sub = lane >> 3;          // 0..7 row within the wave's group
+  const int c0 = (lane & 7) * 8;      // 8 elements per lane
+  for (int64_t base = ((int64_t)blockIdx.x * wpb + wave) * 8; base < nrows;
+       base += (int64_t)gridDim.x * wpb * 8) {
+    const int64_t row = base + sub;
+    const int64_t rc = row < nrows ? row : nrows - 1;
+    const int64_t b = rc / ((int64_t)H * L);
+    const int h = (int)((rc / L) % H);
+    const int l = (int)(rc % L);
     const bf16* a = dO + b * do_sb + h * do_sh + (int64_t)l * do_sl;
     const bf16* o = O + b * o_sb + h * o_sh + (int64_t)l * o_sl;
+    float av[8], ov[8];
+    vload_f32<8>(a + c0, av);
+    vload_f32<8>(o + c0, ov);
     float acc = 0.f;
-    if (lane < 8) {  // 8 lanes x 8 elements = 64
-      float av[8], ov[8];
-      vload_f32<8>(a + lane * 8, av);
-      vload_f32<8>(o + lane * 8, ov);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc += av[j] * ov[j];
-    }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) Dv[row] = acc;
+    for (int j = 0; j < 8; ++j) acc += av[j] * ov[j];
+    // reduce across the 8 lanes of this sub-row
+#pragma unroll
+    for (int off = 1; off < 8; off <<= 1) acc += __shfl_xor(acc, off, WAVE);
+    if ((lane & 7) == 0 && row < nrows) Dv[row] = acc;
   }
 }
 
@@ -85,17 +89,22 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dk_sb, int64_t dk_sh, int64_t dk_sl, int64_t dv_sb, int64_t dv_sh, int64_t dv_sl) {
-  // LDS: K [64][72], V [64][72] staged once (row-major); Q^T and dO^T
-  // DOUBLE-buffered [2][64][72] each (T14 split staging: next tile's global
-  // loads issue before this tile's MFMA, the scatter-transpose writes land
-  // after the barrier — one barrier per q tile); per-wave P/dS tile
-  // 4x[16][72]. Total ~64 KiB.
+  // LDS: K [64][72], V [64][72] staged once (row-major); per q tile both a
+  // TRANSPOSED image (Q^T, dO^T — B-operands of the dK/dV mfma) and a
+  // ROW-major image (B-operands of the S^T/dP^T mfma) of Q and dO, all four
+  // written from the SAME staging registers (the r06 profile showed this
+  // kernel bandwidth-bound on reading Q/dO twice from global: once staged,
+  // once as global B-fragments — 3.1 GB/layer). T14 split staging, single
+  // buffer, two barriers per q tile. Per-wave P/dS tile 4x[16][72].
+  // Total ~64 KiB.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);
   short* v_lds = k_lds + BLK * PITCH;
-  short* qt_lds0 = v_lds + BLK * PITCH;
-  short* dot_lds0 = qt_lds0 + 2 * D * PITCH;
-  short* p_lds = dot_lds0 + 2 * D * PITCH;
+  short* qt_lds = v_lds + BLK * PITCH;       // Q^T  [64 d][72]
+  short* dot_lds = qt_lds + D * PITCH;       // dO^T [64 d][72]
+  short* qr_lds = dot_lds + D * PITCH;       // Q    [64 q][72]
+  short* dor_lds = qr_lds + BLK * PITCH;     // dO   [64 q][72]
+  short* p_lds = dor_lds + BLK * PITCH;
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -164,48 +173,43 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       doreg[hh] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qr * do_sl + st_c0 + hh * 8);
     }
   };
-  auto write_stage = [&](int buf) {
-    short* qt_lds = qt_lds0 + buf * D * PITCH;
-    short* dot_lds = dot_lds0 + buf * D * PITCH;
+  auto write_stage = [&]() {
     if (st_valid) {
 #pragma unroll
       for (int hh = 0; hh < 2; ++hh) {
+        *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + st_c0 + hh * 8) = qreg[hh];
+        *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + st_c0 + hh * 8) = doreg[hh];
 #pragma unroll
         for (int i = 0; i < 8; ++i) qt_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = qreg[hh][i];
 #pragma unroll
         for (int i = 0; i < 8; ++i) dot_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = doreg[hh][i];
       }
     } else {
-      for (int i = 0; i < 16; ++i) qt_lds[(st_c0 + i) * PITCH + st_row] = 0;
-      for (int i = 0; i < 16; ++i) dot_lds[(st_c0 + i) * PITCH + st_row] = 0;
+      for (int i = 0; i < 16; ++i) {
+        qr_lds[st_row * PITCH + st_c0 + i] = 0;
+        dor_lds[st_row * PITCH + st_c0 + i] = 0;
+        qt_lds[(st_c0 + i) * PITCH + st_row] = 0;
+        dot_lds[(st_c0 + i) * PITCH + st_row] = 0;
+      }
     }
   };
 
   load_stage_regs(q_start);
-  write_stage(0);
+  write_stage();
   if (ntiles > 1) load_stage_regs(q_start + BLK);
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int q0 = q_start + it * BLK;
-    const int buf = it & 1;
-    short* qt_lds = qt_lds0 + buf * D * PITCH;
-    short* dot_lds = dot_lds0 + buf * D * PITCH;
-    if (it + 1 < ntiles) {
-      write_stage(buf ^ 1);
-      if (it + 2 < ntiles) load_stage_regs(q0 + 2 * BLK);
-    }
-
-    // ---- B-fragments of Q^T and dO^T from global (b128, L2-resident) ------
+    // ---- B-fragments of Q^T and dO^T from the LDS row images --------------
     // B[k=d][j=q] element = Q[q0 + 16*qt + lo][32*s + hi*8 + jj]
     bf16x8_t qb[4][2], dob[4][2];
 #pragma unroll
     for (int qt = 0; qt < 4; ++qt) {
-      const int qi = min(q0 + 16 * qt + lo, Lq - 1);  // clamped; masked via P=0
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qi * q_sl + 32 * s + hi * 8);
-        dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qi * do_sl + 32 * s + hi * 8);
+        qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qr_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+        dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dor_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
       }
     }
 
@@ -268,7 +272,14 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
         acc_dk[dt] = MFMA16(dsa, bfrag, acc_dk[dt]);
       }
     }
-    __syncthreads();  // qt_lds/dot_lds re-staged next iteration
+    // T14: all waves done reading tile `it`'s images -> overwrite with tile
+    // it+1 (regs loaded one iteration ago), issue tile it+2's loads
+    __syncthreads();
+    if (it + 1 < ntiles) {
+      write_stage();
+      if (it + 2 < ntiles) load_stage_regs(q_start + (it + 2) * BLK);
+      __syncthreads();
+    }
   }
 
   // ---- store dK, dV (strided, bf16) ---------------------------------------
@@ -298,11 +309,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dq_sb, int64_t dq_sh, int64_t dq_sl) {
-  // LDS: K^T double-buffered [2][64][72] (T14 split staging, one barrier
-  // per kv tile); per-wave dS tile 4x[16][72].
+  // LDS: per kv tile a TRANSPOSED K image (B-operand of the dQ mfma) plus
+  // ROW-major K and V images (B-operands of S and dP — read from LDS, not
+  // re-fetched from global); T14 split staging, single buffer, two barriers
+  // per kv tile; per-wave dS tile 4x[16][72].
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* kt_lds0 = reinterpret_cast<short*>(smem);
-  short* ds_lds = kt_lds0 + 2 * D * PITCH;
+  short* kt_lds = reinterpret_cast<short*>(smem);     // K^T [64 d][72]
+  short* kr_lds = kt_lds + D * PITCH;                 // K   [64 key][72]
+  short* vr_lds = kr_lds + BLK * PITCH;               // V   [64 key][72]
+  short* ds_lds = vr_lds + BLK * PITCH;
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -346,52 +361,52 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
-  bf16x8_t kreg[2];
+  bf16x8_t kreg[2], vreg[2];
   bool st_valid;
   auto load_stage_regs = [&](int kv0) {
     const int key = kv0 + st_row;
     st_valid = key < Lk;
     const int kr = min(key, Lk - 1);
 #pragma unroll
-    for (int hh = 0; hh < 2; ++hh)
+    for (int hh = 0; hh < 2; ++hh) {
       kreg[hh] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)kr * k_sl + st_c0 + hh * 8);
+      vreg[hh] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)kr * v_sl + st_c0 + hh * 8);
+    }
   };
-  auto write_stage = [&](int buf) {
-    short* kt_lds = kt_lds0 + buf * D * PITCH;
+  auto write_stage = [&]() {
     if (st_valid) {
 #pragma unroll
-      for (int hh = 0; hh < 2; ++hh)
+      for (int hh = 0; hh < 2; ++hh) {
+        *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + st_c0 + hh * 8) = kreg[hh];
+        *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + st_c0 + hh * 8) = vreg[hh];
 #pragma unroll
         for (int i = 0; i < 8; ++i) kt_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = kreg[hh][i];
+      }
     } else {
-      for (int i = 0; i < 16; ++i) kt_lds[(st_c0 + i) * PITCH + st_row] = 0;
+      for (int i = 0; i < 16; ++i) {
+        kr_lds[st_row * PITCH + st_c0 + i] = 0;
+        vr_lds[st_row * PITCH + st_c0 + i] = 0;
+        kt_lds[(st_c0 + i) * PITCH + st_row] = 0;
+      }
     }
   };
 
   load_stage_regs(0);
-  write_stage(0);
+  write_stage();
   if (ntiles > 1) load_stage_regs(BLK);
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * BLK;
-    const int buf = it & 1;
-    short* kt_lds = kt_lds0 + buf * D * PITCH;
-    if (it + 1 < ntiles) {
-      write_stage(buf ^ 1);
-      if (it + 2 < ntiles) load_stage_regs(kv0 + 2 * BLK);
-    }
-
-    // ---- B-fragments of K^T and V^T from global --------------------------
+    // ---- B-fragments of K^T and V^T from the LDS row images --------------
     // B[k=d][j=key] element = K[kv0 + 16*kt + lo][32*s + hi*8 + jj]
     bf16x8_t kb[4][2], vb[4][2];
 #pragma unroll
     for (int kt = 0; kt < 4; ++kt) {
-      const int key = min(kv0 + 16 * kt + lo, Lk - 1);  // clamped; masked below
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + 32 * s + hi * 8);
-        vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + 32 * s + hi * 8);
+        kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+        vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
       }
     }
 
@@ -430,7 +445,14 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         acc_dq[dt] = MFMA16(dsa, bfrag, acc_dq[dt]);
       }
     }
-    __syncthreads();  // kt_lds re-staged next iteration
+    // T14: overwrite the single-buffer images with tile it+1 after everyone
+    // is done reading, then issue tile it+2's loads
+    __syncthreads();
+    if (it + 1 < ntiles) {
+      write_stage();
+      if (it + 2 < ntiles) load_stage_regs(kv0 + 2 * BLK);
+      __syncthreads();
+    }
   }
 
   // ---- store dQ (strided, bf16) -------------------------------------------
@@ -472,7 +494,7 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   }
 
   const size_t shmem_dkv = (6 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
-  const size_t shmem_dq = (2 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const size_t shmem_dq = (3 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
   const dim3 grid_dkv((Lk + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
   const dim3 grid_dq((Lq + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
 
