@@ -34,7 +34,8 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
-constexpr int QBLK = 64;    // q rows per workgroup (16 per wave)
+constexpr int QBLK = 64;    // q rows per strip (16 per wave)
+constexpr int NSTRIP = 4;   // q strips per workgroup (VGPR-bounded)
 constexpr int KVBLK = 64;   // keys per LDS tile
 constexpr int D = 64;       // head_dim (checked host-side)
 constexpr int LDS_PITCH = D + 8;  // +8 shorts: bank-conflict pad for b128 reads
@@ -48,14 +49,16 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   // (sb, sh, sl) = element strides of the (B,H,L,64) view; innermost dim is
   // contiguous. Covers contiguous BHLD, (B,L,H,64) and fused-qkv (B,L,3,H,64)
   // layouts with no permute copies.
-  // LDS carve (double-buffered K/V^T, T14 split staging — issue next tile's
-  // global loads before this tile's compute, ds_write after the barrier;
-  // ONE barrier per KV tile): K [2][64][72], V^T [2][64][72],
-  // P per-wave 4x[16][72] (shorts)
+  //
+  // Each workgroup covers up to NSTRIP x 64 q rows of one (b,h): K/V tiles
+  // are staged ONCE per kv tile for all strips (the 1-strip version staged
+  // them once per 64-q workgroup — 4x redundant global traffic at L=197,
+  // ~125 of its 213 us bandwidth-bound). T14 split staging + single LDS
+  // buffer; each wave keeps NSTRIP online-softmax states.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds0 = reinterpret_cast<short*>(smem);                      // 2*64*72
-  short* vt_lds0 = k_lds0 + 2 * KVBLK * LDS_PITCH;                     // 2*64*72
-  short* p_lds = vt_lds0 + 2 * D * LDS_PITCH;                          // 4*16*72
+  short* k_lds = reinterpret_cast<short*>(smem);                       // 64*72
+  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // 64*72
+  short* p_lds = vt_lds + D * LDS_PITCH;                               // 4*16*72
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -64,180 +67,188 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int hi = lane >> 4;   // 0..3
   const int64_t bh = blockIdx.y;
   const int64_t b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QBLK + wave * 16;  // this wave's first q row
+  const int q_base = blockIdx.x * (NSTRIP * QBLK);
 
   const bf16* qp = q + b * q_sb + h * q_sh;
   const bf16* kp = k + b * k_sb + h * k_sh;
   const bf16* vp = v + b * v_sb + h * v_sh;
   bf16* op = o + b * o_sb + h * o_sh;
 
-  // ---- load Q fragments (B-operand of the swapped QK^T) -------------------
-  // B[d][q] = Q[q0+lo][d = 32*s + hi*8 + j]
-  bf16x8_t qb[2];
-  {
-    const int qrow = min(q0 + lo, Lq - 1);
+  // ---- load Q fragments for every strip (B-operand of the swapped QK^T) --
+  // strip t covers q rows [q_base + t*64, +64); this wave's rows:
+  // q0(t) = q_base + t*64 + wave*16; B[d][q] = Q[q0+lo][32*s + hi*8 + j]
+  bf16x8_t qb[NSTRIP][2];
+#pragma unroll
+  for (int t = 0; t < NSTRIP; ++t) {
+    const int qrow = min(q_base + t * QBLK + wave * 16 + lo, Lq - 1);
 #pragma unroll
     for (int s = 0; s < 2; ++s)
-      qb[s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
+      qb[t][s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
   }
 
   short* my_p = p_lds + wave * 16 * LDS_PITCH;
 
-  f32x4_t acc_o[4] = {};  // O tile: rows q = hi*4+r, cols d = 16*dt + lo
-  float m_run = -INFINITY;
-  float l_run = 0.f;
+  f32x4_t acc_o[NSTRIP][4] = {};  // O tiles: rows q = hi*4+r, cols d = 16*dt+lo
+  float m_run[NSTRIP], l_run[NSTRIP];
+#pragma unroll
+  for (int t = 0; t < NSTRIP; ++t) {
+    m_run[t] = -INFINITY;
+    l_run[t] = 0.f;
+  }
 
-  const int kv_end = CAUSAL ? min(Lk, blockIdx.x * QBLK + QBLK) : Lk;
+  const int kv_end = CAUSAL ? min(Lk, q_base + NSTRIP * QBLK) : Lk;
 
-  // this thread's staging slot: row = key within tile, 16-short chunk
+  // T14 staging state
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
-
   bf16x8_t kreg[2], vreg[2];
-  bool reg_valid;
+  bool st_valid;
   auto load_tile_regs = [&](int kv0) {
     const int key = kv0 + st_row;
-    reg_valid = key < Lk;
+    st_valid = key < Lk;
     const int krow = min(key, Lk - 1);
     kreg[0] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)krow * k_sl + st_c0);
     kreg[1] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)krow * k_sl + st_c0 + 8);
     vreg[0] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)krow * v_sl + st_c0);
     vreg[1] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)krow * v_sl + st_c0 + 8);
   };
-  auto write_tile = [&](int buf) {
-    short* kd = k_lds0 + buf * KVBLK * LDS_PITCH;
-    short* vd = vt_lds0 + buf * D * LDS_PITCH;
-    if (reg_valid) {
-      *reinterpret_cast<bf16x8_t*>(kd + st_row * LDS_PITCH + st_c0) = kreg[0];
-      *reinterpret_cast<bf16x8_t*>(kd + st_row * LDS_PITCH + st_c0 + 8) = kreg[1];
+  auto write_tile = [&]() {
+    if (st_valid) {
+      *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + st_c0) = kreg[0];
+      *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + st_c0 + 8) = kreg[1];
 #pragma unroll
-      for (int i = 0; i < 8; ++i) vd[(st_c0 + i) * LDS_PITCH + st_row] = vreg[0][i];
+      for (int i = 0; i < 8; ++i) vt_lds[(st_c0 + i) * LDS_PITCH + st_row] = vreg[0][i];
 #pragma unroll
-      for (int i = 0; i < 8; ++i) vd[(st_c0 + 8 + i) * LDS_PITCH + st_row] = vreg[1][i];
+      for (int i = 0; i < 8; ++i) vt_lds[(st_c0 + 8 + i) * LDS_PITCH + st_row] = vreg[1][i];
     } else {
-      // zero-fill so mfma on the padded tail contributes nothing
-      for (int i = 0; i < 16; ++i) kd[st_row * LDS_PITCH + st_c0 + i] = 0;
-      for (int i = 0; i < 16; ++i) vd[(st_c0 + i) * LDS_PITCH + st_row] = 0;
+      for (int i = 0; i < 16; ++i) k_lds[st_row * LDS_PITCH + st_c0 + i] = 0;
+      for (int i = 0; i < 16; ++i) vt_lds[(st_c0 + i) * LDS_PITCH + st_row] = 0;
     }
   };
 
   const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
   load_tile_regs(0);
-  write_tile(0);
+  write_tile();
   if (ntiles > 1) load_tile_regs(KVBLK);
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * KVBLK;
-    const int buf = it & 1;
-    short* k_lds = k_lds0 + buf * KVBLK * LDS_PITCH;
-    short* vt_lds = vt_lds0 + buf * D * LDS_PITCH;
-    // T14 write-after-barrier: tile it+1 (regs loaded last iteration) goes
-    // into the other buffer, then tile it+2's loads are issued immediately
-    if (it + 1 < ntiles) {
-      write_tile(buf ^ 1);
-      if (it + 2 < ntiles) load_tile_regs(kv0 + 2 * KVBLK);
-    }
-
-    // ---- S^T = K . Q^T : 4 key tiles x 2 d-steps -------------------------
-    __builtin_amdgcn_s_setprio(1);
-    f32x4_t sc[4] = {};
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
+    for (int st = 0; st < NSTRIP; ++st) {
+      const int q0 = q_base + st * QBLK + wave * 16;
+      if (q0 >= Lq) continue;
+      // ---- S^T = K . Q^T : 4 key tiles x 2 d-steps ------------------------
+      __builtin_amdgcn_s_setprio(1);
+      f32x4_t sc[4] = {};
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          const bf16x8_t ka =
+              *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * t + lo) * LDS_PITCH + 32 * s + hi * 8);
+          sc[t] = MFMA16(ka, qb[st][s], sc[t]);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- masked, scaled scores; per-row (q = lo) online softmax --------
+      float sv[16];
+      const int q_idx = q0 + lo;
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = kv0 + 16 * t + hi * 4 + r;
+          float x = sc[t][r] * scale;
+          if (key >= Lk || (CAUSAL && key > q_idx)) x = -INFINITY;
+          sv[4 * t + r] = x;
+        }
+      }
+      float mt = sv[0];
+#pragma unroll
+      for (int i = 1; i < 16; ++i) mt = fmaxf(mt, sv[i]);
+      mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE));
+
+      const float m_new = fmaxf(m_run[st], mt);
+      // alpha=0 on the first tile (m_run = -inf) starts O from zero
+      const float alpha = (m_new == -INFINITY) ? 0.f : __expf(m_run[st] - m_new);
+      m_run[st] = m_new;
+
+      float psum = 0.f;
+#pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const float p = (sv[i] == -INFINITY) ? 0.f : __expf(sv[i] - m_new);
+        sv[i] = p;
+        psum += p;
+      }
+      psum += __shfl_xor(psum, 16, WAVE);
+      psum += __shfl_xor(psum, 32, WAVE);
+      l_run[st] = l_run[st] * alpha + psum;
+
+      // ---- write P (bf16) into this wave's LDS tile: [q=lo][k=16t+4hi+r] -
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        bf16x4 pk;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) pk[r] = f2bfs(sv[4 * t + r]);
+        *reinterpret_cast<bf16x4*>(my_p + lo * LDS_PITCH + 16 * t + hi * 4) = pk;
+      }
+
+      // ---- rescale O accumulators (rows q = hi*4+r need alpha of lane q) -
+      float alpha_r[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) alpha_r[r] = __shfl(alpha, hi * 4 + r, WAVE);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) acc_o[st][dt][r] *= alpha_r[r];
+
+      // ---- O += P . V : A = P (from LDS), B = V^T reads ------------------
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        const bf16x8_t ka =
-            *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * t + lo) * LDS_PITCH + 32 * s + hi * 8);
-        sc[t] = MFMA16(ka, qb[s], sc[t]);
+        const bf16x8_t pa =
+            *reinterpret_cast<const bf16x8_t*>(my_p + lo * LDS_PITCH + 32 * s + hi * 8);
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          const bf16x8_t vb =
+              *reinterpret_cast<const bf16x8_t*>(vt_lds + (16 * dt + lo) * LDS_PITCH + 32 * s + hi * 8);
+          acc_o[st][dt] = MFMA16(pa, vb, acc_o[st][dt]);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
-
-    // ---- masked, scaled scores; per-row (q = lo) online softmax ----------
-    float sv[16];
-    const int q_idx = q0 + lo;
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = kv0 + 16 * t + hi * 4 + r;
-        float x = sc[t][r] * scale;
-        if (key >= Lk || (CAUSAL && key > q_idx)) x = -INFINITY;
-        sv[4 * t + r] = x;
-      }
+    // T14: overwrite the K/V tiles with tile it+1 after everyone is done
+    __syncthreads();
+    if (it + 1 < ntiles) {
+      write_tile();
+      if (it + 2 < ntiles) load_tile_regs(kv0 + 2 * KVBLK);
+      __syncthreads();
     }
-    float mt = sv[0];
-#pragma unroll
-    for (int i = 1; i < 16; ++i) mt = fmaxf(mt, sv[i]);
-    mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE));
-    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE));
-
-    const float m_new = fmaxf(m_run, mt);
-    // alpha=0 on the first tile (m_run = -inf) starts O from zero
-    const float alpha = (m_new == -INFINITY) ? 0.f : __expf(m_run - m_new);
-    m_run = m_new;
-
-    float psum = 0.f;
-#pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const float p = (sv[i] == -INFINITY) ? 0.f : __expf(sv[i] - m_new);
-      sv[i] = p;
-      psum += p;
-    }
-    psum += __shfl_xor(psum, 16, WAVE);
-    psum += __shfl_xor(psum, 32, WAVE);
-    l_run = l_run * alpha + psum;
-
-    // ---- write P (bf16) into this wave's LDS tile: [q=lo][k=16t+hi*4+r] --
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      bf16x4 pk;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) pk[r] = f2bfs(sv[4 * t + r]);
-      *reinterpret_cast<bf16x4*>(my_p + lo * LDS_PITCH + 16 * t + hi * 4) = pk;
-    }
-
-    // ---- rescale O accumulators (rows q = hi*4+r need alpha from lane q) -
-    float alpha_r[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) alpha_r[r] = __shfl(alpha, hi * 4 + r, WAVE);
-#pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) acc_o[dt][r] *= alpha_r[r];
-
-    // ---- O += P . V : A = P (from LDS), B = V^T reads --------------------
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const bf16x8_t pa =
-          *reinterpret_cast<const bf16x8_t*>(my_p + lo * LDS_PITCH + 32 * s + hi * 8);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const bf16x8_t vb =
-            *reinterpret_cast<const bf16x8_t*>(vt_lds + (16 * dt + lo) * LDS_PITCH + 32 * s + hi * 8);
-        acc_o[dt] = MFMA16(pa, vb, acc_o[dt]);
-      }
-    }
-    __builtin_amdgcn_s_setprio(0);
-    __syncthreads();  // next tile's ds_writes (into the other buffer) + reuse
   }
 
   // ---- epilogue: O /= l, store O and lse ----------------------------------
-  const float invl = (l_run > 0.f) ? 1.f / l_run : 0.f;
-  float invl_r[4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) invl_r[r] = __shfl(invl, hi * 4 + r, WAVE);
+  for (int st = 0; st < NSTRIP; ++st) {
+    const int q0 = q_base + st * QBLK + wave * 16;
+    if (q0 >= Lq) continue;
+    const float invl = (l_run[st] > 0.f) ? 1.f / l_run[st] : 0.f;
+    float invl_r[4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qrow = q0 + hi * 4 + r;
-    if (qrow >= Lq) continue;
+    for (int r = 0; r < 4; ++r) invl_r[r] = __shfl(invl, hi * 4 + r, WAVE);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
-      op[(int64_t)qrow * o_sl + 16 * dt + lo] = f2bf(acc_o[dt][r] * invl_r[r]);
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + hi * 4 + r;
+      if (qrow >= Lq) continue;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+        op[(int64_t)qrow * o_sl + 16 * dt + lo] = f2bf(acc_o[st][dt][r] * invl_r[r]);
+    }
+    if (hi == 0 && q0 + lo < Lq)
+      lse[bh * Lq + q0 + lo] = m_run[st] + __logf(l_run[st]);
   }
-  if (hi == 0 && q0 + lo < Lq && wave * 16 + lo < QBLK)
-    lse[bh * Lq + q0 + lo] = m_run + __logf(l_run);
 }
 
 }  // namespace
@@ -259,7 +270,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto o_storage = torch::empty({B, Lq, H, 64}, q.options());
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
-  const dim3 grid((Lq + QBLK - 1) / QBLK, (unsigned)((int64_t)B * H));
+  const dim3 grid((Lq + NSTRIP * QBLK - 1) / (NSTRIP * QBLK), (unsigned)((int64_t)B * H));
   const size_t shmem = (2 * KVBLK * LDS_PITCH + 2 * D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
 #define ATTN_LAUNCH(C)                                                                     \
