@@ -67,7 +67,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int hi = lane >> 4;   // 0..3
   const int64_t bh = blockIdx.y;
   const int64_t b = bh / H, h = bh % H;
-  const int q_base = blockIdx.x * (NSTRIP * QBLK);
+  // cyclic strip -> q-tile mapping: strip t covers q tile
+  // (blockIdx.x + t*gridDim.x) — balances ragged L (a blocked mapping gives
+  // the last workgroup a nearly-empty supertile, e.g. 1 of 256 rows at L=257)
+  const int ntq = (Lq + QBLK - 1) / QBLK;
 
   const bf16* qp = q + b * q_sb + h * q_sh;
   const bf16* kp = k + b * k_sb + h * k_sh;
@@ -78,9 +81,12 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   // strip t covers q rows [q_base + t*64, +64); this wave's rows:
   // q0(t) = q_base + t*64 + wave*16; B[d][q] = Q[q0+lo][32*s + hi*8 + j]
   bf16x8_t qb[NSTRIP][2];
+  int nactive = 0;
 #pragma unroll
   for (int t = 0; t < NSTRIP; ++t) {
-    const int qrow = min(q_base + t * QBLK + wave * 16 + lo, Lq - 1);
+    const int ti = blockIdx.x + t * gridDim.x;
+    if (ti < ntq) nactive = t + 1;
+    const int qrow = min(min(ti, ntq - 1) * QBLK + wave * 16 + lo, Lq - 1);
 #pragma unroll
     for (int s = 0; s < 2; ++s)
       qb[t][s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
@@ -96,7 +102,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     l_run[t] = 0.f;
   }
 
-  const int kv_end = CAUSAL ? min(Lk, q_base + NSTRIP * QBLK) : Lk;
+  const int ti_max = blockIdx.x + (nactive - 1) * gridDim.x;  // last active tile
+  const int kv_end = CAUSAL ? min(Lk, (ti_max + 1) * QBLK) : Lk;
 
   // T14 staging state
   const int st_row = tid / 4;
@@ -136,8 +143,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const int kv0 = it * KVBLK;
 #pragma unroll
     for (int st = 0; st < NSTRIP; ++st) {
-      const int q0 = q_base + st * QBLK + wave * 16;
-      if (q0 >= Lq) continue;
+      const int ti = blockIdx.x + st * gridDim.x;
+      const int q0 = ti * QBLK + wave * 16;
+      if (ti >= ntq || q0 >= Lq) continue;
+      if (CAUSAL && kv0 >= (ti + 1) * QBLK) continue;  // fully masked strip
       // ---- S^T = K . Q^T : 4 key tiles x 2 d-steps ------------------------
       __builtin_amdgcn_s_setprio(1);
       f32x4_t sc[4] = {};
@@ -232,8 +241,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   // ---- epilogue: O /= l, store O and lse ----------------------------------
 #pragma unroll
   for (int st = 0; st < NSTRIP; ++st) {
-    const int q0 = q_base + st * QBLK + wave * 16;
-    if (q0 >= Lq) continue;
+    const int ti = blockIdx.x + st * gridDim.x;
+    const int q0 = ti * QBLK + wave * 16;
+    if (ti >= ntq || q0 >= Lq) continue;
     const float invl = (l_run[st] > 0.f) ? 1.f / l_run[st] : 0.f;
     float invl_r[4];
 #pragma unroll
@@ -270,7 +280,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto o_storage = torch::empty({B, Lq, H, 64}, q.options());
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
-  const dim3 grid((Lq + NSTRIP * QBLK - 1) / (NSTRIP * QBLK), (unsigned)((int64_t)B * H));
+  const int ntq = (Lq + QBLK - 1) / QBLK;
+  const dim3 grid((ntq + NSTRIP - 1) / NSTRIP, (unsigned)((int64_t)B * H));
   const size_t shmem = (2 * KVBLK * LDS_PITCH + 2 * D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
 #define ATTN_LAUNCH(C)                                                                     \
