@@ -25,6 +25,7 @@ def _parse_config(cfg: dict) -> dict:
         vision_width=vc["hidden_size"],
         vision_patch_size=vc.get("patch_size", 16),
         vision_mlp_dim=vc.get("intermediate_size"),
+        vision_heads=vc.get("num_attention_heads"),
         context_length=tc.get("max_position_embeddings", 64),
         vocab_size=tc.get("vocab_size", 32000),
         transformer_width=tc["hidden_size"],
@@ -37,7 +38,9 @@ def _parse_config(cfg: dict) -> dict:
 def _infer_config(sd: dict[str, torch.Tensor], image_size: int | None = None) -> dict:
     """Shape inference (reference siglip.py:193-207 — config only for image_size)."""
     conv_w = sd["vision_model.embeddings.patch_embedding.weight"]
-    vision_width, patch = conv_w.shape[0], conv_w.shape[-1]
+    vision_width = conv_w.shape[0]
+    # v1: conv (H, 3, P, P); v2: linear (H, P*P*3) over channels-last patches
+    patch = conv_w.shape[-1] if conv_w.dim() == 4 else int(math.isqrt(conv_w.shape[1] // 3))
     v_pos = sd["vision_model.embeddings.position_embedding.weight"].shape[0]
     img = image_size or int(math.isqrt(v_pos)) * patch
     n_layers = lambda pref: 1 + max(int(k.split(".")[3]) for k in sd if k.startswith(pref + ".encoder.layers."))
@@ -63,8 +66,16 @@ def map_siglip(sd: dict[str, torch.Tensor], vision_layers: int, text_layers: int
     m.put("logit_bias", m.take("logit_bias").reshape(()))
     m.copy("text_projection.weight", "text_model.head.weight")
     m.copy("text_projection.bias", "text_model.head.bias")
-    # vision tower (no CLS token, no pre-norm)
-    m.copy("vision_model.patch_weight", "vision_model.embeddings.patch_embedding.weight")
+    # vision tower (no CLS token, no pre-norm). SigLIP2 stores the patch
+    # embedding as a Linear over channels-LAST flattened patches
+    # (H, P*P*C) — transformers' Siglip2 image processor packs each patch
+    # as image[py, px, c]; reshape to the conv layout (H, C, P, P).
+    pw = m.take("vision_model.embeddings.patch_embedding.weight")
+    if pw.dim() == 2:
+        H_, feat = pw.shape
+        P = int(math.isqrt(feat // 3))
+        pw = pw.view(H_, P, P, 3).permute(0, 3, 1, 2).contiguous()
+    m.put("vision_model.patch_weight", pw)
     m.copy("vision_model.patch_bias", "vision_model.embeddings.patch_embedding.bias")
     m.put("vision_model.pos_embedding", m.take("vision_model.embeddings.position_embedding.weight").unsqueeze(0))
     m.copy("vision_model.ln_post.weight", "vision_model.post_layernorm.weight")

@@ -172,3 +172,43 @@ def test_vit_shape_inference(tmp_path):
     assert cfg["img_size"] == 32
     assert cfg["mlp_dim"] == 64
     assert cfg["num_classes"] == 3
+
+
+def test_siglip2_vision_parity(tmp_path):
+    """SigLIP2 fixed-resolution checkpoints load through the same SigLIP
+    class (reference supports 'SigLIP v1/v2', README.md:6-15): the linear
+    channels-last patch embedding is reshaped to the conv layout."""
+    from transformers import Siglip2Config, Siglip2Model, Siglip2TextConfig, Siglip2VisionConfig
+
+    vc = Siglip2VisionConfig(hidden_size=32, num_hidden_layers=2, num_attention_heads=2,
+                             intermediate_size=64, image_size=32, patch_size=16, num_patches=4)
+    tc = Siglip2TextConfig(hidden_size=32, num_hidden_layers=2, num_attention_heads=2,
+                           intermediate_size=64, vocab_size=99, max_position_embeddings=8,
+                           bos_token_id=None, eos_token_id=None)
+    hf = Siglip2Model(Siglip2Config(vision_config=vc.to_dict(), text_config=tc.to_dict())).eval()
+    for p in hf.parameters():
+        p.data.normal_(0, 0.02)
+    d = tmp_path / "siglip2"
+    hf.save_pretrained(d, safe_serialization=True)
+
+    model = jimm_amd.SigLIP.from_pretrained(str(d)).eval()
+    torch.manual_seed(0)
+    img = torch.randn(2, 3, 32, 32)
+
+    # oracle: pack the image into channels-last patches the Siglip2 way
+    P = 16
+    patches = img.permute(0, 2, 3, 1)                       # (B, H, W, C)
+    patches = patches.reshape(2, 2, P, 2, P, 3).permute(0, 1, 3, 2, 4, 5).reshape(2, 4, P * P * 3)
+    spatial = torch.tensor([[2, 2], [2, 2]])
+    mask = torch.ones(2, 4, dtype=torch.long)
+    with torch.no_grad():
+        ref = hf.vision_model(pixel_values=patches, pixel_attention_mask=mask,
+                              spatial_shapes=spatial).pooler_output
+        out = model.encode_image(img)
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+
+    ids = torch.randint(0, 99, (2, 8))
+    with torch.no_grad():
+        ref_t = hf.text_model(input_ids=ids).pooler_output
+        out_t = model.encode_text(ids)
+    assert torch.allclose(out_t, ref_t, atol=1e-4), (out_t - ref_t).abs().max()
