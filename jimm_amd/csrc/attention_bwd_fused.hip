@@ -38,6 +38,7 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
 constexpr int BLK = 64;            // q rows / keys per workgroup tile
+constexpr int NKV = 2;             // kv tiles per dkv workgroup (VGPR-bounded; 4 spills)
 constexpr int D = 64;              // head_dim (checked host-side)
 constexpr int PITCH = D + 8;       // bank-conflict pad for b128 reads
 
@@ -82,28 +83,24 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
 // ---------------------------------------------------------------------------
 
 template <bool CAUSAL>
-__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int Lq, int Lk, float scale, int H,
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dk_sb, int64_t dk_sh, int64_t dk_sl, int64_t dv_sb, int64_t dv_sh, int64_t dv_sl) {
-  // LDS: K [64][72], V [64][72] staged once (row-major); per q tile both a
-  // TRANSPOSED image (Q^T, dO^T — B-operands of the dK/dV mfma) and a
-  // ROW-major image (B-operands of the S^T/dP^T mfma) of Q and dO, all four
-  // written from the SAME staging registers (the r06 profile showed this
-  // kernel bandwidth-bound on reading Q/dO twice from global: once staged,
-  // once as global B-fragments — 3.1 GB/layer). T14 split staging, single
-  // buffer, two barriers per q tile. Per-wave P/dS tile 4x[16][72].
-  // Total ~64 KiB.
+  // v3: each workgroup owns up to NKV kv tiles (cyclic tile mapping, like
+  // the forward kernel) so the Q/dO images are staged ONCE for all of them
+  // (v2 staged them once per kv-tile workgroup: 4x redundant at L=197).
+  // K/V live as per-strip A-FRAGMENTS in registers — no K/V LDS at all.
+  // LDS: Q^T/dO^T (transposed) + Q/dO (row) images of the current q tile,
+  // per-wave P/dS tile. ~46 KiB.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds = reinterpret_cast<short*>(smem);
-  short* v_lds = k_lds + BLK * PITCH;
-  short* qt_lds = v_lds + BLK * PITCH;       // Q^T  [64 d][72]
-  short* dot_lds = qt_lds + D * PITCH;       // dO^T [64 d][72]
-  short* qr_lds = dot_lds + D * PITCH;       // Q    [64 q][72]
-  short* dor_lds = qr_lds + BLK * PITCH;     // dO   [64 q][72]
+  short* qt_lds = reinterpret_cast<short*>(smem);   // Q^T  [64 d][72]
+  short* dot_lds = qt_lds + D * PITCH;              // dO^T [64 d][72]
+  short* qr_lds = dot_lds + D * PITCH;              // Q    [64 q][72]
+  short* dor_lds = qr_lds + BLK * PITCH;            // dO   [64 q][72]
   short* p_lds = dor_lds + BLK * PITCH;
 
   const int tid = threadIdx.x;
@@ -112,7 +109,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   const int lo = lane & 15, hi = lane >> 4;
   const int64_t bh = blockIdx.y;
   const int64_t b = bh / H, h = bh % H;
-  const int kv0 = blockIdx.x * BLK;
+  const int ntk = (Lk + BLK - 1) / BLK;
 
   const bf16* qp = q + b * q_sb + h * q_sh;
   const bf16* kp = k + b * k_sb + h * k_sh;
@@ -121,44 +118,36 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   const float* lsep = lse + bh * Lq;
   const float* dvp_row = Dv + bh * Lq;
 
-  // ---- stage K and V tiles once (row-major, zero-padded tail) -------------
-  {
-    const int row = tid / 4;
-    const int c0 = (tid % 4) * 16;
-    const int key = kv0 + row;
-    if (key < Lk) {
+  // ---- per-strip K/V A-fragments straight from global (no LDS) ------------
+  // strip s covers kv tile ti(s) = blockIdx.x + s*gridDim.x; this wave's 16
+  // keys of that tile: rows 16*wave + lo. Invalid rows clamp; their P/dS
+  // contributions are masked to zero and their stores are guarded.
+  bf16x8_t ka[NKV][2], va[NKV][2];
+  int kvbase[NKV];
+  int nactive = 0;
 #pragma unroll
-      for (int hh = 0; hh < 2; ++hh) {
-        *reinterpret_cast<bf16x8_t*>(k_lds + row * PITCH + c0 + hh * 8) =
-            *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + c0 + hh * 8);
-        *reinterpret_cast<bf16x8_t*>(v_lds + row * PITCH + c0 + hh * 8) =
-            *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + c0 + hh * 8);
-      }
-    } else {
-      for (int i = 0; i < 16; ++i) k_lds[row * PITCH + c0 + i] = 0;
-      for (int i = 0; i < 16; ++i) v_lds[row * PITCH + c0 + i] = 0;
+  for (int s = 0; s < NKV; ++s) {
+    const int ti = blockIdx.x + s * gridDim.x;
+    kvbase[s] = ti * BLK;
+    if (ti < ntk) nactive = s + 1;
+    const int key = min(min(ti, ntk - 1) * BLK + 16 * wave + lo, Lk - 1);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      ka[s][t] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + 32 * t + hi * 8);
+      va[s][t] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + 32 * t + hi * 8);
     }
   }
-  __syncthreads();
-
-  // A-fragments of K and V for this wave's 16 keys (fixed for whole kernel)
-  bf16x8_t ka[2], va[2];
-#pragma unroll
-  for (int s = 0; s < 2; ++s) {
-    ka[s] = *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * wave + lo) * PITCH + 32 * s + hi * 8);
-    va[s] = *reinterpret_cast<const bf16x8_t*>(v_lds + (16 * wave + lo) * PITCH + 32 * s + hi * 8);
-  }
-  __syncthreads();  // qt/dot staging below reuses nothing, but keep tiles safe
 
   short* my_p = p_lds + wave * 16 * PITCH;
 
-  f32x4_t acc_dk[4] = {};  // rows key = 16*wave + hi*4+r, cols d = 16*dt + lo
-  f32x4_t acc_dv[4] = {};
+  f32x4_t acc_dk[NKV][4] = {};  // rows key = 16*wave + hi*4+r, cols d = 16*dt+lo
+  f32x4_t acc_dv[NKV][4] = {};
 
-  const int q_start = CAUSAL ? (kv0 / BLK) * BLK : 0;
+  // causal: the earliest kv tile of this WG bounds the first useful q tile
+  const int q_start = CAUSAL ? (kvbase[0] / BLK) * BLK : 0;
   const int ntiles = (Lq - q_start + BLK - 1) / BLK;
 
-  // per-thread staging slot for the scatter-transpose
+  // per-thread staging slot for the Q/dO images
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
   bf16x8_t qreg[2], doreg[2];
@@ -201,8 +190,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
 
   for (int it = 0; it < ntiles; ++it) {
     const int q0 = q_start + it * BLK;
-    // ---- B-fragments of Q^T and dO^T from the LDS row images --------------
-    // B[k=d][j=q] element = Q[q0 + 16*qt + lo][32*s + hi*8 + jj]
+    // ---- B-fragments of Q^T and dO^T from the LDS row images (shared by
+    // every kv strip) ------------------------------------------------------
     bf16x8_t qb[4][2], dob[4][2];
 #pragma unroll
     for (int qt = 0; qt < 4; ++qt) {
@@ -213,67 +202,76 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       }
     }
 
-    // ---- S^T = K . Q^T ; P^T = exp(scale*S^T - lse[q]) -------------------
-    // C rows = key = 16*wave + hi*4+r, cols = q = 16*qt + lo
-    bf16x4 ds_stash[4];
 #pragma unroll
-    for (int qt = 0; qt < 4; ++qt) {
-      f32x4_t sc = {};
-#pragma unroll
-      for (int s = 0; s < 2; ++s) sc = MFMA16(ka[s], qb[qt][s], sc);
-      const int qi = q0 + 16 * qt + lo;
-      const float l = lsep[min(qi, Lq - 1)];
-      f32x4_t dpc = {};
-#pragma unroll
-      for (int s = 0; s < 2; ++s) dpc = MFMA16(va[s], dob[qt][s], dpc);
-      const float dcoef = dvp_row[min(qi, Lq - 1)];
-      bf16x4 pk, dsk;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = kv0 + 16 * wave + hi * 4 + r;
-        float p = __expf(sc[r] * scale - l);
-        if (qi >= Lq || (CAUSAL && key > qi)) p = 0.f;
-        pk[r] = f2bfs(p);
-        dsk[r] = f2bfs(p * (dpc[r] - dcoef) * scale);
-      }
-      // route P^T through the per-wave LDS tile now (A-operand of the dV
-      // mfma); dS^T is stashed in registers and reuses the same tile after
-      // the dV mfma has consumed P^T.
-#pragma unroll
-      for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = pk[r];
-      ds_stash[qt] = dsk;
-    }
+    for (int sidx = 0; sidx < NKV; ++sidx) {
+      if (sidx >= nactive) continue;
+      const int kv0 = kvbase[sidx];
+      if (CAUSAL && q0 + BLK - 1 < kv0) continue;  // whole tile above diagonal
 
-    // ---- dV += P^T . dO  (A = P^T via LDS, B = dO^T rows) ----------------
+      // ---- S^T = K . Q^T ; P^T = exp(scale*S^T - lse[q]) -----------------
+      bf16x4 ds_stash[4];
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
+      for (int qt = 0; qt < 4; ++qt) {
+        __builtin_amdgcn_s_setprio(1);
+        f32x4_t sc = {};
+        f32x4_t dpc = {};
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const bf16x8_t bfrag =
-            *reinterpret_cast<const bf16x8_t*>(dot_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
-        acc_dv[dt] = MFMA16(pa, bfrag, acc_dv[dt]);
+        for (int s = 0; s < 2; ++s) {
+          sc = MFMA16(ka[sidx][s], qb[qt][s], sc);
+          dpc = MFMA16(va[sidx][s], dob[qt][s], dpc);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        const int qi = q0 + 16 * qt + lo;
+        const float l = lsep[min(qi, Lq - 1)];
+        const float dcoef = dvp_row[min(qi, Lq - 1)];
+        bf16x4 pk, dsk;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = kv0 + 16 * wave + hi * 4 + r;
+          float p = __expf(sc[r] * scale - l);
+          if (qi >= Lq || key >= Lk || (CAUSAL && key > qi)) p = 0.f;
+          pk[r] = f2bfs(p);
+          dsk[r] = f2bfs(p * (dpc[r] - dcoef) * scale);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = pk[r];
+        ds_stash[qt] = dsk;
       }
-    }
 
-    // ---- overwrite the wave tile with dS^T, then dK += dS^T . Q ----------
+      // ---- dV += P^T . dO  (A = P^T via LDS, B = dO^T rows) --------------
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int qt = 0; qt < 4; ++qt) {
+      for (int s = 0; s < 2; ++s) {
+        const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
-      for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = ds_stash[qt][r];
-    }
-#pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const bf16x8_t bfrag =
-            *reinterpret_cast<const bf16x8_t*>(qt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
-        acc_dk[dt] = MFMA16(dsa, bfrag, acc_dk[dt]);
+        for (int dt = 0; dt < 4; ++dt) {
+          const bf16x8_t bfrag =
+              *reinterpret_cast<const bf16x8_t*>(dot_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
+          acc_dv[sidx][dt] = MFMA16(pa, bfrag, acc_dv[sidx][dt]);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- overwrite the wave tile with dS^T, then dK += dS^T . Q --------
+#pragma unroll
+      for (int qt = 0; qt < 4; ++qt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = ds_stash[qt][r];
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          const bf16x8_t bfrag =
+              *reinterpret_cast<const bf16x8_t*>(qt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
+          acc_dk[sidx][dt] = MFMA16(dsa, bfrag, acc_dk[sidx][dt]);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    // T14: all waves done reading tile `it`'s images -> overwrite with tile
-    // it+1 (regs loaded one iteration ago), issue tile it+2's loads
+    // T14: all waves done reading this q tile -> overwrite with tile it+1
     __syncthreads();
     if (it + 1 < ntiles) {
       write_stage();
@@ -286,13 +284,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   bf16* dkp = dk + b * dk_sb + h * dk_sh;
   bf16* dvp = dv + b * dv_sb + h * dv_sh;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int key = kv0 + 16 * wave + hi * 4 + r;
-    if (key >= Lk) continue;
+  for (int sidx = 0; sidx < NKV; ++sidx) {
+    if (sidx >= nactive) continue;
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      dkp[(int64_t)key * dk_sl + 16 * dt + lo] = f2bf(acc_dk[dt][r]);
-      dvp[(int64_t)key * dv_sl + 16 * dt + lo] = f2bf(acc_dv[dt][r]);
+    for (int r = 0; r < 4; ++r) {
+      const int key = kvbase[sidx] + 16 * wave + hi * 4 + r;
+      if (key >= Lk) continue;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        dkp[(int64_t)key * dk_sl + 16 * dt + lo] = f2bf(acc_dk[sidx][dt][r]);
+        dvp[(int64_t)key * dv_sl + 16 * dt + lo] = f2bf(acc_dv[sidx][dt][r]);
+      }
     }
   }
 }
@@ -493,9 +495,10 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
                        o.stride(0), o.stride(1), o.stride(2));
   }
 
-  const size_t shmem_dkv = (6 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const size_t shmem_dkv = (4 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
   const size_t shmem_dq = (3 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
-  const dim3 grid_dkv((Lk + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
+  const int ntk = (Lk + BLK - 1) / BLK;
+  const dim3 grid_dkv((ntk + NKV - 1) / NKV, (unsigned)((int64_t)B * H));
   const dim3 grid_dq((Lq + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
 
 #define DKV_LAUNCH(C)                                                                        \
