@@ -39,6 +39,7 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
 constexpr int BLK = 64;            // q rows / keys per workgroup tile
 constexpr int NKV = 2;             // kv tiles per dkv workgroup (VGPR-bounded; 4 spills)
+constexpr int NQS = 2;             // q tiles per dq workgroup
 constexpr int D = 64;              // head_dim (checked host-side)
 constexpr int PITCH = D + 8;       // bank-conflict pad for b128 reads
 
@@ -304,17 +305,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 // ---------------------------------------------------------------------------
 
 template <bool CAUSAL>
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
     bf16* __restrict__ dq, int Lq, int Lk, float scale, int H,
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dq_sb, int64_t dq_sh, int64_t dq_sl) {
-  // LDS: per kv tile a TRANSPOSED K image (B-operand of the dQ mfma) plus
-  // ROW-major K and V images (B-operands of S and dP — read from LDS, not
-  // re-fetched from global); T14 split staging, single buffer, two barriers
-  // per kv tile; per-wave dS tile 4x[16][72].
+  // v3: each workgroup owns up to NQS q tiles (cyclic mapping) so the K
+  // images are staged ONCE for all of them; Q/dO A-fragments live in
+  // registers per strip. LDS: K^T + K/V row images of the current kv tile,
+  // per-wave dS tile.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* kt_lds = reinterpret_cast<short*>(smem);     // K^T [64 d][72]
   short* kr_lds = kt_lds + D * PITCH;                 // K   [64 key][72]
@@ -327,7 +328,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int lo = lane & 15, hi = lane >> 4;
   const int64_t bh = blockIdx.y;
   const int64_t b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * BLK + wave * 16;  // this wave's first q row
+  const int ntq = (Lq + BLK - 1) / BLK;
 
   const bf16* qp = q + b * q_sb + h * q_sh;
   const bf16* kp = k + b * k_sb + h * k_sh;
@@ -336,29 +337,39 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const float* lsep = lse + bh * Lq;
   const float* dvp_row = Dv + bh * Lq;
 
-  // A-fragments of Q and dO for this wave's 16 q rows (fixed)
-  bf16x8_t qa[2], doa[2];
-  {
+  // ---- per-strip Q/dO A-fragments + per-row lse/D -------------------------
+  // strip s covers q tile ti(s) = blockIdx.x + s*gridDim.x; this wave's 16
+  // q rows of that tile: rows 16*wave + ...
+  bf16x8_t qa[NQS][2], doa[NQS][2];
+  float lse_r[NQS][4], d_r[NQS][4];
+  int qbase[NQS];
+  int nactive = 0;
+#pragma unroll
+  for (int s = 0; s < NQS; ++s) {
+    const int ti = blockIdx.x + s * gridDim.x;
+    qbase[s] = ti * BLK;
+    if (ti < ntq) nactive = s + 1;
+    const int q0 = min(ti, ntq - 1) * BLK + wave * 16;
     const int qrow = min(q0 + lo, Lq - 1);
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      qa[s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
-      doa[s] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qrow * do_sl + 32 * s + hi * 8);
+    for (int t = 0; t < 2; ++t) {
+      qa[s][t] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * t + hi * 8);
+      doa[s][t] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qrow * do_sl + 32 * t + hi * 8);
     }
-  }
-  // per-row lse and D for rows q = q0 + hi*4 + r
-  float lse_r[4], d_r[4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qi = min(q0 + hi * 4 + r, Lq - 1);
-    lse_r[r] = lsep[qi];
-    d_r[r] = dvp_row[qi];
+    for (int r = 0; r < 4; ++r) {
+      const int qi = min(q0 + hi * 4 + r, Lq - 1);
+      lse_r[s][r] = lsep[qi];
+      d_r[s][r] = dvp_row[qi];
+    }
   }
 
   short* my_ds = ds_lds + wave * 16 * PITCH;
-  f32x4_t acc_dq[4] = {};  // rows q = hi*4+r, cols d = 16*dt + lo
+  f32x4_t acc_dq[NQS][4] = {};  // rows q = hi*4+r, cols d = 16*dt + lo
 
-  const int kv_end = CAUSAL ? min(Lk, blockIdx.x * BLK + BLK) : Lk;
+  // causal: only kv tiles up to the LAST active strip's diagonal are needed
+  const int ti_max = blockIdx.x + (nactive - 1) * gridDim.x;
+  const int kv_end = CAUSAL ? min(Lk, (ti_max + 1) * BLK) : Lk;
   const int ntiles = (kv_end + BLK - 1) / BLK;
 
   const int st_row = tid / 4;
@@ -400,8 +411,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * BLK;
-    // ---- B-fragments of K^T and V^T from the LDS row images --------------
-    // B[k=d][j=key] element = K[kv0 + 16*kt + lo][32*s + hi*8 + jj]
+    // ---- B-fragments of K^T and V^T from the LDS row images (shared) -----
     bf16x8_t kb[4][2], vb[4][2];
 #pragma unroll
     for (int kt = 0; kt < 4; ++kt) {
@@ -412,40 +422,50 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       }
     }
 
-    // ---- S, P, dP, dS per 16-key tile ------------------------------------
-    // C rows = q = hi*4+r, cols = key = 16*kt + lo
 #pragma unroll
-    for (int kt = 0; kt < 4; ++kt) {
-      f32x4_t sc = {};
-      f32x4_t dpc = {};
+    for (int sidx = 0; sidx < NQS; ++sidx) {
+      if (sidx >= nactive) continue;
+      const int q0s = qbase[sidx] + wave * 16;
+      if (CAUSAL && kv0 >= qbase[sidx] + BLK) continue;  // tile above diagonal
+
+      // ---- S, P, dP, dS per 16-key tile ----------------------------------
+#pragma unroll
+      for (int kt = 0; kt < 4; ++kt) {
+        __builtin_amdgcn_s_setprio(1);
+        f32x4_t sc = {};
+        f32x4_t dpc = {};
+#pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          sc = MFMA16(qa[sidx][s], kb[kt][s], sc);
+          dpc = MFMA16(doa[sidx][s], vb[kt][s], dpc);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        const int key = kv0 + 16 * kt + lo;
+        bf16x4 dsk;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qi = q0s + hi * 4 + r;
+          float p = __expf(sc[r] * scale - lse_r[sidx][r]);
+          if (key >= Lk || (CAUSAL && key > qi)) p = 0.f;
+          dsk[r] = f2bfs(p * (dpc[r] - d_r[sidx][r]) * scale);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) my_ds[(hi * 4 + r) * PITCH + 16 * kt + lo] = dsk[r];
+      }
+
+      // ---- dQ += dS . K (A = dS via LDS, B = K from K^T rows) ------------
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        sc = MFMA16(qa[s], kb[kt][s], sc);
-        dpc = MFMA16(doa[s], vb[kt][s], dpc);
+        const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_ds + lo * PITCH + 32 * s + hi * 8);
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          const bf16x8_t bfrag =
+              *reinterpret_cast<const bf16x8_t*>(kt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
+          acc_dq[sidx][dt] = MFMA16(dsa, bfrag, acc_dq[sidx][dt]);
+        }
       }
-      const int key = kv0 + 16 * kt + lo;
-      bf16x4 dsk;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qi = q0 + hi * 4 + r;
-        float p = __expf(sc[r] * scale - lse_r[r]);
-        if (key >= Lk || (CAUSAL && key > qi)) p = 0.f;
-        dsk[r] = f2bfs(p * (dpc[r] - d_r[r]) * scale);
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) my_ds[(hi * 4 + r) * PITCH + 16 * kt + lo] = dsk[r];
-    }
-
-    // ---- dQ += dS . K (A = dS via LDS, B = K from K^T rows) --------------
-#pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_ds + lo * PITCH + 32 * s + hi * 8);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const bf16x8_t bfrag =
-            *reinterpret_cast<const bf16x8_t*>(kt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
-        acc_dq[dt] = MFMA16(dsa, bfrag, acc_dq[dt]);
-      }
+      __builtin_amdgcn_s_setprio(0);
     }
     // T14: overwrite the single-buffer images with tile it+1 after everyone
     // is done reading, then issue tile it+2's loads
@@ -460,12 +480,16 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   // ---- store dQ (strided, bf16) -------------------------------------------
   bf16* dqp = dq + b * dq_sb + h * dq_sh;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qi = q0 + hi * 4 + r;
-    if (qi >= Lq) continue;
+  for (int sidx = 0; sidx < NQS; ++sidx) {
+    if (sidx >= nactive) continue;
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
-      dqp[(int64_t)qi * dq_sl + 16 * dt + lo] = f2bf(acc_dq[dt][r]);
+    for (int r = 0; r < 4; ++r) {
+      const int qi = qbase[sidx] + wave * 16 + hi * 4 + r;
+      if (qi >= Lq) continue;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+        dqp[(int64_t)qi * dq_sl + 16 * dt + lo] = f2bf(acc_dq[sidx][dt][r]);
+    }
   }
 }
 
@@ -499,7 +523,8 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   const size_t shmem_dq = (3 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
   const int ntk = (Lk + BLK - 1) / BLK;
   const dim3 grid_dkv((ntk + NKV - 1) / NKV, (unsigned)((int64_t)B * H));
-  const dim3 grid_dq((Lq + BLK - 1) / BLK, (unsigned)((int64_t)B * H));
+  const int ntq = (Lq + BLK - 1) / BLK;
+  const dim3 grid_dq((ntq + NQS - 1) / NQS, (unsigned)((int64_t)B * H));
 
 #define DKV_LAUNCH(C)                                                                        \
   hipLaunchKernelGGL((attn_bwd_dkv_kernel<C>), grid_dkv, dim3(256), shmem_dkv, stream,       \
