@@ -74,7 +74,12 @@ class DataParallelGrads:
         self._reset_counters()
 
     def _make_bucket(self, params: list[torch.nn.Parameter]) -> None:
-        numel = sum(p.numel() for p in params)
+        # offsets aligned to 8 elements so every grad view is at least
+        # 16-B aligned — the fused Adam kernel uses vectorized accesses
+        def aligned(n: int) -> int:
+            return (n + 7) & ~7
+
+        numel = sum(aligned(p.numel()) for p in params)
         p0 = params[0]
         flat = torch.zeros(numel, dtype=p0.dtype, device=p0.device)
         offset = 0
@@ -82,7 +87,7 @@ class DataParallelGrads:
             n = p.numel()
             # p.grad aliases the bucket: backward accumulates in place
             p.grad = flat[offset : offset + n].view_as(p)
-            offset += n
+            offset += aligned(n)
         self.buckets.append({"params": params, "flat": flat, "pending": len(params), "work": None})
 
     def _reset_counters(self) -> None:
