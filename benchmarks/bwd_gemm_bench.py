@@ -64,7 +64,8 @@ def main():
             tun.set_filename(args.tune)  # auto-written at process exit
             tun.enable(True)
             tun.tuning_enable(True)
-            tun.set_max_tuning_duration(100)
+            tun.set_max_tuning_duration(int(os.environ.get("TUNE_MS", "100")))
+            tun.set_max_tuning_iterations(int(os.environ.get("TUNE_ITERS", "100")))
         else:
             tun.enable(True)
             tun.tuning_enable(False)
