@@ -518,3 +518,29 @@ def test_fused_block_vs_composite():
     assert rel_err(dx_f, dx_c) < 2e-2, rel_err(dx_f, dx_c)
     for n in g_f:
         assert rel_err(g_f[n], g_c[n]) < 3e-2, (n, rel_err(g_f[n], g_c[n]))
+
+
+@pytest.mark.parametrize("which", ["clip", "siglip"])
+def test_dual_tower_gpu_vs_cpu_parity(which):
+    """CLIP/SigLIP logits: GPU bf16 HIP path vs CPU fp32 reference path
+    (covers causal text attention, EOT/last pooling and the MAP head)."""
+    torch.manual_seed(0)
+    if which == "clip":
+        model = jimm_amd.CLIP(embed_dim=64, image_resolution=64, vision_layers=2,
+                              vision_width=128, vision_patch_size=32, context_length=12,
+                              vocab_size=99, transformer_width=64, transformer_heads=1,
+                              transformer_layers=2).eval()
+    else:
+        model = jimm_amd.SigLIP(image_resolution=64, vision_layers=2, vision_width=128,
+                                vision_patch_size=16, context_length=12, vocab_size=99,
+                                transformer_width=128, transformer_heads=2,
+                                transformer_layers=2).eval()
+    imgs = torch.randn(3, 3, 64, 64)
+    ids = torch.randint(0, 98, (3, 12))
+    ids[:, -1] = 98  # EOT = max id for CLIP pooling
+    with torch.no_grad():
+        ref, _ = model(imgs, ids)
+        gm = model.to(dev(), torch.bfloat16)
+        out, _ = gm(imgs.to(dev(), torch.bfloat16), ids.to(dev()))
+    err = (out.float().cpu() - ref).abs().max().item()
+    assert err < 0.25, err  # bf16 end-to-end vs fp32; logits are O(1-10)
