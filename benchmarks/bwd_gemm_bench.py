@@ -72,7 +72,9 @@ def main():
             assert tun.read_file(args.load), f"failed to read {args.load}"
 
     dev = torch.device("cuda:0")
-    for M, N, K in SHAPES:
+    skip = int(os.environ.get("TUNE_SKIP", "0"))
+    only = int(os.environ.get("TUNE_COUNT", str(len(SHAPES))))
+    for M, N, K in SHAPES[skip:skip + only]:
         x = torch.randn(M, K, device=dev).bfloat16()
         w = torch.randn(N, K, device=dev).bfloat16() / K**0.5
         dz = torch.randn(M, N, device=dev).bfloat16()
