@@ -126,8 +126,10 @@ def main() -> None:
         try:
             trainer.enable_graph(next(it))
         except Exception as e:  # fall back to eager launches
+            import sys
+
             if rank == 0:
-                print(f"# graph capture failed ({e!r}); running eager", flush=True)
+                print(f"# graph capture failed ({e!r}); running eager", file=sys.stderr, flush=True)
     for _ in range(args.warmup):
         trainer.train_step(next(it))
     sync()
