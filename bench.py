@@ -121,15 +121,25 @@ def main() -> None:
             torch.cuda.synchronize()
 
     it = iter(data)
-    use_graph = on_gpu and (args.graph == "1" or (args.graph == "auto" and world == 1))
+    # hipGraph-capture the whole train step (RCCL collectives capture too —
+    # verified by benchmarks/graph_nccl_probe.py). Fallback is COORDINATED:
+    # if any rank fails to capture, every rank runs eager.
+    use_graph = on_gpu and args.graph != "0"
     if use_graph:
+        ok = True
         try:
             trainer.enable_graph(next(it))
-        except Exception as e:  # fall back to eager launches
+        except Exception as e:
+            ok = False
             import sys
 
             if rank == 0:
                 print(f"# graph capture failed ({e!r}); running eager", file=sys.stderr, flush=True)
+        if dist.is_initialized():
+            flag = torch.tensor([1 if ok else 0], dtype=torch.int32, device=device)
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            if flag.item() == 0:
+                trainer._graph = None  # someone failed: everyone eager
     for _ in range(args.warmup):
         trainer.train_step(next(it))
     sync()
