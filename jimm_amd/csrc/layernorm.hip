@@ -67,7 +67,8 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                               const float* __restrict__ w, const float* __restrict__ mean,
                               const float* __restrict__ rstd, const T* __restrict__ addend,
                               T* __restrict__ dx, float* __restrict__ dw,
-                              float* __restrict__ db, int64_t nrows, int H) {
+                              float* __restrict__ db, float* __restrict__ ws,
+                              int64_t nrows, int H) {
   // per-wave fp32 partial slabs: [waves][H] for dw and db — no atomics in
   // the row loop (each wave owns its slab; each lane its columns)
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -198,7 +199,9 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   }
   }
   __syncthreads();
-  // merge the per-wave slabs, one global atomic per column per block
+  // merge the per-wave slabs; default: one global atomic per column per
+  // block (order nondeterministic); deterministic mode (ws != nullptr,
+  // JIMM_AMD_DETERMINISTIC=1): per-block partial rows reduced by the host
   float* base_dw = reinterpret_cast<float*>(smem);
   float* base_db = base_dw + (size_t)waves_per_block * H;
   for (int i = threadIdx.x; i < H; i += blockDim.x) {
@@ -207,8 +210,13 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
       adw += base_dw[(size_t)wv * H + i];
       adb += base_db[(size_t)wv * H + i];
     }
-    atomicAdd(&dw[i], adw);
-    atomicAdd(&db[i], adb);
+    if (ws) {
+      ws[(size_t)blockIdx.x * 2 * H + i] = adw;
+      ws[(size_t)blockIdx.x * 2 * H + H + i] = adb;
+    } else {
+      atomicAdd(&dw[i], adw);
+      atomicAdd(&db[i], adb);
+    }
   }
 }
 
@@ -233,19 +241,18 @@ void launch_ln_fwd(const T* x, const float* w, const float* b, T* y, float* mean
 template <typename T>
 void launch_ln_bwd(const T* dy, const T* x, const float* w, const float* mean,
                    const float* rstd, const T* addend, T* dx, float* dw, float* db,
-                   int64_t nrows, int H, hipStream_t stream) {
+                   float* ws, int grid, int64_t nrows, int H, hipStream_t stream) {
   const int block = 256;
   const int waves_per_block = block / WAVE;
-  int grid = (int)std::min<int64_t>((nrows + waves_per_block - 1) / waves_per_block, 1024);
   size_t shmem = 2 * (size_t)waves_per_block * H * sizeof(float);
   auto pick = [&](auto vec_tag) {
     constexpr int V = decltype(vec_tag)::value;
     if (addend)
       hipLaunchKernelGGL((ln_bwd_kernel<T, V, true>), dim3(grid), dim3(block), shmem,
-                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, nrows, H);
+                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, ws, nrows, H);
     else
       hipLaunchKernelGGL((ln_bwd_kernel<T, V, false>), dim3(grid), dim3(block), shmem,
-                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, nrows, H);
+                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, ws, nrows, H);
   };
   if (H % (WAVE * 4) == 0) pick(std::integral_constant<int, 4>{});
   else if (H % (WAVE * 2) == 0) pick(std::integral_constant<int, 2>{});
@@ -301,20 +308,39 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
                 addend->numel() == x.numel());
     addp = addend->data_ptr();
   }
+  // JIMM_AMD_DETERMINISTIC=1: per-block partials + ordered host reduce
+  // instead of cross-block fp32 atomics (SURVEY §5 deterministic mode)
+  static const bool deterministic = [] {
+    const char* e = getenv("JIMM_AMD_DETERMINISTIC");
+    return e && std::string(e) == "1";
+  }();
+  const int waves_per_block = 256 / WAVE;
+  const int grid = (int)std::min<int64_t>((nrows + waves_per_block - 1) / waves_per_block, 1024);
+  torch::Tensor wsbuf;
+  float* wsp = nullptr;
+  if (deterministic) {
+    wsbuf = torch::empty({(int64_t)grid, 2, (int64_t)H}, x.options().dtype(torch::kFloat32));
+    wsp = wsbuf.data_ptr<float>();
+  }
   if (x.scalar_type() == torch::kBFloat16) {
     launch_ln_bwd(reinterpret_cast<const bf16*>(dy.data_ptr()),
                   reinterpret_cast<const bf16*>(x.data_ptr()), wf.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
                   reinterpret_cast<const bf16*>(addp),
                   reinterpret_cast<bf16*>(dx.data_ptr()), dw.data_ptr<float>(),
-                  db.data_ptr<float>(), nrows, H, stream);
+                  db.data_ptr<float>(), wsp, grid, nrows, H, stream);
   } else if (x.scalar_type() == torch::kFloat32) {
     launch_ln_bwd(dy.data_ptr<float>(), x.data_ptr<float>(), wf.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
                   reinterpret_cast<const float*>(addp), dx.data_ptr<float>(),
-                  dw.data_ptr<float>(), db.data_ptr<float>(), nrows, H, stream);
+                  dw.data_ptr<float>(), db.data_ptr<float>(), wsp, grid, nrows, H, stream);
   } else {
     TORCH_CHECK(false, "layernorm: unsupported dtype ", x.scalar_type());
+  }
+  if (deterministic) {
+    auto sums = wsbuf.sum(0);  // fixed-order tree reduce
+    dw = sums[0];
+    db = sums[1];
   }
   // cast param grads to param dtype at the python layer if needed
   return {dx, dw.to(w.scalar_type()), db.to(w.scalar_type())};
