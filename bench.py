@@ -92,7 +92,7 @@ def main() -> None:
     rank, world, local_rank, device = init_distributed()
     on_gpu = device.type == "cuda"
     dtype = torch.bfloat16 if on_gpu else torch.float32
-    batch = args.batch or ({"vit": 256, "vitl384": 128, "clip": 512, "siglip": 256}[args.task] if on_gpu else 4)
+    batch = args.batch or ({"vit": 1024, "vitl384": 128, "clip": 1024, "siglip": 512}[args.task] if on_gpu else 4)
 
     torch.manual_seed(1234 + rank)
     model, trainer, model_name = build(args.task, device, dtype)
