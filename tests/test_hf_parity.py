@@ -212,3 +212,56 @@ def test_siglip2_vision_parity(tmp_path):
         ref_t = hf.text_model(input_ids=ids).pooler_output
         out_t = model.encode_text(ids)
     assert torch.allclose(out_t, ref_t, atol=1e-4), (out_t - ref_t).abs().max()
+
+
+def test_checkpoint_coverage_asserts(tmp_path):
+    """The bidirectional coverage asserts of the reference's from_pretrained
+    (zero nonvisited, zero leftover, shape checks) fire."""
+    from safetensors.torch import load_file, save_file
+
+    hf, d = _tiny_vit(tmp_path, old_keys=False)
+    sd_path = d / "model.safetensors"
+    good = load_file(sd_path)
+
+    # extra unconsumed key -> error
+    bad = dict(good)
+    bad["vit.mystery_weight"] = torch.zeros(3)
+    save_file(bad, str(sd_path))
+    with pytest.raises(ValueError, match="unconsumed"):
+        jimm_amd.VisionTransformer.from_pretrained(str(d))
+
+    # missing key -> error
+    bad = {k: v for k, v in good.items() if "classifier.bias" not in k}
+    save_file(bad, str(sd_path))
+    with pytest.raises((ValueError, KeyError)):
+        jimm_amd.VisionTransformer.from_pretrained(str(d))
+
+    # position_ids buffers are tolerated (IGNORE_PATTERNS)
+    ok = dict(good)
+    ok["vit.embeddings.position_ids"] = torch.arange(5)
+    save_file(ok, str(sd_path))
+    jimm_amd.VisionTransformer.from_pretrained(str(d))
+
+
+def test_roundtrip_save_load(tmp_path):
+    """save_pretrained -> from_pretrained is the identity (all three models)."""
+    torch.manual_seed(0)
+    models = {
+        "vit": jimm_amd.VisionTransformer(num_classes=3, img_size=32, patch_size=16,
+                                          num_layers=1, num_heads=2, mlp_dim=64, hidden_size=32),
+        "clip": jimm_amd.CLIP(embed_dim=16, image_resolution=32, vision_layers=1,
+                              vision_width=64, vision_patch_size=16, context_length=5,
+                              vocab_size=33, transformer_width=32, transformer_heads=2,
+                              transformer_layers=1),
+        "siglip": jimm_amd.SigLIP(image_resolution=32, vision_layers=1, vision_width=64,
+                                  vision_patch_size=16, context_length=5, vocab_size=33,
+                                  transformer_width=64, transformer_heads=1,
+                                  transformer_layers=1, vision_heads=2),
+    }
+    for name, m in models.items():
+        d = tmp_path / name
+        m.save_pretrained(str(d))
+        m2 = type(m).from_pretrained(str(d))
+        for (k, a), (k2, b) in zip(sorted(m.state_dict().items()), sorted(m2.state_dict().items())):
+            assert k == k2
+            assert torch.equal(a, b), (name, k)
