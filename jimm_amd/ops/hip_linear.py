@@ -1,14 +1,20 @@
 """GPU linear / patch-embed path: GEMM + fused bias/activation/residual epilogue.
 
 GEMM engine selection (env ``JIMM_AMD_GEMM``):
-  * ``hip``  — in-house MFMA bf16 GEMM kernel (csrc/gemm.hip), epilogue fused
-               into the GEMM itself (bias + gelu/quickgelu + residual).
-  * ``blas`` — rocBLAS via ``torch.matmul`` (plain library GEMM) followed by
-               the fused HIP bias+act(+residual) elementwise kernel.
+  * ``blas`` (default) — hipBLASLt/rocBLAS (plain library GEMMs, with the
+               committed MI355X TunableOp table; bias-only linears use the
+               hipBLASLt bias epilogue via ``torch.addmm``), followed by the
+               fused HIP bias+act(+residual) elementwise kernel where needed.
+               Measured faster than the in-house GEMMs on the model-zoo
+               shapes (benchmarks/gemm_bench.py).
+  * ``hip``  — in-house MFMA bf16 GEMM kernels (csrc/gemm.hip, 128x128 tile;
+               csrc/gemm256.hip, 256x256 tile) with the epilogue fused into
+               the GEMM itself.
+  * fp8      — ``jimm_amd.ops.set_fp8(True)``: e4m3 scaled_mm forward path.
 
-Backward GEMMs (dx = dz@W, dW = dz^T@x) always run through rocBLAS — they are
-plain GEMMs with no epilogue to fuse; dz itself comes from the fused HIP
-``act_bwd`` kernel.
+Backward GEMMs: dX runs through rocBLAS; dW defaults to rocBLAS (TunableOp
+split-K picks) with the in-house split-K TN kernel opt-in (JIMM_AMD_DW=hip);
+dz comes from the fused HIP ``act_bwd`` kernel and db from ``colsum``.
 """
 
 from __future__ import annotations
