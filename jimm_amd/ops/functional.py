@@ -188,10 +188,13 @@ class _AttentionQKVFn(torch.autograd.Function):
 
 
 def attention_qkv(qkv: torch.Tensor, *, causal: bool = False, scale: float | None = None) -> torch.Tensor:
-    """qkv (B, L, 3, H, D) -> (B, H, L, D). GPU fast path for self-attention."""
+    """qkv (B, L, 3, H, D) -> (B, H, L, D). GPU fast path for self-attention.
+
+    The flash kernels are bf16/head_dim-64 by design; other dtypes/head dims
+    run the composite torch math (on GPU or CPU alike)."""
     if scale is None:
         scale = 1.0 / math.sqrt(qkv.shape[-1])
-    if _backend.use_hip(qkv):
+    if _backend.use_hip(qkv) and qkv.dtype == torch.bfloat16 and qkv.shape[-1] == 64:
         return _AttentionQKVFn.apply(qkv, causal, scale)
     q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
     return attention(q, k, v, causal=causal, scale=scale)
@@ -207,9 +210,9 @@ def attention(
 ) -> torch.Tensor:
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _backend.use_hip(q):
+    if _backend.use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] == 64:
         return _AttentionFn.apply(q, k, v, causal, scale)
-    # CPU reference (fp32 math; float64 preserved for gradcheck)
+    # composite reference (fp32 math; float64 preserved for gradcheck)
     ref_dtype = torch.float64 if q.dtype == torch.float64 else torch.float32
     qf, kf, vf = q.to(ref_dtype), k.to(ref_dtype), v.to(ref_dtype)
     s = torch.matmul(qf, kf.transpose(-1, -2)) * scale

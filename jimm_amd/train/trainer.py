@@ -33,6 +33,9 @@ class TrainConfig:
 
 class Trainer:
     def __init__(self, model: torch.nn.Module, cfg: TrainConfig, *, process_group=None):
+        from jimm_amd.ops._backend import maybe_enable_tunableop
+
+        maybe_enable_tunableop()  # committed hipBLASLt algo table (MI355X)
         self.model = model
         self.cfg = cfg
         self.ddp = DataParallelGrads(model, bucket_bytes=cfg.bucket_bytes, process_group=process_group)

@@ -544,3 +544,15 @@ def test_dual_tower_gpu_vs_cpu_parity(which):
         out, _ = gm(imgs.to(dev(), torch.bfloat16), ids.to(dev()))
     err = (out.float().cpu() - ref).abs().max().item()
     assert err < 0.25, err  # bf16 end-to-end vs fp32; logits are O(1-10)
+
+
+def test_fp32_model_on_gpu_runs():
+    """fp32 GPU inference routes attention through the composite path
+    (the flash kernels are bf16-only by design) without errors."""
+    torch.manual_seed(0)
+    m = jimm_amd.VisionTransformer(num_classes=4, img_size=64, patch_size=16,
+                                   num_layers=1, num_heads=2, mlp_dim=128,
+                                   hidden_size=128).to(dev()).eval()
+    with torch.no_grad():
+        out = m(torch.randn(2, 3, 64, 64, device=dev()))
+    assert torch.isfinite(out).all()
