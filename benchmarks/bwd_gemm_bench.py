@@ -38,6 +38,18 @@ SHAPES = [
     (65536, 2304, 768), (65536, 768, 768), (65536, 3072, 768), (65536, 768, 3072),
     (16384, 2304, 768), (16384, 768, 768), (16384, 3072, 768), (16384, 768, 3072),
     (256, 768, 768),               # text head / MAP-head linears
+    # default-batch shapes (vit b1024, clip b1024, siglip b512, vitl384 b128)
+    (201728, 2304, 768), (201728, 768, 768), (201728, 3072, 768), (201728, 768, 3072),
+    (200704, 768, 768),
+    (73856, 3072, 1024), (73856, 1024, 1024), (73856, 4096, 1024), (73856, 1024, 4096),
+    (73728, 1024, 768),
+    (51200, 2304, 768), (51200, 768, 768), (51200, 3072, 768), (51200, 768, 3072),
+    (50176, 768, 3072),
+    (78848, 1536, 512), (78848, 512, 512), (78848, 2048, 512), (78848, 512, 2048),
+    (1024, 512, 768), (1024, 512, 512), (1024, 1000, 768),
+    (131072, 2304, 768), (131072, 768, 768), (131072, 3072, 768), (131072, 768, 3072),
+    (32768, 2304, 768), (32768, 768, 768), (32768, 3072, 768), (32768, 768, 3072),
+    (512, 768, 768),
 ]
 
 
