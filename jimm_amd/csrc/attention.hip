@@ -35,12 +35,14 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
 constexpr int QBLK = 64;    // q rows per strip (16 per wave)
-constexpr int NSTRIP = 4;   // q strips per workgroup (VGPR-bounded)
+// NSTRIP (q strips per workgroup) is a template param: 4 = fewer K/V
+// staging passes, 2 waves/SIMD; 2 = 3 waves/SIMD, twice the staging
+// (JIMM_AMD_ATTN_NSTRIP selects; default 4)
 constexpr int KVBLK = 64;   // keys per LDS tile
 constexpr int D = 64;       // head_dim (checked host-side)
 constexpr int LDS_PITCH = D + 8;  // +8 shorts: bank-conflict pad for b128 reads
 
-template <bool CAUSAL>
+template <bool CAUSAL, int NSTRIP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale, int H,
@@ -280,12 +282,27 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto o_storage = torch::empty({B, Lq, H, 64}, q.options());
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
+  static const int nstrip = [] {
+    const char* e = getenv("JIMM_AMD_ATTN_NSTRIP");
+    return (e && std::string(e) == "2") ? 2 : 4;
+  }();
   const int ntq = (Lq + QBLK - 1) / QBLK;
-  const dim3 grid((ntq + NSTRIP - 1) / NSTRIP, (unsigned)((int64_t)B * H));
+  const dim3 grid((ntq + nstrip - 1) / nstrip, (unsigned)((int64_t)B * H));
   const size_t shmem = (2 * KVBLK * LDS_PITCH + 2 * D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
 #define ATTN_LAUNCH(C)                                                                     \
-  hipLaunchKernelGGL((attn_fwd_kernel<C>), grid, dim3(256), shmem, stream,                 \
+  if (nstrip == 2)                                                                         \
+    hipLaunchKernelGGL((attn_fwd_kernel<C, 2>), grid, dim3(256), shmem, stream,            \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                          \
+                     reinterpret_cast<bf16*>(o_storage.data_ptr()), lse.data_ptr<float>(), \
+                     Lq, Lk, (float)scale, H, q.stride(0), q.stride(1), q.stride(2),       \
+                     k.stride(0), k.stride(1), k.stride(2),                                \
+                     v.stride(0), v.stride(1), v.stride(2),                                \
+                     o.stride(0), o.stride(1), o.stride(2));                               \
+  else                                                                                     \
+  hipLaunchKernelGGL((attn_fwd_kernel<C, 4>), grid, dim3(256), shmem, stream,              \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                          \
                      reinterpret_cast<const bf16*>(v.data_ptr()),                          \
