@@ -282,11 +282,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto o_storage = torch::empty({B, Lq, H, 64}, q.options());
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
-  static const int nstrip = [] {
-    const char* e = getenv("JIMM_AMD_ATTN_NSTRIP");
-    return (e && std::string(e) == "2") ? 2 : 4;
-  }();
+  // 4 strips only when the q tiles divide evenly (measured: 4 wins at
+  // L=197 — one staging pass; 2 wins elsewhere via 3 waves/SIMD occupancy:
+  // L=577 261->298 TF/s, L=257 184->216, L=77 46->60)
   const int ntq = (Lq + QBLK - 1) / QBLK;
+  int nstrip = (ntq % 4 == 0) ? 4 : 2;
+  if (const char* e = getenv("JIMM_AMD_ATTN_NSTRIP")) nstrip = (std::string(e) == "2") ? 2 : 4;
   const dim3 grid((ntq + nstrip - 1) / nstrip, (unsigned)((int64_t)B * H));
   const size_t shmem = (2 * KVBLK * LDS_PITCH + 2 * D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
