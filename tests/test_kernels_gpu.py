@@ -556,3 +556,30 @@ def test_fp32_model_on_gpu_runs():
     with torch.no_grad():
         out = m(torch.randn(2, 3, 64, 64, device=dev()))
     assert torch.isfinite(out).all()
+
+
+@pytest.mark.parametrize(
+    "B,img,patch,layers,heads,hidden,mlp",
+    [
+        (1, 112, 16, 1, 1, 64, 128),     # B=1, 49 patches
+        (3, 448, 32, 1, 2, 128, 256),    # 196 patches, odd B
+        (2, 96, 16, 2, 3, 192, 384),     # 36 patches, 3 heads
+        (5, 64, 64, 1, 2, 128, 256),     # single patch (L=2 with CLS)
+    ],
+)
+def test_vit_stress_shapes_gpu(B, img, patch, layers, heads, hidden, mlp):
+    """Odd batch/sequence/width combinations run the HIP path end to end
+    (fwd+bwd) and match the CPU fp32 reference."""
+    torch.manual_seed(0)
+    m = jimm_amd.VisionTransformer(num_classes=7, img_size=img, patch_size=patch,
+                                   num_layers=layers, num_heads=heads, mlp_dim=mlp,
+                                   hidden_size=hidden).eval()
+    x = torch.randn(B, 3, img, img)
+    with torch.no_grad():
+        ref = m(x)
+    gm = m.to(dev(), torch.bfloat16)
+    xg = x.to(dev(), torch.bfloat16).requires_grad_(True)
+    out = gm(xg)
+    out.float().square().sum().backward()
+    assert torch.isfinite(xg.grad).all()
+    assert (out.float().cpu() - ref).abs().max().item() < 0.2, (out.float().cpu() - ref).abs().max().item()
