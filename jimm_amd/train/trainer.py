@@ -149,13 +149,20 @@ class Trainer:
 
     @torch.no_grad()
     def eval_step(self, batch) -> dict[str, torch.Tensor]:
-        if self.cfg.task != "vit":
-            raise NotImplementedError
-        images, labels = batch
-        logits = self.model(images)
+        if self.cfg.task == "vit":
+            images, labels = batch
+            logits = self.model(images)
+            return {
+                "loss": L.softmax_cross_entropy(logits, labels),
+                "accuracy": (logits.argmax(-1) == labels).float().mean(),
+            }
+        # clip/siglip: local-batch image->text retrieval accuracy (diagonal)
+        images, ids = batch
+        logits_per_image, _ = self.model(images, ids)
+        labels = torch.arange(logits_per_image.shape[0], device=logits_per_image.device)
         return {
-            "loss": L.softmax_cross_entropy(logits, labels),
-            "accuracy": (logits.argmax(-1) == labels).float().mean(),
+            "retrieval_i2t": (logits_per_image.argmax(-1) == labels).float().mean(),
+            "retrieval_t2i": (logits_per_image.argmax(0) == labels).float().mean(),
         }
 
 

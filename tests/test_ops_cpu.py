@@ -122,3 +122,17 @@ def test_trainer_checkpoint_resume(tmp_path):
         tr2.train_step(b)
     for p, q in zip(ref, tr2.model.parameters()):
         assert torch.allclose(p, q, atol=1e-6), (p - q).abs().max()
+
+
+def test_eval_step_contrastive():
+    import jimm_amd
+    from jimm_amd.train import TrainConfig, Trainer
+
+    torch.manual_seed(0)
+    m = jimm_amd.CLIP(embed_dim=16, image_resolution=32, vision_layers=1, vision_width=64,
+                      vision_patch_size=16, context_length=5, vocab_size=33,
+                      transformer_width=32, transformer_heads=2, transformer_layers=1)
+    tr = Trainer(m, TrainConfig(task="clip"))
+    out = tr.eval_step((torch.randn(4, 3, 32, 32), torch.randint(0, 33, (4, 5))))
+    assert set(out) == {"retrieval_i2t", "retrieval_t2i"}
+    assert 0.0 <= out["retrieval_i2t"].item() <= 1.0
