@@ -38,7 +38,8 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
 constexpr int BLK = 64;            // q rows / keys per workgroup tile
-constexpr int NKV = 2;             // kv tiles per dkv workgroup (VGPR-bounded; 4 spills)
+// NKV (kv tiles per dkv workgroup) is a template param: 2 = half the Q/dO
+// staging passes at 2 waves/SIMD; 1 = 3 waves/SIMD occupancy
 constexpr int NQS = 2;             // q tiles per dq workgroup
 constexpr int D = 64;              // head_dim (checked host-side)
 constexpr int PITCH = D + 8;       // bank-conflict pad for b128 reads
@@ -83,8 +84,8 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
 // dK/dV kernel: workgroup owns keys [kv0, kv0+64); loops q tiles
 // ---------------------------------------------------------------------------
 
-template <bool CAUSAL>
-__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
+template <bool CAUSAL, int NKV>
+__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int Lq, int Lk, float scale, int H,
@@ -522,12 +523,30 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   const size_t shmem_dkv = (4 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
   const size_t shmem_dq = (3 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
   const int ntk = (Lk + BLK - 1) / BLK;
-  const dim3 grid_dkv((ntk + NKV - 1) / NKV, (unsigned)((int64_t)B * H));
+  static const int nkv_env = [] {
+    const char* e = getenv("JIMM_AMD_ATTN_NKV");
+    return e ? atoi(e) : 0;
+  }();
+  const int nkv = nkv_env ? nkv_env : 2;
+  const dim3 grid_dkv((ntk + nkv - 1) / nkv, (unsigned)((int64_t)B * H));
   const int ntq = (Lq + BLK - 1) / BLK;
   const dim3 grid_dq((ntq + NQS - 1) / NQS, (unsigned)((int64_t)B * H));
 
 #define DKV_LAUNCH(C)                                                                        \
-  hipLaunchKernelGGL((attn_bwd_dkv_kernel<C>), grid_dkv, dim3(256), shmem_dkv, stream,       \
+  if (nkv == 1)                                                                              \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 1>), grid_dkv, dim3(256), shmem_dkv, stream,  \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
+                     Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dk.data_ptr()),           \
+                     reinterpret_cast<bf16*>(dv.data_ptr()), Lq, Lk, (float)scale, H,        \
+                     q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),        \
+                     k.stride(2), v.stride(0), v.stride(1), v.stride(2), dO.stride(0),       \
+                     dO.stride(1), dO.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),   \
+                     dv.stride(0), dv.stride(1), dv.stride(2));                              \
+  else                                                                                       \
+  hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 2>), grid_dkv, dim3(256), shmem_dkv, stream,    \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                            \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                            \
                      reinterpret_cast<const bf16*>(v.data_ptr()),                            \
