@@ -85,7 +85,7 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
 // ---------------------------------------------------------------------------
 
 template <bool CAUSAL, int NKV>
-__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int Lq, int Lk, float scale, int H,
