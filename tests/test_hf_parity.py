@@ -265,3 +265,34 @@ def test_roundtrip_save_load(tmp_path):
         for (k, a), (k2, b) in zip(sorted(m.state_dict().items()), sorted(m2.state_dict().items())):
             assert k == k2
             assert torch.equal(a, b), (name, k)
+
+
+def test_roundtrip_fuzz_configs(tmp_path):
+    """Randomized model configs survive save_pretrained -> from_pretrained
+    exactly (hypothesis-style sweep without the decorator overhead)."""
+    import random
+
+    rng = random.Random(0)
+    for trial in range(6):
+        hidden = rng.choice([32, 64, 96])
+        heads = rng.choice([1, 2]) if hidden % 64 else 2
+        if hidden % heads:
+            heads = 1
+        layers = rng.choice([1, 2, 3])
+        patch = rng.choice([8, 16])
+        img = patch * rng.choice([2, 3, 4])
+        ncls = rng.choice([2, 7, 10])
+        m = jimm_amd.VisionTransformer(num_classes=ncls, img_size=img, patch_size=patch,
+                                       num_layers=layers, num_heads=heads,
+                                       mlp_dim=hidden * 2, hidden_size=hidden)
+        d = tmp_path / f"t{trial}"
+        m.save_pretrained(str(d))
+        m2 = jimm_amd.VisionTransformer.from_pretrained(str(d))
+        sd, sd2 = m.state_dict(), m2.state_dict()
+        assert set(sd) == set(sd2)
+        for k in sd:
+            assert torch.equal(sd[k], sd2[k]), (trial, k)
+        # and the forward agrees
+        x = torch.randn(2, 3, img, img)
+        with torch.no_grad():
+            assert torch.allclose(m(x), m2(x), atol=1e-6)
