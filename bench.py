@@ -79,6 +79,7 @@ def main() -> None:
     p.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16",
                    help="fp8: e4m3 forward GEMMs via MFMA fp8 (bf16 backward)")
     p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
+    p.add_argument("--ckpt", action="store_true", help="gradient checkpointing (huge batches)")
     p.add_argument("--graph", choices=["auto", "1", "0"], default="auto",
                    help="hipGraph-capture the train step (auto: single-GPU only)")
     args = p.parse_args()
@@ -96,6 +97,8 @@ def main() -> None:
 
     torch.manual_seed(1234 + rank)
     model, trainer, model_name = build(args.task, device, dtype)
+    if args.ckpt:
+        model.gradient_checkpointing_enable()
 
     if args.dtype == "fp8" and on_gpu:
         from jimm_amd.ops import set_fp8

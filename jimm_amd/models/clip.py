@@ -72,6 +72,10 @@ class CLIP(nn.Module):
         self.text_projection = nn.Linear(transformer_width, embed_dim, bias=False)
         self.logit_scale = nn.Parameter(torch.tensor(math.log(1 / 0.07)))
 
+    def gradient_checkpointing_enable(self) -> None:
+        self.vision_model.encoder.gradient_checkpointing = True
+        self.text_model.encoder.gradient_checkpointing = True
+
     def encode_image(self, images: torch.Tensor) -> torch.Tensor:
         return ops.linear(self.vision_model(images), self.visual_projection.weight)
 

@@ -143,6 +143,16 @@ class Encoder(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if getattr(self, "gradient_checkpointing", False) and torch.is_grad_enabled():
+            from torch.utils.checkpoint import checkpoint
+
+            for layer in self.layers:
+                # recompute the block's forward during backward: activation
+                # memory per block drops from ~16 to ~1 tensors of (B,L,H)
+                # (needed for e.g. SigLIP at 4096 pairs/GPU = the 32k-global
+                # BASELINE config on 8 GPUs; without it b4096 OOMs 288 GB)
+                x = checkpoint(layer, x, use_reentrant=False)
+            return x
         for layer in self.layers:
             x = layer(x)
         return x

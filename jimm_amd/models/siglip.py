@@ -69,6 +69,12 @@ class SigLIP(nn.Module):
         self.logit_scale = nn.Parameter(torch.tensor(1.0))
         self.logit_bias = nn.Parameter(torch.tensor(0.0))
 
+    def gradient_checkpointing_enable(self) -> None:
+        """Recompute encoder blocks in backward (both towers) — trades ~35%
+        step time for ~10x less activation memory (huge-batch training)."""
+        self.vision_model.encoder.gradient_checkpointing = True
+        self.text_model.encoder.gradient_checkpointing = True
+
     def encode_image(self, images: torch.Tensor) -> torch.Tensor:
         return self.vision_model(images)  # no visual projection (siglip.py:123-133)
 

@@ -56,6 +56,9 @@ class VisionTransformer(nn.Module):
         else:
             self.classifier = None
 
+    def gradient_checkpointing_enable(self) -> None:
+        self.vision.encoder.gradient_checkpointing = True
+
     def forward(self, images: torch.Tensor) -> torch.Tensor:
         x = self.vision(images)  # (B, H) CLS- or MAP-pooled
         if self.do_classification:

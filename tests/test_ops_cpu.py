@@ -136,3 +136,26 @@ def test_eval_step_contrastive():
     out = tr.eval_step((torch.randn(4, 3, 32, 32), torch.randint(0, 33, (4, 5))))
     assert set(out) == {"retrieval_i2t", "retrieval_t2i"}
     assert 0.0 <= out["retrieval_i2t"].item() <= 1.0
+
+
+def test_gradient_checkpointing_matches():
+    """Checkpointed blocks give the same loss/grads as the plain path."""
+    import jimm_amd
+
+    torch.manual_seed(0)
+    def run(ckpt):
+        torch.manual_seed(1)
+        m = jimm_amd.VisionTransformer(num_classes=5, img_size=32, patch_size=16,
+                                       num_layers=2, num_heads=2, mlp_dim=64, hidden_size=32)
+        if ckpt:
+            m.gradient_checkpointing_enable()
+        x = torch.randn(2, 3, 32, 32)
+        out = m(x)
+        out.square().sum().backward()
+        return out.detach(), [p.grad.clone() for p in m.parameters()]
+
+    o1, g1 = run(False)
+    o2, g2 = run(True)
+    assert torch.allclose(o1, o2, atol=1e-6)
+    for a, b in zip(g1, g2):
+        assert torch.allclose(a, b, atol=1e-5)
