@@ -32,6 +32,7 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
                     torch::Tensor dO, torch::Tensor lse, torch::Tensor dq, torch::Tensor dk,
                     torch::Tensor dv, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor B);
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype);
 std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x);
 torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor rinv);
@@ -64,6 +65,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_ds", &attn_ds, "attention bwd: dP -> dS in place (K15)");
   m.def("attn_bwd_fused", &attn_bwd_fused, "fused flash attention backward (K15)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
+  m.def("mfma_probe32", &mfma_probe32, "MFMA 32x32x16 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
   m.def("gemm_dw_supported", &gemm_dw_supported, "split-K dW GEMM shape check");
   m.def("l2norm_fwd", &l2norm_fwd, "row L2-normalize forward (K12)");

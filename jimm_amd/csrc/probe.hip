@@ -35,7 +35,46 @@ __global__ void mfma_probe_kernel(const bf16* __restrict__ A, const bf16* __rest
   for (int r = 0; r < 4; ++r) C[(hi * 4 + r) * 16 + lo] = c[r];
 }
 
+typedef float f32x16_t __attribute__((ext_vector_type(16)));
+
+__global__ void mfma_probe32_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                                    float* __restrict__ C) {
+  // v_mfma_f32_32x32x16_bf16 — assumed layouts (the x16 extension of the
+  // CDNA3 32x32x8 mapping), verified on hardware by test_mfma_probe32:
+  //   A[i][k]: lane l holds A[l&31][(l>>5)*8 + j]          j=0..7
+  //   B[k][j]: lane l holds B[(l>>5)*8 + j][l&31]
+  //   C[i][j]: lane l holds C[8*b + (l>>5)*4 + r][l&31]    b=0..3, r=0..3,
+  //            vector index 4*b + r
+  const int lane = threadIdx.x % WAVE;
+  const int lo = lane & 31, hi = lane >> 5;
+  bf16x8_t a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = *reinterpret_cast<const short*>(&A[lo * 16 + hi * 8 + j]);      // A[32][16] row-major
+    b[j] = *reinterpret_cast<const short*>(&B[(hi * 8 + j) * 32 + lo]);    // B[16][32] row-major
+  }
+  f32x16_t c = {};
+  c = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int blk = 0; blk < 4; ++blk)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      C[(8 * blk + hi * 4 + r) * 32 + lo] = c[4 * blk + r];
+}
+
 }  // namespace
+
+torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({32, 16}) && B.sizes() == torch::IntArrayRef({16, 32}));
+  auto C = torch::empty({32, 32}, A.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, stream,
+                     reinterpret_cast<const bf16*>(A.contiguous().data_ptr()),
+                     reinterpret_cast<const bf16*>(B.contiguous().data_ptr()),
+                     C.data_ptr<float>());
+  return C;
+}
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);

@@ -606,3 +606,12 @@ def test_gradient_checkpointing_gpu():
     b = run(True)
     for x, y in zip(a, b):
         assert abs(x - y) < 5e-2, (a, b)
+
+
+def test_mfma_probe32():
+    torch.manual_seed(0)
+    A = torch.randn(32, 16, device=dev()).bfloat16()
+    B = torch.randn(16, 32, device=dev()).bfloat16()
+    C = EXT.mfma_probe32(A, B)
+    ref = A.float() @ B.float()
+    assert (C - ref).abs().max().item() < 1e-2, (C - ref).abs().max().item()
