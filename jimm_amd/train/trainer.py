@@ -28,6 +28,12 @@ class TrainConfig:
     weight_decay: float = 0.0
     bucket_bytes: int = 25 * 1024 * 1024
     log_every: int = 50
+    # LR schedule: None = constant; "cosine" decays lr -> min_lr over
+    # total_steps after a linear warmup_steps ramp
+    lr_schedule: str | None = None
+    warmup_steps: int = 0
+    total_steps: int = 0
+    min_lr: float = 0.0
     extra: dict = field(default_factory=dict)
 
 
@@ -51,7 +57,34 @@ class Trainer:
         self._static_batch = None
         self._static_out = None
 
+    def _current_lr(self) -> float:
+        c = self.cfg
+        if c.lr_schedule is None:
+            return c.lr
+        step = self.step_idx
+        if c.warmup_steps and step < c.warmup_steps:
+            return c.lr * (step + 1) / c.warmup_steps
+        if c.lr_schedule == "cosine" and c.total_steps:
+            import math
+
+            t = min(max(step - c.warmup_steps, 0), max(c.total_steps - c.warmup_steps, 1))
+            f = 0.5 * (1 + math.cos(math.pi * t / max(c.total_steps - c.warmup_steps, 1)))
+            return c.min_lr + (c.lr - c.min_lr) * f
+        return c.lr
+
+    def _apply_lr(self) -> None:
+        lr = self._current_lr()
+        for g in self.opt.param_groups:
+            g["lr"] = lr
+        # under graph replay the captured Adam reads lr from device memory:
+        # refresh it OUTSIDE the graph
+        if self._graph is not None and getattr(self.opt, "_prepared", None):
+            self.opt._prepared[2].fill_(lr)
+            self.opt._last_lr = lr
+
     def train_step(self, batch) -> dict[str, torch.Tensor]:
+        if self.cfg.lr_schedule is not None:
+            self._apply_lr()
         if self._roctx:
             torch.cuda.nvtx.range_push(f"train_step_{self.cfg.task}")  # roctx range on ROCm
         if self._graph is not None:

@@ -159,3 +159,23 @@ def test_gradient_checkpointing_matches():
     assert torch.allclose(o1, o2, atol=1e-6)
     for a, b in zip(g1, g2):
         assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_lr_schedule():
+    import jimm_amd
+    from jimm_amd.train import SyntheticImages, TrainConfig, Trainer
+
+    torch.manual_seed(0)
+    m = jimm_amd.VisionTransformer(num_classes=5, img_size=32, patch_size=16,
+                                   num_layers=1, num_heads=2, mlp_dim=64, hidden_size=32)
+    tr = Trainer(m, TrainConfig(task="vit", lr=1e-3, lr_schedule="cosine",
+                                warmup_steps=2, total_steps=10, min_lr=1e-5))
+    data = SyntheticImages(2, 32, 5, torch.device("cpu"))
+    it = iter(data)
+    lrs = []
+    for _ in range(10):
+        tr.train_step(next(it))
+        lrs.append(tr.opt.param_groups[0]["lr"])
+    assert lrs[0] < lrs[1] <= 1e-3          # warmup ramps
+    assert lrs[-1] < lrs[3]                 # cosine decays
+    assert lrs[-1] >= 1e-5
