@@ -17,14 +17,15 @@ it = iter(data)
 tr.enable_graph(next(it))
 losses = []
 t0 = time.perf_counter()
-for step in range(500):
+STEPS = int(os.environ.get("SOAK_STEPS", "500"))
+for step in range(STEPS):
     out = tr.train_step(next(it))
-    if step % 100 == 0 or step == 499:
+    if step % 100 == 0 or step == STEPS - 1:
         torch.cuda.synchronize()
         losses.append(out["loss"].item())
         print(f"step {step}: loss {losses[-1]:.4f} mem {torch.cuda.memory_allocated()/2**30:.2f} GiB "
               f"peak {torch.cuda.max_memory_allocated()/2**30:.2f} GiB", flush=True)
 dt = time.perf_counter() - t0
-print(f"500 steps in {dt:.1f}s ({500*256/dt:.0f} img/s); loss {losses[0]:.3f} -> {losses[-1]:.3f}")
+print(f"{STEPS} steps in {dt:.1f}s ({STEPS*256/dt:.0f} img/s); loss {losses[0]:.3f} -> {losses[-1]:.3f}")
 assert losses[-1] < losses[0], "loss did not decrease"
 print("SOAK OK")
