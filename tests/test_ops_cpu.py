@@ -179,3 +179,29 @@ def test_lr_schedule():
     assert lrs[0] < lrs[1] <= 1e-3          # warmup ramps
     assert lrs[-1] < lrs[3]                 # cosine decays
     assert lrs[-1] >= 1e-5
+
+
+def test_siglip_loss_chunking_invariant():
+    """Column-chunked sigmoid loss equals the unchunked computation."""
+    from jimm_amd.ops import losses as L
+
+    torch.manual_seed(0)
+    img = torch.randn(16, 32, requires_grad=True)
+    txt = torch.randn(16, 32, requires_grad=True)
+    s, b = torch.tensor(0.5), torch.tensor(-1.0)
+    l1 = L.siglip_sigmoid_loss(img, txt, s, b, gather=False, chunk_size=5)
+    l2 = L.siglip_sigmoid_loss(img, txt, s, b, gather=False, chunk_size=1 << 20)
+    assert torch.allclose(l1, l2, atol=1e-5), (l1, l2)
+
+
+def test_clip_loss_gather_flag_world1():
+    """With no process group, gather=True must equal gather=False."""
+    from jimm_amd.ops import losses as L
+
+    torch.manual_seed(0)
+    img = torch.randn(8, 16)
+    txt = torch.randn(8, 16)
+    s = torch.tensor(0.3)
+    a = L.clip_contrastive_loss(img, txt, s, gather=True)
+    b = L.clip_contrastive_loss(img, txt, s, gather=False)
+    assert torch.allclose(a, b)
