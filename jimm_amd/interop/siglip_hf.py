@@ -54,7 +54,7 @@ def _infer_config(sd: dict[str, torch.Tensor], image_size: int | None = None) ->
         context_length=sd["text_model.embeddings.position_embedding.weight"].shape[0],
         vocab_size=sd["text_model.embeddings.token_embedding.weight"].shape[0],
         transformer_width=tw,
-        transformer_heads=tw // 64,  # siglip.py:217
+        transformer_heads=max(1, tw // 64),  # siglip.py:217
         transformer_layers=n_layers("text_model"),
         transformer_mlp_dim=sd["text_model.encoder.layers.0.mlp.fc1.weight"].shape[0],
     )

@@ -55,7 +55,7 @@ def _infer_config(sd: dict[str, torch.Tensor]) -> dict:
     return dict(
         hidden_size=hidden,
         num_layers=max(layer_ids) + 1,
-        num_heads=hidden // 64,  # assumed head_dim 64 (vit.py:156-157)
+        num_heads=max(1, hidden // 64),  # assumed head_dim 64 (vit.py:156-157)
         mlp_dim=sd[fc1_key].shape[0],
         img_size=img,
         patch_size=patch,

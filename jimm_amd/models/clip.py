@@ -42,7 +42,7 @@ class CLIP(nn.Module):
         super().__init__()
         self.embed_dim = embed_dim
         self.context_length = context_length
-        vision_heads = vision_width // 64  # clip.py:60
+        vision_heads = max(1, vision_width // 64)  # clip.py:60
         self.vision_model = VisionTransformerBase(
             img_size=image_resolution,
             patch_size=vision_patch_size,

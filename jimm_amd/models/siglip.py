@@ -39,7 +39,7 @@ class SigLIP(nn.Module):
         layernorm_epsilon: float = 1e-6,
     ) -> None:
         super().__init__()
-        vision_heads = vision_heads or vision_width // 64  # siglip.py:59
+        vision_heads = vision_heads or max(1, vision_width // 64)  # siglip.py:59
         self.vision_model = VisionTransformerBase(
             img_size=image_resolution,
             patch_size=vision_patch_size,

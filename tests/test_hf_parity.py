@@ -296,3 +296,45 @@ def test_roundtrip_fuzz_configs(tmp_path):
         x = torch.randn(2, 3, img, img)
         with torch.no_grad():
             assert torch.allclose(m(x), m2(x), atol=1e-6)
+
+
+def test_bare_safetensors_shape_inference(tmp_path):
+    """A bare .safetensors file (no config.json) loads via shape inference
+    (reference models/vit.py:144-164, clip.py:208-247, siglip.py:193-207)."""
+    from safetensors.torch import load_file, save_file
+
+    # ViT
+    hf, d = _tiny_vit(tmp_path, old_keys=False)
+    bare = tmp_path / "vit_bare.safetensors"
+    save_file(load_file(d / "model.safetensors"), str(bare))
+    m = jimm_amd.VisionTransformer.from_pretrained(str(bare)).eval()
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        ref = hf(x).logits
+        out = m(x)
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+
+    # CLIP and SigLIP roundtrip through a bare file
+    clip = jimm_amd.CLIP(embed_dim=16, image_resolution=32, vision_layers=1,
+                         vision_width=64, vision_patch_size=16, context_length=5,
+                         vocab_size=33, transformer_width=64, transformer_heads=1,
+                         transformer_layers=1)
+    d2 = tmp_path / "clip"
+    clip.save_pretrained(str(d2))
+    bare2 = tmp_path / "clip_bare.safetensors"
+    save_file(load_file(d2 / "model.safetensors"), str(bare2))
+    clip2 = jimm_amd.CLIP.from_pretrained(str(bare2))
+    for k, v in clip.state_dict().items():
+        assert torch.equal(v, clip2.state_dict()[k]), k
+
+    sig = jimm_amd.SigLIP(image_resolution=32, vision_layers=1, vision_width=64,
+                          vision_patch_size=16, context_length=5, vocab_size=33,
+                          transformer_width=64, transformer_heads=1,
+                          transformer_layers=1, vision_heads=1)
+    d3 = tmp_path / "sig"
+    sig.save_pretrained(str(d3))
+    bare3 = tmp_path / "sig_bare.safetensors"
+    save_file(load_file(d3 / "model.safetensors"), str(bare3))
+    sig2 = jimm_amd.SigLIP.from_pretrained(str(bare3))
+    for k, v in sig.state_dict().items():
+        assert torch.equal(v, sig2.state_dict()[k]), k
