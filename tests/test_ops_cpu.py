@@ -205,3 +205,28 @@ def test_clip_loss_gather_flag_world1():
     a = L.clip_contrastive_loss(img, txt, s, gather=True)
     b = L.clip_contrastive_loss(img, txt, s, gather=False)
     assert torch.allclose(a, b)
+
+
+def test_synthetic_text_eot_invariant():
+    """CLIP pools at argmax(ids): the synthetic pipeline must place exactly
+    one EOS (= max id) per row so EOT pooling is well-defined."""
+    from jimm_amd.train import SyntheticImageText
+
+    data = SyntheticImageText(8, 32, 12, 100, torch.device("cpu"), seed=4)
+    _, ids = next(iter(data))
+    assert (ids == 99).sum(dim=1).eq(1).all()
+    assert ids.max() == 99
+
+
+def test_meter_jsonl(tmp_path):
+    import json
+
+    from jimm_amd.train import Meter
+
+    p = tmp_path / "m.jsonl"
+    m = Meter(str(p), rank=0)
+    m.log(1, loss=0.5, acc=0.9)
+    m.log(2, loss=0.4)
+    m.close()
+    recs = [json.loads(l) for l in open(p)]
+    assert recs[0]["loss"] == 0.5 and recs[1]["step"] == 2
