@@ -259,3 +259,7 @@ def _check_trainer_dp_invariance(rank, world):
 
 def test_trainer_dp_invariance():
     spawn(_check_trainer_dp_invariance, port=29517)
+
+
+def test_ddp_average_world4():
+    spawn(_check_ddp_average, world=4, port=29519)

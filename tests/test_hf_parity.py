@@ -338,3 +338,20 @@ def test_bare_safetensors_shape_inference(tmp_path):
     sig2 = jimm_amd.SigLIP.from_pretrained(str(bare3))
     for k, v in sig.state_dict().items():
         assert torch.equal(v, sig2.state_dict()[k]), k
+
+
+def test_pytorch_bin_format(tmp_path):
+    """pytorch_model.bin checkpoints load via use_pytorch=True (the
+    reference parametrizes safetensors vs torch-format — utils.py:56,70)."""
+    from safetensors.torch import load_file
+
+    hf, d = _tiny_vit(tmp_path, old_keys=False)
+    sd = load_file(d / "model.safetensors")
+    torch.save(sd, d / "pytorch_model.bin")
+    (d / "model.safetensors").unlink()
+    m = jimm_amd.VisionTransformer.from_pretrained(str(d), use_pytorch=True).eval()
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        ref = hf(x).logits
+        out = m(x)
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
