@@ -22,7 +22,6 @@ import torch.nn.functional as F
 import jimm_amd
 from jimm_amd.train import Meter, TrainConfig, Trainer, init_distributed
 
-EPOCHS = 5
 BATCH = 64
 LR = 1e-4          # reference: Adam 1e-4 (vit_training.py:202)
 IMG = 32           # digits 8x8 upscaled
@@ -44,6 +43,12 @@ def load_digits_batches(batch, rank, world, seed=0):
 
 
 def main():
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--epochs", type=int, default=5)
+    ap.add_argument("--jsonl", default=None, help="metrics JSONL output path")
+    args = ap.parse_args()
     rank, world, local_rank, device = init_distributed()
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     (tr_x, tr_y), (te_x, te_y) = load_digits_batches(BATCH, rank, world)
@@ -54,13 +59,13 @@ def main():
         hidden_size=128, num_heads=2, mlp_dim=256,
     ).to(device, dtype)
     trainer = Trainer(model, TrainConfig(task="vit", lr=LR))
-    meter = Meter(rank=rank)
+    meter = Meter(args.jsonl, rank=rank)
 
     n_train = len(tr_y)
     steps_per_epoch = n_train // (BATCH * world)
     g = torch.Generator().manual_seed(1)
     step = 0
-    for epoch in range(EPOCHS):
+    for epoch in range(args.epochs):
         perm = torch.randperm(n_train, generator=g)
         for i in range(steps_per_epoch):
             idx = perm[(i * world + rank) * BATCH : (i * world + rank + 1) * BATCH]
