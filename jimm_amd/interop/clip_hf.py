@@ -30,8 +30,12 @@ def _map_tower_layers(m: KeyMap, hf_prefix: str, native_prefix: str, num_layers:
 
 def _parse_config(cfg: dict) -> dict:
     tc, vc = cfg["text_config"], cfg["vision_config"]
+    act = (tc.get("hidden_act") or "quick_gelu")
+    hidden_act = {"quick_gelu": "quickgelu", "gelu": "gelu",
+                  "gelu_new": "gelu_tanh", "gelu_pytorch_tanh": "gelu_tanh"}.get(act, "quickgelu")
     return dict(
         embed_dim=cfg.get("projection_dim", 512),
+        hidden_act=hidden_act,
         image_resolution=vc.get("image_size", 224),
         vision_layers=vc["num_hidden_layers"],
         vision_width=vc["hidden_size"],

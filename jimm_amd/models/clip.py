@@ -37,6 +37,7 @@ class CLIP(nn.Module):
         vision_mlp_dim: int | None = None,
         transformer_mlp_dim: int | None = None,
         *,
+        hidden_act: str = "quickgelu",  # OpenAI CLIP; LAION variants use "gelu"
         layernorm_epsilon: float = 1e-5,
     ) -> None:
         super().__init__()
@@ -52,7 +53,7 @@ class CLIP(nn.Module):
             mlp_dim=vision_mlp_dim or vision_width * 4,
             use_pre_norm=True,
             use_patch_bias=False,
-            hidden_act="quickgelu",
+            hidden_act=hidden_act,
             pooling="CLS",
             layernorm_epsilon=layernorm_epsilon,
         )
@@ -66,7 +67,7 @@ class CLIP(nn.Module):
             mlp_dim=transformer_mlp_dim or transformer_width * 4,
             causal=True,
             pooling="EOT",
-            hidden_act="quickgelu",
+            hidden_act=hidden_act,
             layernorm_epsilon=layernorm_epsilon,
         )
         self.text_projection = nn.Linear(transformer_width, embed_dim, bias=False)

@@ -355,3 +355,31 @@ def test_pytorch_bin_format(tmp_path):
         ref = hf(x).logits
         out = m(x)
     assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+
+
+def test_clip_gelu_variant(tmp_path):
+    """LAION-style CLIP configs (hidden_act='gelu') load with plain GELU
+    instead of the OpenAI QuickGELU default."""
+    from transformers import CLIPConfig, CLIPModel, CLIPTextConfig, CLIPVisionConfig
+
+    vc = CLIPVisionConfig(hidden_size=64, num_hidden_layers=1, num_attention_heads=1,
+                          intermediate_size=128, image_size=32, patch_size=16,
+                          hidden_act="gelu")
+    tc = CLIPTextConfig(hidden_size=32, num_hidden_layers=1, num_attention_heads=2,
+                        intermediate_size=64, max_position_embeddings=6, vocab_size=50,
+                        hidden_act="gelu", bos_token_id=0, eos_token_id=49)
+    hf = CLIPModel(CLIPConfig(text_config=tc.to_dict(), vision_config=vc.to_dict(),
+                              projection_dim=16)).eval()
+    for p in hf.parameters():
+        p.data.normal_(0, 0.02)
+    d = tmp_path / "clip_gelu"
+    hf.save_pretrained(d, safe_serialization=True)
+    m = jimm_amd.CLIP.from_pretrained(str(d)).eval()
+    assert m.vision_model.encoder.layers[0].act == "gelu"
+    img = torch.randn(2, 3, 32, 32)
+    ids = torch.randint(0, 49, (2, 6))
+    ids[:, -1] = 49
+    with torch.no_grad():
+        ref = hf(pixel_values=img, input_ids=ids).logits_per_image
+        out, _ = m(img, ids)
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
