@@ -263,3 +263,7 @@ def test_trainer_dp_invariance():
 
 def test_ddp_average_world4():
     spawn(_check_ddp_average, world=4, port=29519)
+
+
+def test_tensor_parallel_world4():
+    spawn(_check_tensor_parallel, world=4, port=29521)
