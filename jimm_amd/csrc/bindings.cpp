@@ -11,6 +11,8 @@ torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, s
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act);
 torch::Tensor colsum(torch::Tensor dz);
 torch::Tensor im2col_patch(torch::Tensor img, int64_t patch);
+torch::Tensor cls_pos_fwd(torch::Tensor x, c10::optional<torch::Tensor> cls, torch::Tensor pos);
+torch::Tensor embed_pos_fwd(torch::Tensor ids, torch::Tensor emb, torch::Tensor pos);
 torch::Tensor col2im_patch(torch::Tensor cols, std::vector<int64_t> img_shape, int64_t patch);
 void adam_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
@@ -55,6 +57,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_bwd", &act_bwd, "activation backward: dy * act'(z)");
   m.def("colsum", &colsum, "column sum -> fp32 (bias gradient, K15)");
   m.def("im2col_patch", &im2col_patch, "patch-embed unfold (K1)");
+  m.def("cls_pos_fwd", &cls_pos_fwd, "fused CLS concat + pos-emb add (K2)");
+  m.def("embed_pos_fwd", &embed_pos_fwd, "fused token-embed gather + pos-emb add (K10)");
   m.def("col2im_patch", &col2im_patch, "patch-embed fold backward (K1/K15)");
   m.def("adam_step", &adam_step, "fused multi-tensor Adam (K14)");
   m.def("adam_prepare", &adam_prepare, "build device chunk descriptors once (K14)");

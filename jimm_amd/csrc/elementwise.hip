@@ -160,6 +160,58 @@ __global__ void col2im_kernel(const T* __restrict__ cols, T* __restrict__ img, i
   }
 }
 
+// K2: y[b,0,:] = cls + pos[0]; y[b,1+l,:] = x[b,l,:] + pos[1+l,:]
+// (CLS-token concat + position-embedding add in one pass; without cls it is
+// a plain broadcast pos add). Reference: common/vit.py:232-241.
+template <typename T, bool HAS_CLS>
+__global__ void cls_pos_kernel(const T* __restrict__ x, const T* __restrict__ cls,
+                               const T* __restrict__ pos, T* __restrict__ y, int B,
+                               int Lout, int Hv) {
+  // Hv = H/8 vector groups; one thread handles one 8-group
+  constexpr int V = 8;
+  const int64_t total = (int64_t)B * Lout * Hv;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int hv = (int)(idx % Hv);
+    const int l = (int)((idx / Hv) % Lout);
+    const int64_t b = idx / ((int64_t)Hv * Lout);
+    float pv[V], sv[V], o[V];
+    vload_f32<V>(pos + ((int64_t)l * Hv + hv) * V, pv);
+    if (HAS_CLS && l == 0) {
+      vload_f32<V>(cls + (int64_t)hv * V, sv);
+    } else {
+      const int lx = HAS_CLS ? l - 1 : l;
+      vload_f32<V>(x + ((b * (int64_t)(Lout - (HAS_CLS ? 1 : 0)) + lx) * Hv + hv) * V, sv);
+    }
+#pragma unroll
+    for (int j = 0; j < V; ++j) o[j] = sv[j] + pv[j];
+    vstore_f32<V>(y + (idx * V), o);
+  }
+}
+
+// K10: y[b,l,:] = emb[ids[b,l],:] + pos[l,:] — token-embedding gather fused
+// with the position add. Reference: clip.py:159-160, siglip.py:146-147.
+template <typename T>
+__global__ void embed_pos_kernel(const int64_t* __restrict__ ids, const T* __restrict__ emb,
+                                 const T* __restrict__ pos, T* __restrict__ y, int64_t BL,
+                                 int L, int Hv) {
+  constexpr int V = 8;
+  const int64_t total = BL * Hv;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int hv = (int)(idx % Hv);
+    const int64_t bl = idx / Hv;
+    const int l = (int)(bl % L);
+    const int64_t tok = ids[bl];
+    float ev[V], pv[V], o[V];
+    vload_f32<V>(emb + (tok * Hv + hv) * V, ev);
+    vload_f32<V>(pos + ((int64_t)l * Hv + hv) * V, pv);
+#pragma unroll
+    for (int j = 0; j < V; ++j) o[j] = ev[j] + pv[j];
+    vstore_f32<V>(y + idx * V, o);
+  }
+}
+
 int act_code(const std::string& act) {
   if (act.empty()) return ACT_NONE;
   if (act == "gelu") return ACT_GELU;
@@ -280,6 +332,72 @@ torch::Tensor colsum(torch::Tensor dz) {
     TORCH_CHECK(false, "colsum: unsupported dtype");
   }
   return ws.sum(0);  // (gy*rstep, N) fp32 reduce — tiny
+}
+
+torch::Tensor cls_pos_fwd(torch::Tensor x, c10::optional<torch::Tensor> cls,
+                          torch::Tensor pos) {
+  // x (B,L,H); cls (1,1,H) optional; pos (1,Lout,H) with Lout = L + (cls?1:0)
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const int B = x.size(0), L = x.size(1), H = x.size(2);
+  TORCH_CHECK(H % 8 == 0, "cls_pos: H % 8 != 0");
+  const int Lout = L + (cls ? 1 : 0);
+  auto posc = pos.contiguous();
+  TORCH_CHECK(posc.numel() >= (int64_t)Lout * H, "cls_pos: pos too short");
+  TORCH_CHECK(posc.scalar_type() == x.scalar_type());
+  auto y = torch::empty({B, Lout, H}, x.options());
+  const int64_t n = y.numel() / 8;
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n + block - 1) / block, kMaxGrid);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto t_tag) {
+    using T = decltype(t_tag);
+    const T* clsp = cls ? reinterpret_cast<const T*>(cls->contiguous().data_ptr()) : nullptr;
+    if (cls)
+      hipLaunchKernelGGL((cls_pos_kernel<T, true>), dim3(grid), dim3(block), 0, stream,
+                         reinterpret_cast<const T*>(x.data_ptr()), clsp,
+                         reinterpret_cast<const T*>(posc.data_ptr()),
+                         reinterpret_cast<T*>(y.data_ptr()), B, Lout, H / 8);
+    else
+      hipLaunchKernelGGL((cls_pos_kernel<T, false>), dim3(grid), dim3(block), 0, stream,
+                         reinterpret_cast<const T*>(x.data_ptr()), clsp,
+                         reinterpret_cast<const T*>(posc.data_ptr()),
+                         reinterpret_cast<T*>(y.data_ptr()), B, Lout, H / 8);
+  };
+  if (x.scalar_type() == torch::kBFloat16) launch(bf16{});
+  else if (x.scalar_type() == torch::kFloat32) launch(0.f);
+  else TORCH_CHECK(false, "cls_pos: unsupported dtype");
+  return y;
+}
+
+torch::Tensor embed_pos_fwd(torch::Tensor ids, torch::Tensor emb, torch::Tensor pos) {
+  // ids (B,L) int64; emb (Vocab,H); pos (L_max,H) -> y (B,L,H)
+  TORCH_CHECK(ids.is_cuda() && ids.scalar_type() == torch::kInt64);
+  auto idc = ids.contiguous();
+  const int64_t BL = idc.numel();
+  const int L = ids.size(-1);
+  const int H = emb.size(-1);
+  TORCH_CHECK(H % 8 == 0, "embed_pos: H % 8 != 0");
+  auto embc = emb.contiguous();
+  auto posc = pos.contiguous();
+  TORCH_CHECK(embc.scalar_type() == posc.scalar_type());
+  auto sizes = ids.sizes().vec();
+  sizes.push_back(H);
+  auto y = torch::empty(sizes, embc.options());
+  const int64_t n = BL * (H / 8);
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n + block - 1) / block, kMaxGrid);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto t_tag) {
+    using T = decltype(t_tag);
+    hipLaunchKernelGGL((embed_pos_kernel<T>), dim3(grid), dim3(block), 0, stream,
+                       idc.data_ptr<int64_t>(), reinterpret_cast<const T*>(embc.data_ptr()),
+                       reinterpret_cast<const T*>(posc.data_ptr()),
+                       reinterpret_cast<T*>(y.data_ptr()), BL, L, H / 8);
+  };
+  if (embc.scalar_type() == torch::kBFloat16) launch(bf16{});
+  else if (embc.scalar_type() == torch::kFloat32) launch(0.f);
+  else TORCH_CHECK(false, "embed_pos: unsupported dtype");
+  return y;
 }
 
 torch::Tensor im2col_patch(torch::Tensor img, int64_t patch) {

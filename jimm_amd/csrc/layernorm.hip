@@ -62,7 +62,7 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
   }
 }
 
-template <typename T, int VEC, bool HAS_ADD>
+template <typename T, int VEC, bool HAS_ADD, bool REGCACHE = true>
 __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                               const float* __restrict__ w, const float* __restrict__ mean,
                               const float* __restrict__ rstd, const T* __restrict__ addend,
@@ -92,7 +92,7 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   constexpr int MAXC = 24;            // supports H <= WAVE*4*6 (= 1536)
   constexpr int MAXCH = MAXC / VEC;
   const int nchunk = H / (WAVE * VEC);
-  if constexpr (VEC == 4) {
+  if constexpr (VEC == 4 && REGCACHE) {
   float wv[MAXC];
 #pragma unroll
   for (int c = 0; c < MAXCH; ++c) {
@@ -245,14 +245,23 @@ void launch_ln_bwd(const T* dy, const T* x, const float* w, const float* mean,
   const int block = 256;
   const int waves_per_block = block / WAVE;
   size_t shmem = 2 * (size_t)waves_per_block * H * sizeof(float);
+  // H > 1536 exceeds the VEC==4 register cache; route through the streaming
+  // 2-pass body instead of erroring (hidden 1664 / ViT-G class).
+  const bool regcache = H <= WAVE * 24;
   auto pick = [&](auto vec_tag) {
     constexpr int V = decltype(vec_tag)::value;
-    if (addend)
-      hipLaunchKernelGGL((ln_bwd_kernel<T, V, true>), dim3(grid), dim3(block), shmem,
-                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, ws, nrows, H);
-    else
-      hipLaunchKernelGGL((ln_bwd_kernel<T, V, false>), dim3(grid), dim3(block), shmem,
-                         stream, dy, x, w, mean, rstd, addend, dx, dw, db, ws, nrows, H);
+    auto pick2 = [&](auto add_tag, auto reg_tag) {
+      hipLaunchKernelGGL(
+          (ln_bwd_kernel<T, V, decltype(add_tag)::value, decltype(reg_tag)::value>),
+          dim3(grid), dim3(block), shmem, stream, dy, x, w, mean, rstd, addend, dx, dw,
+          db, ws, nrows, H);
+    };
+    using tt = std::true_type;
+    using ft = std::false_type;
+    if (addend && regcache) pick2(tt{}, tt{});
+    else if (addend) pick2(tt{}, ft{});
+    else if (regcache) pick2(ft{}, tt{});
+    else pick2(ft{}, ft{});
   };
   if (H % (WAVE * 4) == 0) pick(std::integral_constant<int, 4>{});
   else if (H % (WAVE * 2) == 0) pick(std::integral_constant<int, 2>{});
@@ -295,7 +304,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
   // gradient of a pre-LN transformer block lands here for free
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
   const int H = x.size(-1);
-  TORCH_CHECK(H <= WAVE * 24, "layernorm_bwd: H must be <= 1536 (register cache)");
+  // H <= 1536 runs the one-pass register-cached body; larger H streams
+  // 2-pass. The per-wave LDS partial slabs (2*4*H floats) bound H at 5120.
+  TORCH_CHECK(H <= 5120, "layernorm_bwd: H must be <= 5120 (LDS partial slabs)");
   const int64_t nrows = x.numel() / H;
   auto wf = w.contiguous().to(torch::kFloat32);
   auto dx = torch::empty_like(x);

@@ -113,6 +113,10 @@ __global__ void sigmoid_loss_ew_kernel(const float* __restrict__ logits,
                                        int cols, int64_t diag0) {
   // z = +1 where col == diag0 + row else -1
   // loss += -logsigmoid(z * x); dloss/dx = -z * sigmoid(-z * x)
+  // Per-block partial sums land in loss_parts[blockIdx.x] (fixed in-block
+  // reduce order); the host sums them in fixed order — the loss value is
+  // deterministic without atomics.
+  __shared__ float s_acc[4];
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int wpb = blockDim.x / WAVE;
@@ -129,7 +133,13 @@ __global__ void sigmoid_loss_ew_kernel(const float* __restrict__ logits,
     }
   }
   acc = wave_reduce_sum(acc);
-  if (lane == 0) atomicAdd(loss_parts, acc);
+  if (lane == 0) s_acc[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.f;
+    for (int wv = 0; wv < wpb; ++wv) total += s_acc[wv];
+    loss_parts[blockIdx.x] = total;
+  }
 }
 
 }  // namespace
@@ -207,11 +217,12 @@ std::vector<torch::Tensor> sigmoid_loss_ew(torch::Tensor logits, int64_t diag0) 
   const int64_t rows = logits.size(0);
   const int cols = logits.size(1);
   auto dlogits = torch::empty_like(logits);
-  auto loss = torch::zeros({1}, logits.options());
   const int grid = (int)std::min<int64_t>((rows + 3) / 4, 2048);
+  auto parts = torch::empty({(int64_t)grid}, logits.options());
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(sigmoid_loss_ew_kernel, dim3(grid), dim3(256), 0, stream,
                      logits.data_ptr<float>(), dlogits.data_ptr<float>(),
-                     loss.data_ptr<float>(), rows, cols, diag0);
+                     parts.data_ptr<float>(), rows, cols, diag0);
+  auto loss = parts.sum().reshape({1});  // fixed-order reduce: deterministic
   return {loss, dlogits};
 }

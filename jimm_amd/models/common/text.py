@@ -60,7 +60,7 @@ class TextTransformer(nn.Module):
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         """input_ids (B, L) -> pooled (B, H)."""
-        x = self.token_embedding(input_ids) + self.pos_embedding[:, : input_ids.shape[1]]  # K10
+        x = ops.embed_pos(input_ids, self.token_embedding.weight, self.pos_embedding)  # K10 fused
         x = self.encoder(x)
         x = ops.layer_norm(x, self.ln_final.weight, self.ln_final.bias, self.eps)
         if self.pooling == "EOT":

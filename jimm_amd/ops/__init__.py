@@ -2,6 +2,7 @@ from jimm_amd.ops.functional import (  # noqa: F401
     add_cls_pos,
     attention,
     attention_qkv,
+    embed_pos,
     layer_norm,
     linear,
     patch_embed,
@@ -16,6 +17,7 @@ __all__ = [
     "add_cls_pos",
     "attention",
     "attention_qkv",
+    "embed_pos",
     "layer_norm",
     "linear",
     "patch_embed",

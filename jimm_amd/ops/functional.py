@@ -278,19 +278,80 @@ def patch_embed(
     return y.flatten(2).transpose(1, 2)  # (B, n_patches, hidden)
 
 
+class _ClsPosFn(torch.autograd.Function):
+    """K2 fused on GPU: one kernel does CLS-tile + concat + pos add."""
+
+    @staticmethod
+    def forward(ctx, x, cls_token, pos_emb):
+        ctx.has_cls = cls_token is not None
+        cls = cls_token.to(x.dtype).view(1, 1, -1) if cls_token is not None else None
+        Lout = x.shape[1] + (1 if ctx.has_cls else 0)
+        pos = pos_emb[0, :Lout].to(x.dtype)
+        return _backend.ext().cls_pos_fwd(x.contiguous(), cls, pos)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.has_cls:
+            dx = dy[:, 1:].contiguous()
+            dcls = dy[:, :1].sum(dim=0, keepdim=True)
+        else:
+            dx, dcls = dy, None
+        dpos = dy.sum(dim=0, keepdim=True)
+        return dx, dcls, dpos
+
+
 def add_cls_pos(
     x: torch.Tensor,
     cls_token: torch.Tensor | None,
     pos_emb: torch.Tensor,
 ) -> torch.Tensor:
-    """[CLS concat] + position-embedding add (K2): pure bandwidth, fused on GPU.
+    """[CLS concat] + position-embedding add (K2): pure bandwidth.
 
+    On GPU a single HIP kernel (csrc/elementwise.hip cls_pos_kernel) tiles the
+    CLS token, concatenates, and adds the position embedding in one pass; the
+    CPU path is the eager cat+add oracle.
     Reference semantics: /root/reference/src/jimm/common/vit.py:232-241.
     """
+    if _backend.use_hip(x) and x.shape[-1] % 8 == 0:
+        return _ClsPosFn.apply(x, cls_token, pos_emb)
     if cls_token is not None:
         cls = cls_token.expand(x.shape[0], -1, -1).to(x.dtype)
         x = torch.cat([cls, x], dim=1)
     return x + pos_emb[:, : x.shape[1]].to(x.dtype)
+
+
+class _EmbedPosFn(torch.autograd.Function):
+    """K10 fused on GPU: token-embedding gather + pos-emb add in one kernel.
+
+    Reference semantics: /root/reference/src/jimm/models/clip.py:159-160,
+    siglip.py:146-147.
+    """
+
+    @staticmethod
+    def forward(ctx, ids, emb, pos_emb):
+        ctx.save_for_backward(ids)
+        ctx.vocab = emb.shape[0]
+        pos = pos_emb[0, : ids.shape[-1]].to(emb.dtype)
+        return _backend.ext().embed_pos_fwd(ids, emb, pos)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (ids,) = ctx.saved_tensors
+        demb = None
+        if ctx.needs_input_grad[1]:
+            demb = torch.zeros(
+                ctx.vocab, dy.shape[-1], dtype=dy.dtype, device=dy.device
+            )
+            demb.index_add_(0, ids.reshape(-1), dy.reshape(-1, dy.shape[-1]))
+        dpos = dy.sum(dim=0).unsqueeze(0) if ctx.needs_input_grad[2] else None
+        return None, demb, dpos
+
+
+def embed_pos(ids: torch.Tensor, emb: torch.Tensor, pos_emb: torch.Tensor) -> torch.Tensor:
+    """Token-embedding lookup + position-embedding add (K10), fused on GPU."""
+    if _backend.use_hip(emb) and emb.shape[-1] % 8 == 0:
+        return _EmbedPosFn.apply(ids, emb, pos_emb)
+    return torch.nn.functional.embedding(ids, emb) + pos_emb[:, : ids.shape[-1]].to(emb.dtype)
 
 
 __all__ = [
@@ -301,4 +362,5 @@ __all__ = [
     "linear",
     "patch_embed",
     "add_cls_pos",
+    "embed_pos",
 ]

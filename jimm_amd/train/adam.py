@@ -33,8 +33,11 @@ class Adam(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self.master_weights = master_weights
         # graph-capturable fast path: chunk descriptors uploaded once, lr and
-        # step live on-device (csrc/adam.hip adam_prepare/adam_apply)
-        self._prepared = None  # (desc_dev, nchunks, lr_dev, step_dev)
+        # step live on-device (csrc/adam.hip adam_prepare/adam_apply).
+        # Keyed per param group: each group gets its own descriptor table,
+        # lr and device-side step counter.
+        self._prepared: dict[int, tuple] = {}
+        self._last_lr: dict[int, float] = {}
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -42,7 +45,7 @@ class Adam(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
             lr, (b1, b2), eps, wd = group["lr"], group["betas"], group["eps"], group["weight_decay"]
             ps, gs, ms, vs, masters = [], [], [], [], []
             step_t = None
@@ -67,7 +70,7 @@ class Adam(torch.optim.Optimizer):
                 continue
             if ps[0].is_cuda and not _backend.force_eager():
                 ext = _backend.ext()
-                if self._prepared is None:
+                if gi not in self._prepared:
                     if step_t == 1:
                         # first step initializes state lazily; take it eagerly
                         # then freeze the descriptor table (pointers stable)
@@ -78,32 +81,44 @@ class Adam(torch.optim.Optimizer):
                     # apply computes bias correction for the right step
                     seed = step_t if step_t == 1 else step_t - 1
                     step_dev = torch.full((1,), seed, dtype=torch.int32, device=ps[0].device)
-                    self._prepared = (desc, int(nchunks.item()), lr_dev, step_dev)
+                    self._prepared[gi] = (desc, int(nchunks.item()), lr_dev, step_dev)
                     if step_t == 1:
                         continue
-                desc, nchunks, lr_dev, step_dev = self._prepared
-                if lr != self._last_lr:
+                desc, nchunks, lr_dev, step_dev = self._prepared[gi]
+                if lr != self._last_lr.get(gi):
                     lr_dev.fill_(lr)
-                    self._last_lr = lr
+                    self._last_lr[gi] = lr
                 ext.adam_apply(desc, nchunks, lr_dev, step_dev, b1, b2, eps, wd)
             else:
                 self._ref_step(ps, gs, ms, vs, masters, lr, b1, b2, eps, wd, step_t)
         return loss
 
-    _last_lr: float = float("nan")
-
     def load_state_dict(self, state_dict) -> None:
         super().load_state_dict(state_dict)
-        self._prepared = None  # moment tensors were replaced: descriptors stale
+        # The base class casts loaded state to the param dtype; for bf16
+        # params that would silently demote the fp32 moments/master weights
+        # (and break the fused kernel's data_ptr<float>()). Restore fp32.
+        for group in self.param_groups:
+            for p in group["params"]:
+                st = self.state.get(p)
+                if not st:
+                    continue
+                for key in ("m", "v", "master"):
+                    if key in st and st[key] is not None and st[key].dtype != torch.float32:
+                        st[key] = st[key].float()
+        self._prepared = {}  # moment tensors were replaced: descriptors stale
+        self._last_lr = {}
 
     @torch.no_grad()
     def sync_step_from_device(self) -> None:
         """Copy the on-device step counter back into per-param state (needed
         after graph-replayed training, where host counters do not advance)."""
-        if self._prepared is None:
+        if not self._prepared:
             return
-        step = int(self._prepared[3].item())
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
+            if gi not in self._prepared:
+                continue
+            step = int(self._prepared[gi][3].item())
             for p in group["params"]:
                 if self.state[p]:
                     self.state[p]["step"] = step
