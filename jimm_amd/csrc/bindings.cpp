@@ -43,6 +43,7 @@ torch::Tensor xent_rows_bwd(torch::Tensor logits, torch::Tensor labels, torch::T
                             double gscale);
 std::vector<torch::Tensor> sigmoid_loss_ew(torch::Tensor logits, int64_t diag0);
 bool gemm_dw_supported(int64_t M, int64_t N, int64_t K);
+bool gemm8p_supported(int64_t M, int64_t N, int64_t K);
 torch::Tensor gemm_tn_splitk(torch::Tensor dz, torch::Tensor x);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias, std::string act,
@@ -72,6 +73,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe32", &mfma_probe32, "MFMA 32x32x16 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
   m.def("gemm_dw_supported", &gemm_dw_supported, "split-K dW GEMM shape check");
+  m.def("gemm8p_supported", &gemm8p_supported, "8-phase 256-tile GEMM shape check");
   m.def("l2norm_fwd", &l2norm_fwd, "row L2-normalize forward (K12)");
   m.def("l2norm_bwd", &l2norm_bwd, "row L2-normalize backward (K12)");
   m.def("xent_rows_fwd", &xent_rows_fwd, "fused softmax-CE forward (K13)");

@@ -56,13 +56,34 @@ __device__ __forceinline__ float wave_reduce_max(float v) {
 #define ACT_GELU_TANH 2
 #define ACT_QUICKGELU 3
 
+// Fast erf (Abramowitz-Stegun 7.1.26, |err| <= 1.5e-7 — below bf16
+// resolution): ~12 VALU + 1 v_rcp + 1 v_exp vs libm erff's branchy
+// polynomial, which measured ~75 lane-cycles/element as a GEMM epilogue.
+__device__ __forceinline__ float fast_erff(float x) {
+  const float ax = fabsf(x);
+  const float t = __frcp_rn(1.0f + 0.3275911f * ax);
+  float p = 1.061405429f;
+  p = p * t - 1.453152027f;
+  p = p * t + 1.421413741f;
+  p = p * t - 0.284496736f;
+  p = p * t + 0.254829592f;
+  const float e = 1.0f - p * t * __expf(-ax * ax);
+  return copysignf(e, x);
+}
+
+__device__ __forceinline__ float fast_tanhf(float x) {
+  // tanh(x) = 1 - 2/(exp(2x)+1); clamp avoids exp overflow (|x|>10 -> +-1)
+  const float cx = fminf(fmaxf(x, -10.0f), 10.0f);
+  return 1.0f - 2.0f / (__expf(2.0f * cx) + 1.0f);
+}
+
 __device__ __forceinline__ float act_fwd(float x, int act) {
   switch (act) {
     case ACT_GELU:
-      return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f));
+      return 0.5f * x * (1.0f + fast_erff(x * 0.70710678118654752440f));
     case ACT_GELU_TANH: {
       float x3 = x * x * x;
-      float t = tanhf(0.7978845608028654f * (x + 0.044715f * x3));
+      float t = fast_tanhf(0.7978845608028654f * (x + 0.044715f * x3));
       return 0.5f * x * (1.0f + t);
     }
     case ACT_QUICKGELU: {
@@ -78,14 +99,14 @@ __device__ __forceinline__ float act_grad(float x, int act) {
   switch (act) {
     case ACT_GELU: {
       // d/dx [x * Phi(x)] = Phi(x) + x * phi(x)
-      float cdf = 0.5f * (1.0f + erff(x * 0.70710678118654752440f));
+      float cdf = 0.5f * (1.0f + fast_erff(x * 0.70710678118654752440f));
       float pdf = 0.3989422804014327f * __expf(-0.5f * x * x);
       return cdf + x * pdf;
     }
     case ACT_GELU_TANH: {
       float x2 = x * x;
       float inner = 0.7978845608028654f * (x + 0.044715f * x * x2);
-      float t = tanhf(inner);
+      float t = fast_tanhf(inner);
       float dinner = 0.7978845608028654f * (1.0f + 3.0f * 0.044715f * x2);
       return 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * dinner;
     }

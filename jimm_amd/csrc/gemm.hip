@@ -250,11 +250,25 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 
 }  // namespace
 
-// 256x256-tile kernel (gemm256.hip) — used when the shape tiles evenly
+// 256x256-tile kernels: 8-phase (gemm8p.hip, the default — deep-pipelined
+// counted-vmcnt schedule) and the older 2-phase (gemm256.hip, kept for A/B
+// via JIMM_AMD_GEMM_TILE=256).
 bool gemm256_supported(int64_t M, int64_t N, int64_t K);
 void gemm_nt_256(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias_f32,
                  std::string act, c10::optional<torch::Tensor> residual, torch::Tensor y,
                  c10::optional<torch::Tensor> z);
+bool gemm8p_supported(int64_t M, int64_t N, int64_t K);
+void gemm_nt_8p(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias_f32,
+                std::string act, c10::optional<torch::Tensor> residual, torch::Tensor y,
+                c10::optional<torch::Tensor> z);
+
+static int gemm_tile_mode() {
+  // 0 = auto (8-phase when supported), 256 = force 2-phase 256, 128 = force
+  // the 128-tile kernel (debug/A-B only)
+  const char* env = getenv("JIMM_AMD_GEMM_TILE");
+  if (!env) return 0;
+  return atoi(env);
+}
 
 bool gemm_supported(int64_t M, int64_t N, int64_t K, std::string dtype) {
   if (dtype != "torch.bfloat16") return false;
@@ -282,7 +296,15 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
   else if (act == "quickgelu") act_code = ACT_QUICKGELU;
   else TORCH_CHECK(act.empty(), "unknown act ", act);
 
-  if (gemm256_supported(M, N, K)) {
+  const int tile_mode = gemm_tile_mode();
+  if (tile_mode == 0 && gemm8p_supported(M, N, K)) {
+    c10::optional<torch::Tensor> zopt;
+    if (save_z) zopt = z;
+    gemm_nt_8p(x, w, bf, act, residual, y, zopt);
+    if (save_z) return {y, z};
+    return {y, torch::Tensor()};
+  }
+  if (tile_mode != 128 && gemm256_supported(M, N, K)) {
     c10::optional<torch::Tensor> zopt;
     if (save_z) zopt = z;
     gemm_nt_256(x, w, bf, act, residual, y, zopt);

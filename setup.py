@@ -31,6 +31,7 @@ setup(
                 "jimm_amd/csrc/attention_bwd_fused.hip",
                 "jimm_amd/csrc/gemm.hip",
                 "jimm_amd/csrc/gemm256.hip",
+                "jimm_amd/csrc/gemm8p.hip",
                 "jimm_amd/csrc/gemm_dw.hip",
                 "jimm_amd/csrc/losses.hip",
                 "jimm_amd/csrc/probe.hip",
