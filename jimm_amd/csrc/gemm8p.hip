@@ -1,35 +1,48 @@
 // K4/K6/K7/K8 — 8-phase 256x256-tile MFMA bf16 GEMM with fused epilogue.
 //
-// Implements the CDNA4 guide's deep-pipelined 256-square structure
-// (cdna_hip_programming.md §5 "The 256² 8-phase template"): the K-step is
-// split into 4 sub-phases per K-tile, each {12x ds_read_b128 fragment loads |
-// 1 half-tile global_load_lds prefetch | counted s_waitcnt vmcnt(4) | raw
-// s_barrier | setprio(1) 16x mfma_f32_16x16x32_bf16 setprio(0) | raw
-// s_barrier}.  The counted vmcnt (never 0 in the main loop) keeps 2 staged
-// half-tiles in flight ACROSS barriers — the 2-phase vmcnt(0) structure in
-// gemm256.hip drains the glds queue at every barrier, which is its ~900 TF
-// structural ceiling; this schedule removes that drain (T3+T4, +28-41%, and
-// enables T5 setprio, +21-25%).
+// Deep-pipelined 256-square structure after the CDNA4 guide's 8-phase
+// template (cdna_hip_programming.md §5): four sub-phases per K-tile, ONE raw
+// s_barrier per phase (a wave's next-phase fragment reads overlap another
+// wave's MFMA cluster), counted s_waitcnt vmcnt (never 0 in the main loop)
+// so staged half-tiles stay in flight ACROSS barriers, s_setprio(1) around
+// each 16-MFMA cluster (T5).
 //
-// Geometry: 256x256 tile, BK=64, 512 threads = 8 waves.  Per phase p
-// (quadrant (pr,pc) = (p>>1, p&1)) wave (wm in {0,1}, wn in 0..3) computes
-// the 64x32 strip rows [pr*128+wm*64, +64) x cols [pc*128+wn*32, +32): the
-// quadrant walk makes phases 0-1 consume only A-half0 / phases 2-3 A-half1
-// (and B-half pc), so a half staged at phase f is first read 4 phases later
-// and vmcnt(4) (= 2 half-tiles x 2 glds/wave in flight) certifies it.
+// Phase walk (kh = K-half of the 64-deep K-tile, pr = row-half):
+//   p0=(kh0,pr0)  p1=(kh0,pr1)  p2=(kh1,pr0)  p3=(kh1,pr1)
+// Wave (wm,wn) computes strip rows [pr*128+wm*64,+64) x cols [wn*64,+64) at
+// K-depth 32 per phase: 4 A-frags + 4 B-frags ds_read_b128, 16 MFMA.  The
+// B-fragments of a kh are REUSED in registers across the pr pair, so a
+// K-tile costs 24 fragment reads per wave — the information-theoretic
+// minimum (16 KiB A + 8 KiB B at 1 KiB per b128) — where a (row,col)
+// quadrant walk costs 48.
 //
-// Stage schedule (tile t, phases p0..p3 stage): p0 -> B-half0(t+1),
-// p1 -> B-half1(t+1), p2 -> A-half1(t+1), p3 -> A-half0(t+2).
-// Slot reuse is safe: e.g. A-half0(t+2) lands in the buffer A[t&1][0] whose
-// last reader was phase p1 of tile t (>= 2 barriers earlier).
+// Staging (one half-image per phase; 2 glds per wave):
+//   p0 -> B-half0(t+1)   p1 -> B-half1(t+1)   p2 -> A-half0(t+1)
+//   p3 -> A-half1(t+1)
+// A has 2x2 buffer slots (tile parity); B-half slots cycle mod 3 because B
+// is read in EVERY phase (a 2-slot B would be overwritten one barrier after
+// its last read — a DMA-vs-pending-ds_read race).  All slot-reuse gaps are
+// >= 2 barriers; a wave's reads complete (hipcc lgkmcnt) before its own
+// MFMA, hence before its next barrier arrival.  LDS = 4x16 KiB (A) +
+// 6x16 KiB (B) = 160 KiB (the full CU LDS; occupancy 1 block/CU, 2
+// waves/SIMD — the template's regime).
 //
-// LDS: per operand 2 buffers x 2 half-images of [128 rows][64 shorts]
-// (16 KiB each, XOR-swizzled chunk^=(row&7) with the inverse swizzle on the
-// per-lane glds SOURCE address — §5.4 rule 21) = 128 KiB total.
+// Certification (reads at phase f follow the barrier at f-1, which follows
+// every wave's counted vmcnt): vmcnt allowances {p0:2, p1:4, p2:4, p3:2}
+// keep 1-2 half-tiles in flight and cover each half's stage->first-read lag
+// (A0 staged t.p2 first read t+1.p0; A1 t.p3 -> t+1.p1; B t.p0/p1 ->
+// t+1.p0).  The last tile's p0 drains with vmcnt(0) (its stages were the
+// newest and the stream has ended).
 //
-// Epilogue fuses bias + gelu/gelu_tanh/quickgelu (+residual, +save-pre) like
-// gemm256.hip.  M may be ragged (MGUARD clamps staging rows and predicates
-// stores); N%256==0 and K%64==0, K>=128 required (gemm8p_supported).
+// LDS image swizzle: byte(row, c16) = row*128 + (c16 ^ (row&7))*16 —
+// verified conflict-free for this read pattern (SQ_LDS_BANK_CONFLICT = 0,
+// profiles/r02: the ds_read_b128 16-lane groups mix chunk indices, so the
+// row-XOR spreads them over all 16 slots).  glds writes lane-linear; the
+// inverse permutation goes on the per-lane SOURCE address (§5.4 rule 21).
+//
+// Epilogue fuses bias + gelu/gelu_tanh/quickgelu (+residual, +save-pre).
+// M may be ragged (MGUARD clamps staging rows and predicates stores);
+// N%256==0, K%64==0, K>=128 required (gemm8p_supported).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -51,43 +64,17 @@ __device__ __forceinline__ void raw_barrier() {
   asm volatile("s_barrier" ::: "memory");
 }
 
-// LDS image swizzle (conflict-FREE for the fragment read pattern): the
-// fragment reads put 16 consecutive rows (r0 multiple of 16) on the same
-// 16-B chunk — with a row-major image rows r and r+8 always collide (same
-// chunk xor, same row parity -> same 16-B slot of the 256-B bank row, 2-way
-// on ds_read_b128).  Fix: interleave row PAIRS within one 256-B region and
-// use row bit 3 to select the slot half:
-//   slot(row, c) = (c ^ (row&7)) | (((row&1) ^ ((row>>3)&1)) << 3)
-//   byte(row, c) = (row>>1)*256 + slot*16
-// For rows r0..r0+15 at fixed c the 16 slots are pairwise distinct, so a
-// wave's ds_read_b128 lane group touches 16 distinct 16-B slots = all 64
-// banks (SQ_LDS_BANK_CONFLICT 0).  glds writes lane-linear, so the inverse
-// permutation goes on the per-lane SOURCE address (§5.4 rule 21).
-//
-// Per-lane source base addresses (row/chunk fixed per lane; only the K
-// offset advances) are precomputed once — the K-loop issues glds with
-// base + kt*128 B, no per-phase 64-bit address rebuild.
-//
-// A half-image is staged by 512 threads x 16 B x 2 rounds; round r, thread
-// tid covers image byte off = r*8192 + tid*16.
-template <bool MGUARD, int SWZ>
+// Per-lane glds source base for one half-image (row/chunk fixed per lane;
+// only the K offset advances — the K-loop adds kt*128 B).  A half-image is
+// staged by 512 threads x 16 B x 2 rounds; round r, thread tid covers image
+// byte off = r*8192 + tid*16.
+template <bool MGUARD>
 __device__ __forceinline__ const bf16* stage_base(const bf16* __restrict__ gsrc,
                                                   int64_t ldg, int row0, int rows_total,
                                                   int round, int tid) {
   const int off = round * 8192 + tid * 16;
-  int row, chunk;
-  if constexpr (SWZ == 1) {
-    // invert byte(row,c): pair = off/256, slot = (off/16)&15
-    const int pair = off >> 8;
-    const int slot = (off >> 4) & 15;
-    const int half = slot >> 3;
-    const int b = half ^ ((pair >> 2) & 1);  // row bit 3 = pair bit 2
-    row = 2 * pair + b;
-    chunk = (slot & 7) ^ (row & 7);
-  } else {
-    row = off >> 7;
-    chunk = ((off >> 4) & 7) ^ (row & 7);
-  }
+  const int row = off >> 7;
+  const int chunk = ((off >> 4) & 7) ^ (row & 7);
   int grow = row0 + row;
   if (MGUARD) grow = grow < rows_total ? grow : rows_total - 1;
   return gsrc + (int64_t)grow * ldg + chunk * 8;
@@ -103,25 +90,22 @@ __device__ __forceinline__ bf16x8_t lds_frag8p(const char* base, int byte_off) {
   return *reinterpret_cast<const bf16x8_t*>(base + byte_off);
 }
 
-template <int SWZ>
 __device__ __forceinline__ int frag_off(int row, int chunk) {
-  if constexpr (SWZ == 1) {
-    const int slot = (chunk ^ (row & 7)) | ((((row & 1) ^ ((row >> 3) & 1))) << 3);
-    return (row >> 1) * 256 + slot * 16;
-  } else {
-    return (row << 7) + ((chunk ^ (row & 7)) << 4);
-  }
+  return (row << 7) + ((chunk ^ (row & 7)) << 4);
 }
 
-template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool MGUARD, int SWZ = 1>
+template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool MGUARD>
 __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ W, const float* __restrict__ bias,
     const bf16* __restrict__ res, bf16* __restrict__ Y, bf16* __restrict__ Z,
     int M, int N, int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // A images: [buf][half] at (buf*2+half)*16KiB; B images at +64KiB
+  // A images: [buf 0..1][half 0..1] at (buf*2+half)*16 KiB;
+  // B images: [slot 0..2][half 0..1] at 64 KiB + (slot*2+half)*16 KiB.
   auto As = [&](int buf, int h) { return smem + (buf * 2 + h) * HALF_BYTES; };
-  auto Bs = [&](int buf, int h) { return smem + 4 * HALF_BYTES + (buf * 2 + h) * HALF_BYTES; };
+  auto Bs = [&](int slot, int h) {
+    return smem + 4 * HALF_BYTES + (slot * 2 + h) * HALF_BYTES;
+  };
 
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
@@ -139,114 +123,117 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
   }
   const int m0 = (wg / nt) * BM;
   const int n0 = (wg % nt) * BN;
-  const int wm = (wave >> 2);      // 0..1 — M sub-position inside a quadrant
-  const int wn = (wave & 3);       // 0..3 — N sub-position
+  const int wm = (wave >> 2);  // 0..1
+  const int wn = (wave & 3);   // 0..3
 
-  f32x4_t acc[4][4][2] = {};  // [phase][mi][ni]
+  f32x4_t acc[2][4][4] = {};  // [pr][mi][ni]
 
   const int nk = K / BK;
-  // Per-lane staging source bases (row/chunk fixed per lane; K advances by
-  // 128 B per tile): [half][round] for each operand.
+  // Per-lane staging source bases: [half][round] per operand.
   const bf16* asrc[2][2];
   const bf16* bsrc[2][2];
 #pragma unroll
   for (int h = 0; h < 2; ++h)
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
-      asrc[h][r] = stage_base<MGUARD, SWZ>(X, K, m0 + h * 128, M, r, tid);
-      bsrc[h][r] = stage_base<false, SWZ>(W + (int64_t)n0 * K, K, h * 128, N, r, tid);
+      asrc[h][r] = stage_base<MGUARD>(X, K, m0 + h * 128, M, r, tid);
+      bsrc[h][r] = stage_base<false>(W + (int64_t)n0 * K, K, h * 128, N, r, tid);
     }
-  const int ldst = (tid * 16) & 8191;        // lds dest byte of round 0
+  const int ldst = tid * 16;  // lds dest byte of round 0
   auto stageA = [&](int kt, int h) {
     char* img = As(kt & 1, h);
     glds16(asrc[h][0] + (int64_t)kt * BK, img + ldst);
     glds16(asrc[h][1] + (int64_t)kt * BK, img + 8192 + ldst);
   };
-  auto stageB = [&](int kt, int h) {
-    char* img = Bs(kt & 1, h);
+  auto stageB = [&](int kt, int slot, int h) {
+    char* img = Bs(slot, h);
     glds16(bsrc[h][0] + (int64_t)kt * BK, img + ldst);
     glds16(bsrc[h][1] + (int64_t)kt * BK, img + 8192 + ldst);
   };
 
-  // Per-lane fragment read byte offsets (loop-invariant; image base varies
-  // by phase/buffer only).
-  int offA[2][4], offB[2][2];
+  // Per-lane fragment read byte offsets (loop-invariant).  A-frag (mi, kh):
+  // row wm*64+16mi+lo, chunk 4kh+hi.  B-frag (ni, kh): the wave's 64 cols
+  // live in B-half (wn>>1) at local rows (wn&1)*64 + 16ni + lo.
+  int offA[2][4], offB[2][4];
 #pragma unroll
-  for (int ks = 0; ks < 2; ++ks) {
+  for (int kh = 0; kh < 2; ++kh) {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) offA[ks][mi] = frag_off<SWZ>(wm * 64 + 16 * mi + lo, 4 * ks + hi);
+    for (int mi = 0; mi < 4; ++mi)
+      offA[kh][mi] = frag_off(wm * 64 + 16 * mi + lo, 4 * kh + hi);
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) offB[ks][ni] = frag_off<SWZ>(wn * 32 + 16 * ni + lo, 4 * ks + hi);
+    for (int ni = 0; ni < 4; ++ni)
+      offB[kh][ni] = frag_off((wn & 1) * 64 + 16 * ni + lo, 4 * kh + hi);
   }
+  const int bh = wn >> 1;  // this wave's B half
 
-  // Prologue: A0(0), B0(0), B1(0), A1(0), A0(1) then certify tile 0.
+  // Prologue: stage tile 0 (B0,B1,A0,A1), certify all but A1.
+  stageB(0, 0, 0);
+  stageB(0, 0, 1);
   stageA(0, 0);
-  stageB(0, 0);
-  stageB(0, 1);
   stageA(0, 1);
-  if (nk > 1) {
-    stageA(1, 0);
-    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-  } else {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  }
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
   raw_barrier();
 
-  // ONE barrier per phase: a wave's phase-p+1 fragment reads may overlap
-  // another wave's phase-p MFMA cluster (the LDS array drains under the
-  // matrix pipe instead of strictly after it).  Safety: the slot a glds
-  // overwrites was last read >= 2 barriers earlier, and a wave's reads
-  // complete (hipcc's lgkmcnt) before its own MFMA -> before its next
-  // barrier arrival, so no wave can see a slot mid-overwrite.
+  int slot = 0, slot1 = nk > 1 ? 1 : 0;  // B slots of tiles t, t+1
   for (int t = 0; t < nk; ++t) {
     const int buf = t & 1;
+    const char* a0i = As(buf, 0);
+    const char* a1i = As(buf, 1);
+    const char* bi = Bs(slot, bh);
+    const bool more = t + 1 < nk;
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
-      const int pr = p >> 1, pc = p & 1;
-      // ---- fragment ds_reads for this phase (tile t images) ----
-      bf16x8_t xa[2][4], wb[2][2];
-      {
-        const char* ai = As(buf, pr);
-        const char* bi = Bs(buf, pc);
+    for (int kh = 0; kh < 2; ++kh) {
+      // ---- pr = 0 phase: read A0-frags + B-frags(kh), stage, MFMA ----
+      bf16x8_t xa[4], wb[4];
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
+      for (int mi = 0; mi < 4; ++mi) xa[mi] = lds_frag8p(a0i, offA[kh][mi]);
 #pragma unroll
-          for (int mi = 0; mi < 4; ++mi) xa[ks][mi] = lds_frag8p(ai, offA[ks][mi]);
-#pragma unroll
-          for (int ni = 0; ni < 2; ++ni) wb[ks][ni] = lds_frag8p(bi, offB[ks][ni]);
-        }
+      for (int ni = 0; ni < 4; ++ni) wb[ni] = lds_frag8p(bi, offB[kh][ni]);
+      if (more) {
+        if (kh == 0) stageB(t + 1, slot1, 0);
+        else stageA(t + 1, 0);
       }
-      // ---- stage prefetch: p0->B0(t+1) p1->B1(t+1) p2->A1(t+1) p3->A0(t+2)
-      if (p == 0 && t + 1 < nk) stageB(t + 1, 0);
-      if (p == 1 && t + 1 < nk) stageB(t + 1, 1);
-      if (p == 2 && t + 1 < nk) stageA(t + 1, 1);
-      if (p == 3 && t + 2 < nk) stageA(t + 2, 0);
-      // ---- counted wait: keep 2 half-tiles (4 loads/wave) in flight.
-      // The last tile's first phase drains (its halves were the newest
-      // stages and the stream has ended — vmcnt(4) would not cover them).
-      if (t == nk - 1 && p == 0) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      if (kh == 0) {
+        if (t == nk - 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        else asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       }
       raw_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
+      for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+        for (int ni = 0; ni < 4; ++ni)
+          acc[0][mi][ni] = MFMA16(xa[mi], wb[ni], acc[0][mi][ni]);
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- pr = 1 phase: read A1-frags, REUSE B-frags, stage, MFMA ----
 #pragma unroll
-          for (int ni = 0; ni < 2; ++ni)
-            acc[p][mi][ni] = MFMA16(xa[ks][mi], wb[ks][ni], acc[p][mi][ni]);
+      for (int mi = 0; mi < 4; ++mi) xa[mi] = lds_frag8p(a1i, offA[kh][mi]);
+      if (more) {
+        if (kh == 0) stageB(t + 1, slot1, 1);
+        else stageA(t + 1, 1);
+      }
+      if (kh == 0) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      raw_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[1][mi][ni] = MFMA16(xa[mi], wb[ni], acc[1][mi][ni]);
       __builtin_amdgcn_s_setprio(0);
     }
+    slot = slot1;
+    slot1 = slot1 + 1 == 3 ? 0 : slot1 + 1;
   }
 
-  // Epilogue: per phase strip rows pr*128+wm*64+16mi+hi*4+r, cols
-  // pc*128+wn*32+16ni+lo (C layout: row = hi*4+r within a 16-fragment).
+  // Epilogue: acc[pr][mi][ni] -> rows m0 + pr*128 + wm*64 + 16mi + hi*4 + r,
+  // cols n0 + wn*64 + 16ni + lo.
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int pr = p >> 1, pc = p & 1;
+  for (int pr = 0; pr < 2; ++pr) {
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -254,9 +241,9 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
         const int m = m0 + pr * 128 + wm * 64 + 16 * mi + hi * 4 + r;
         if (MGUARD && m >= M) continue;
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-          const int n = n0 + pc * 128 + wn * 32 + 16 * ni + lo;
-          float vpre = acc[p][mi][ni][r];
+        for (int ni = 0; ni < 4; ++ni) {
+          const int n = n0 + wn * 64 + 16 * ni + lo;
+          float vpre = acc[pr][mi][ni][r];
           if (HAS_BIAS) vpre += bias[n];
           if (SAVE_PRE) Z[(int64_t)m * N + n] = f2bf(vpre);
           float vy = act_fwd(vpre, ACT);
@@ -269,11 +256,6 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
 }
 
 }  // namespace
-
-static int gemm8p_swz() {
-  const char* env = getenv("JIMM_AMD_GEMM_SWZ");
-  return env ? atoi(env) : 1;
-}
 
 bool gemm8p_supported(int64_t M, int64_t N, int64_t K) {
   return M >= 1 && (N % BN == 0) && (K % BK == 0) && K >= 2 * BK;
@@ -293,14 +275,13 @@ void gemm_nt_8p(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> b
   const bf16* resp = residual ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr;
   const float* biasp = bias_f32 ? bias_f32->data_ptr<float>() : nullptr;
   bf16* zp = z ? reinterpret_cast<bf16*>(z->data_ptr()) : nullptr;
-  const size_t shmem = 8 * HALF_BYTES;  // 128 KiB
+  const size_t shmem = 10 * HALF_BYTES;  // 160 KiB (full CU LDS)
   const bool mguard = (M % BM) != 0;
   const int mt = (M + BM - 1) / BM;
 
 #define LAUNCH8P(ACTC, HB, HR, SP, MG)                                                     \
   do {                                                                                     \
-    auto kfn = gemm8p_swz() == 1 ? gemm_nt_8p_kernel<ACTC, HB, HR, SP, MG, 1>          \
-                                 : gemm_nt_8p_kernel<ACTC, HB, HR, SP, MG, 0>;                                    \
+    auto kfn = gemm_nt_8p_kernel<ACTC, HB, HR, SP, MG>;                                    \
     static bool attr_set_##ACTC##HB##HR##SP##MG = [&] {                                    \
       hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),                              \
                           hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);         \
