@@ -94,7 +94,7 @@ __device__ __forceinline__ int frag_off(int row, int chunk) {
   return (row << 7) + ((chunk ^ (row & 7)) << 4);
 }
 
-template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool MGUARD>
+template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool MGUARD, bool GRADM = false>
 __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ W, const float* __restrict__ bias,
     const bf16* __restrict__ res, bf16* __restrict__ Y, bf16* __restrict__ Z,
@@ -246,8 +246,15 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
           float vpre = acc[pr][mi][ni][r];
           if (HAS_BIAS) vpre += bias[n];
           if (SAVE_PRE) Z[(int64_t)m * N + n] = f2bf(vpre);
-          float vy = act_fwd(vpre, ACT);
-          if (HAS_RES) vy += bf2f(res[(int64_t)m * N + n]);
+          float vy;
+          if (GRADM) {
+            // dX epilogue fused with the activation backward: res carries
+            // the saved pre-activation z; vy = (dy @ W) * act'(z)
+            vy = vpre * act_grad(bf2f(res[(int64_t)m * N + n]), ACT);
+          } else {
+            vy = act_fwd(vpre, ACT);
+            if (HAS_RES) vy += bf2f(res[(int64_t)m * N + n]);
+          }
           Y[(int64_t)m * N + n] = f2bf(vy);
         }
       }
@@ -259,6 +266,55 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
 
 bool gemm8p_supported(int64_t M, int64_t N, int64_t K) {
   return M >= 1 && (N % BN == 0) && (K % BK == 0) && K >= 2 * BK;
+}
+
+torch::Tensor gemm_nt_8p_gradact(torch::Tensor dy, torch::Tensor wt, torch::Tensor z,
+                                 std::string act) {
+  // dz = (dy @ wt^T as NT on pre-transposed wt [in_f, out_f]) * act'(z) —
+  // the dX GEMM of an activated linear with the activation backward fused
+  // into the epilogue (saves the separate act_bwd elementwise pass).
+  const int M = dy.size(0), K = dy.size(1), N = wt.size(0);
+  TORCH_CHECK(dy.is_contiguous() && wt.is_contiguous() && z.is_contiguous());
+  TORCH_CHECK(wt.size(1) == K && z.size(0) == M && z.size(1) == N);
+  TORCH_CHECK(gemm8p_supported(M, N, K));
+  int act_code = ACT_NONE;
+  if (act == "gelu") act_code = ACT_GELU;
+  else if (act == "gelu_tanh") act_code = ACT_GELU_TANH;
+  else if (act == "quickgelu") act_code = ACT_QUICKGELU;
+  else TORCH_CHECK(false, "gradact needs an activation, got ", act);
+  auto y = torch::empty({M, N}, dy.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  const size_t shmem = 10 * HALF_BYTES;
+  const bool mguard = (M % BM) != 0;
+  const int mt = (M + BM - 1) / BM;
+#define LAUNCH_GRAD(ACTC, MG)                                                              \
+  do {                                                                                     \
+    auto kfn = gemm_nt_8p_kernel<ACTC, false, true, false, MG, true>;                      \
+    static bool attr_g_##ACTC##MG = [&] {                                                  \
+      hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),                              \
+                          hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);         \
+      return true;                                                                         \
+    }();                                                                                   \
+    (void)attr_g_##ACTC##MG;                                                               \
+    hipLaunchKernelGGL(kfn, dim3(mt * (N / BN)), dim3(NTHREADS), shmem, stream,            \
+                       reinterpret_cast<const bf16*>(dy.data_ptr()),                       \
+                       reinterpret_cast<const bf16*>(wt.data_ptr()), nullptr,              \
+                       reinterpret_cast<const bf16*>(z.data_ptr()),                        \
+                       reinterpret_cast<bf16*>(y.data_ptr()), nullptr, M, N, K);           \
+  } while (0)
+#define DG(ACTC)                                                                           \
+  do {                                                                                     \
+    if (mguard) LAUNCH_GRAD(ACTC, true);                                                   \
+    else LAUNCH_GRAD(ACTC, false);                                                         \
+  } while (0)
+  switch (act_code) {
+    case ACT_GELU: DG(ACT_GELU); break;
+    case ACT_GELU_TANH: DG(ACT_GELU_TANH); break;
+    case ACT_QUICKGELU: DG(ACT_QUICKGELU); break;
+  }
+#undef DG
+#undef LAUNCH_GRAD
+  return y;
 }
 
 void gemm_nt_8p(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias_f32,

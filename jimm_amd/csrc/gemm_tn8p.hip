@@ -1,0 +1,315 @@
+// K15 — TN weight-gradient GEMM: dW[N,Kw] = dz^T @ X (contraction over the
+// huge M = B*L dimension), 256x256 output tiles, split-M, MFMA bf16.
+//
+// Both operands are stored m-major ([M,N] / [M,Kw]) so both MFMA fragments
+// need per-lane m-columns.  Staging: coalesced 16-B global loads into
+// registers, then scattered ds_write_b128 into a transpose-friendly image;
+// fragments come back by ds_read_b64_tr_b16 (the gfx950 hardware transpose
+// read — no builtin, inline asm per guide §5.7 form (i): reads + their
+// s_waitcnt lgkmcnt(0) in ONE statement, outputs "=&v").
+//
+// Image layout per operand per buffer (32 KiB, for one 64-m x 256-col tile):
+//   byte(m, col) = kh*16384 + nt*1024 + qpos*128 + jm*32 + u*16
+//   with kh = m>>5 (K-half), q' = (m>>2)&7, jm = m&3, qpos = (q'&1)*4+(q'>>1)
+//   (even m-quads first, then odd), nt = col>>4, u = (col>>3)&1.
+// A ds_read_b64_tr_b16 at per-lane address base + (lane&15)*2 + (lane>>4)*128
+// delivers lane (lo,hi) the 4 elements m = 8*hi + (0..3) of column
+// nt*16 + lo (the odd quads via immediate offset +512) — exactly the MFMA
+// fragment m-octet per hi group, with the 32-lane groups touching disjoint
+// bank rows (the verified T10 subtile pattern).
+//
+// Phase walk per 64-m tile mirrors gemm8p.hip: (kh, pr) x 4 phases, one
+// __syncthreads() per phase (no glds in this kernel, so syncthreads does
+// not drain the VMEM queue), X-side fragments reused in registers across
+// the pr pair.  Staging for tile t+1: global loads issue at p0, dz-image
+// ds_writes at p1, X-image writes at p2 (>= 2 barriers after the old
+// buffer's last reader).  M-tail rows load zeros (no clamping — this is an
+// accumulation).
+//
+// Split-M: grid.y walks m-chunks (slow index -> the chunk's operand rows
+// stay L3-resident across its output tiles); partial results combine with
+// fp32 atomicAdd into a zeroed C (one-shot store when grid.y == 1).
+// Deterministic mode uses the rocBLAS path instead (ops/hip_linear.py).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+typedef short bf16x4_t __attribute__((ext_vector_type(4)));
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+#define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
+
+constexpr int BN = 256, BKW = 256, BMS = 64;  // n-tile, k-tile, m-step
+constexpr int NTHREADS = 512;
+constexpr int IMG_BYTES = 32768;  // one 64x256 bf16 image
+
+// image byte offset of element (m_local 0..63, col 0..255)
+__device__ __forceinline__ int img_off(int m, int col) {
+  const int kh = m >> 5, q = (m >> 2) & 7, jm = m & 3;
+  const int qpos = (q & 1) * 4 + (q >> 1);
+  return kh * 16384 + (col >> 4) * 1024 + qpos * 128 + jm * 32 + ((col >> 3) & 1) * 16;
+}
+
+// 4 tr reads of one frag-quad column block: returns the two m-octet halves
+// of MFMA operands for fragments f and f+1?  We issue per-phase batches
+// instead — see the asm blocks in the kernel.
+
+template <bool ATOMIC>
+__global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
+    const bf16* __restrict__ DZ, const bf16* __restrict__ X, float* __restrict__ C,
+    int64_t M, int N, int Kw, int64_t chunk_m) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // dz images: [buf]*32 KiB at 0; X images at 64 KiB.
+  auto DZs = [&](int buf) { return smem + buf * IMG_BYTES; };
+  auto Xs = [&](int buf) { return smem + 2 * IMG_BYTES + buf * IMG_BYTES; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid % WAVE;
+  const int wave = tid / WAVE;
+  const int lo = lane & 15, hi = lane >> 4;
+
+  const int nt_n = N / BN, nt_k = Kw / BKW;
+  const int nwg = nt_n * nt_k;
+  int wg = blockIdx.x;
+  {  // XCD-aware bijective remap
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int n0 = (wg / nt_k) * BN;
+  const int k0 = (wg % nt_k) * BKW;
+  const int wm = wave >> 2;  // 0..1 (n sub-position)
+  const int wn = wave & 3;   // 0..3 (k sub-position)
+
+  const int64_t m_begin = (int64_t)blockIdx.y * chunk_m;
+  const int64_t m_end = m_begin + chunk_m < M ? m_begin + chunk_m : M;
+  const int ntile = (int)((m_end - m_begin + BMS - 1) / BMS);
+  if (ntile <= 0) return;
+
+  f32x4_t acc[2][4][4] = {};  // [pr][mi][ni]
+
+  // Staging decode for this thread: 4 rounds, round r covers global chunk
+  // g = r*512 + tid -> m_local = g>>5, col = (g&31)*8.
+  int m_loc[4], col[4], io[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int g = r * NTHREADS + tid;
+    m_loc[r] = g >> 5;
+    col[r] = (g & 31) * 8;
+    io[r] = img_off(m_loc[r], col[r]);
+  }
+
+  typedef short s8 __attribute__((ext_vector_type(8)));
+  s8 ldz[4], lx[4];
+  auto issue_loads = [&](int t) {
+    const int64_t mt0 = m_begin + (int64_t)t * BMS;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int64_t m = mt0 + m_loc[r];
+      if (m < m_end) {
+        ldz[r] = *reinterpret_cast<const s8*>(DZ + m * N + n0 + col[r]);
+        lx[r] = *reinterpret_cast<const s8*>(X + m * Kw + k0 + col[r]);
+      } else {
+        ldz[r] = s8{};
+        lx[r] = s8{};
+      }
+    }
+  };
+  auto write_dz = [&](int buf) {
+    char* img = DZs(buf);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) *reinterpret_cast<s8*>(img + io[r]) = ldz[r];
+  };
+  auto write_x = [&](int buf) {
+    char* img = Xs(buf);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) *reinterpret_cast<s8*>(img + io[r]) = lx[r];
+  };
+
+  // Per-lane tr-read address term: lane*8 B — each lane points at its own
+  // consecutive b64 slot; the hardware transposes within each 16-lane group
+  // (verified by ext.tr16_probe mode 1: lane l elem j reads short
+  // (l>>4)*64 + (l&15) + j*16 relative to lane 0's address).
+  const int lane_term = lane * 8;
+
+  // prologue: stage tile 0
+  issue_loads(0);
+  write_dz(0);
+  write_x(0);
+  __syncthreads();
+
+  for (int t = 0; t < ntile; ++t) {
+    const int buf = t & 1;
+    const bool more = t + 1 < ntile;
+    bf16x8_t xb[4];  // X-side fragments, reused across the pr pair
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      // ---------- phase (kh, pr=0) ----------
+      if (kh == 0 && more) issue_loads(t + 1);
+      {
+        // B (X image) frags ni=0..3 at nt = wn*4+ni, plus A (dz) frags
+        // mi=0..3 at nt = wm*4+mi (pr=0).  8+8 tr reads + lgkmcnt(0) in one
+        // asm (form i).
+        typedef const __attribute__((address_space(3))) char* lds_p;
+        const lds_p bxi = (lds_p)(const void*)(Xs(buf) + kh * 16384 + wn * 4096 + lane_term);
+        const lds_p adi = (lds_p)(const void*)(DZs(buf) + kh * 16384 + wm * 4096 + lane_term);
+        bf16x4_t b0l, b0h, b1l, b1h, b2l, b2h, b3l, b3h;
+        bf16x4_t a0l, a0h, a1l, a1h, a2l, a2h, a3l, a3h;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %16 offset:0\n\t"
+            "ds_read_b64_tr_b16 %1, %16 offset:512\n\t"
+            "ds_read_b64_tr_b16 %2, %16 offset:1024\n\t"
+            "ds_read_b64_tr_b16 %3, %16 offset:1536\n\t"
+            "ds_read_b64_tr_b16 %4, %16 offset:2048\n\t"
+            "ds_read_b64_tr_b16 %5, %16 offset:2560\n\t"
+            "ds_read_b64_tr_b16 %6, %16 offset:3072\n\t"
+            "ds_read_b64_tr_b16 %7, %16 offset:3584\n\t"
+            "ds_read_b64_tr_b16 %8, %17 offset:0\n\t"
+            "ds_read_b64_tr_b16 %9, %17 offset:512\n\t"
+            "ds_read_b64_tr_b16 %10, %17 offset:1024\n\t"
+            "ds_read_b64_tr_b16 %11, %17 offset:1536\n\t"
+            "ds_read_b64_tr_b16 %12, %17 offset:2048\n\t"
+            "ds_read_b64_tr_b16 %13, %17 offset:2560\n\t"
+            "ds_read_b64_tr_b16 %14, %17 offset:3072\n\t"
+            "ds_read_b64_tr_b16 %15, %17 offset:3584\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(b0l), "=&v"(b0h), "=&v"(b1l), "=&v"(b1h), "=&v"(b2l), "=&v"(b2h),
+              "=&v"(b3l), "=&v"(b3h), "=&v"(a0l), "=&v"(a0h), "=&v"(a1l), "=&v"(a1h),
+              "=&v"(a2l), "=&v"(a2h), "=&v"(a3l), "=&v"(a3h)
+            : "v"(bxi), "v"(adi)
+            : "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        xb[0] = __builtin_shufflevector(b0l, b0h, 0, 1, 2, 3, 4, 5, 6, 7);
+        xb[1] = __builtin_shufflevector(b1l, b1h, 0, 1, 2, 3, 4, 5, 6, 7);
+        xb[2] = __builtin_shufflevector(b2l, b2h, 0, 1, 2, 3, 4, 5, 6, 7);
+        xb[3] = __builtin_shufflevector(b3l, b3h, 0, 1, 2, 3, 4, 5, 6, 7);
+        bf16x8_t ad[4];
+        ad[0] = __builtin_shufflevector(a0l, a0h, 0, 1, 2, 3, 4, 5, 6, 7);
+        ad[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
+        ad[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
+        ad[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
+        __syncthreads();
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[0][mi][ni] = MFMA16(ad[mi], xb[ni], acc[0][mi][ni]);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      // ---------- phase (kh, pr=1) ----------
+      {
+        if (kh == 0 && more) write_dz(buf ^ 1);
+        if (kh == 1 && more) write_x(buf ^ 1);
+        typedef const __attribute__((address_space(3))) char* lds_p;
+        const lds_p adi = (lds_p)(const void*)(DZs(buf) + kh * 16384 + 8192 + wm * 4096 + lane_term);
+        bf16x4_t a0l, a0h, a1l, a1h, a2l, a2h, a3l, a3h;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %8 offset:0\n\t"
+            "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+            "ds_read_b64_tr_b16 %2, %8 offset:1024\n\t"
+            "ds_read_b64_tr_b16 %3, %8 offset:1536\n\t"
+            "ds_read_b64_tr_b16 %4, %8 offset:2048\n\t"
+            "ds_read_b64_tr_b16 %5, %8 offset:2560\n\t"
+            "ds_read_b64_tr_b16 %6, %8 offset:3072\n\t"
+            "ds_read_b64_tr_b16 %7, %8 offset:3584\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(a0l), "=&v"(a0h), "=&v"(a1l), "=&v"(a1h), "=&v"(a2l), "=&v"(a2h),
+              "=&v"(a3l), "=&v"(a3h)
+            : "v"(adi)
+            : "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        bf16x8_t ad[4];
+        ad[0] = __builtin_shufflevector(a0l, a0h, 0, 1, 2, 3, 4, 5, 6, 7);
+        ad[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
+        ad[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
+        ad[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
+        __syncthreads();
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[1][mi][ni] = MFMA16(ad[mi], xb[ni], acc[1][mi][ni]);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+  }
+
+  // Epilogue: acc[pr][mi][ni] -> C rows n = n0 + pr*128 + wm*64 + 16mi +
+  // hi*4 + r, cols k = k0 + wn*64 + 16ni + lo.
+#pragma unroll
+  for (int pr = 0; pr < 2; ++pr)
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int n = n0 + pr * 128 + wm * 64 + 16 * mi + hi * 4 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          const int k = k0 + wn * 64 + 16 * ni + lo;
+          float* dst = C + (int64_t)n * Kw + k;
+          if (ATOMIC) atomicAdd(dst, acc[pr][mi][ni][r]);
+          else *dst = acc[pr][mi][ni][r];
+        }
+      }
+}
+
+}  // namespace
+
+bool gemm_tn8p_supported(int64_t M, int64_t N, int64_t K) {
+  return M >= 1 && (N % BN == 0) && (K % BKW == 0);
+}
+
+torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x) {
+  // dz (M, N) bf16, x (M, Kw) bf16 -> dW (N, Kw) fp32 = dz^T @ x
+  TORCH_CHECK(dz.is_cuda() && dz.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(dz.scalar_type() == torch::kBFloat16 && x.scalar_type() == torch::kBFloat16);
+  const int64_t M = dz.size(0);
+  const int N = dz.size(1), Kw = x.size(1);
+  TORCH_CHECK(x.size(0) == M && gemm_tn8p_supported(M, N, Kw));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int tiles = (N / BN) * (Kw / BKW);
+  // split-M so total blocks ~ 2-3 waves of the 256 CUs (JIMM_AMD_DW_SPLITM
+  // overrides for tuning; higher split = more parallelism but splitm x
+  // C-sized fp32 atomic traffic)
+  static const int splitm_env = [] {
+    const char* e = getenv("JIMM_AMD_DW_SPLITM");
+    return e ? atoi(e) : 0;
+  }();
+  int splitm = splitm_env > 0
+                   ? splitm_env
+                   : (int)std::min<int64_t>((640 + tiles - 1) / tiles, (M + BMS - 1) / BMS);
+  if ((int64_t)splitm > (M + BMS - 1) / BMS) splitm = (int)((M + BMS - 1) / BMS);
+  if (splitm < 1) splitm = 1;
+  int64_t chunk = ((M + splitm - 1) / splitm + BMS - 1) / BMS * BMS;
+  splitm = (int)((M + chunk - 1) / chunk);
+  auto C = splitm > 1 ? torch::zeros({(int64_t)N, (int64_t)Kw},
+                                     dz.options().dtype(torch::kFloat32))
+                      : torch::empty({(int64_t)N, (int64_t)Kw},
+                                     dz.options().dtype(torch::kFloat32));
+  const size_t shmem = 4 * IMG_BYTES;  // 128 KiB
+#define LAUNCH_TN(AT)                                                                      \
+  do {                                                                                     \
+    auto kfn = gemm_tn_8p_kernel<AT>;                                                      \
+    static bool attr_##AT = [&] {                                                          \
+      hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),                              \
+                          hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);         \
+      return true;                                                                         \
+    }();                                                                                   \
+    (void)attr_##AT;                                                                       \
+    hipLaunchKernelGGL(kfn, dim3(tiles, splitm), dim3(NTHREADS), shmem, stream,            \
+                       reinterpret_cast<const bf16*>(dz.data_ptr()),                       \
+                       reinterpret_cast<const bf16*>(x.data_ptr()), C.data_ptr<float>(),   \
+                       M, N, Kw, chunk);                                                   \
+  } while (0)
+  if (splitm > 1) LAUNCH_TN(true);
+  else LAUNCH_TN(false);
+#undef LAUNCH_TN
+  return C;
+}

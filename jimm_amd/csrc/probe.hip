@@ -62,7 +62,50 @@ __global__ void mfma_probe32_kernel(const bf16* __restrict__ A, const bf16* __re
       C[(8 * blk + hi * 4 + r) * 32 + lo] = c[4 * blk + r];
 }
 
+typedef short bf16x4_t __attribute__((ext_vector_type(4)));
+
+template <int MODE>
+__global__ void tr16_probe_kernel(const short* __restrict__ src, short* __restrict__ out) {
+  // Fill LDS with src[0..255] (a 512-byte pattern), then ds_read_b64_tr_b16
+  // with per-lane address addr(l) = MODE==0 ? (l&15)*2 + (l>>4)*128
+  //                              : MODE==1 ? l*8
+  //                              : 0 (uniform)
+  // and dump each lane's 4 shorts: out[lane*4 + j].
+  __shared__ __attribute__((aligned(16))) short buf[256];
+  const int lane = threadIdx.x % WAVE;
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) buf[i] = src[i];
+  __syncthreads();
+  int off;
+  if (MODE == 0) off = (lane & 15) * 2 + (lane >> 4) * 128;
+  else if (MODE == 1) off = lane * 8;
+  else off = 0;
+  typedef const __attribute__((address_space(3))) char* lds_p;
+  const lds_p addr = (lds_p)(const void*)(reinterpret_cast<const char*>(buf) + off);
+  bf16x4_t v;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=&v"(v) : "v"(addr) : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = v[j];
+}
+
 }  // namespace
+
+torch::Tensor tr16_probe(torch::Tensor src, int64_t mode) {
+  TORCH_CHECK(src.is_cuda() && src.scalar_type() == torch::kInt16 && src.numel() == 256);
+  auto out = torch::empty({64, 4}, src.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  if (mode == 0)
+    hipLaunchKernelGGL(tr16_probe_kernel<0>, dim3(1), dim3(64), 0, stream,
+                       (const short*)src.contiguous().data_ptr(), (short*)out.data_ptr());
+  else if (mode == 1)
+    hipLaunchKernelGGL(tr16_probe_kernel<1>, dim3(1), dim3(64), 0, stream,
+                       (const short*)src.contiguous().data_ptr(), (short*)out.data_ptr());
+  else
+    hipLaunchKernelGGL(tr16_probe_kernel<2>, dim3(1), dim3(64), 0, stream,
+                       (const short*)src.contiguous().data_ptr(), (short*)out.data_ptr());
+  return out;
+}
 
 torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);

@@ -19,12 +19,19 @@ import os
 import torch
 
 from jimm_amd.ops import _backend
+from jimm_amd.ops.hip_linear import _dw_gemm, _dx_gemm, _gemm_mode
 
 
 def _colsum(ext, dz: torch.Tensor) -> torch.Tensor:
     if dz.shape[-1] % 8 == 0:
         return ext.colsum(dz).to(dz.dtype)
     return dz.sum(dim=0)
+
+
+def _hip_gemms(M: int, H: int) -> bool:
+    """True when the encoder block's GEMMs run on the in-house MFMA kernels
+    (JIMM_AMD_GEMM=hip default; all block shapes have N%256==0, K%64==0)."""
+    return _gemm_mode() == "hip" and H % 256 == 0
 
 
 def fused_block_enabled(x: torch.Tensor, dropout_p: float) -> bool:
@@ -47,22 +54,35 @@ class EncoderBlockFn(torch.autograd.Function):
         B, L, H = x.shape
         x = x.contiguous()
         x2 = x.view(-1, H)
+        hip = _hip_gemms(B * L, H)
         h1, mean1, rstd1 = ext.layernorm_fwd(x, ln1w, ln1b, eps)
-        qkv2 = torch.addmm(bqkv, h1.view(-1, H), wqkv.t())       # (M, 3H) fused bias
+        if hip:
+            # in-house MFMA GEMMs with the bias / act / residual epilogues
+            # fused into the GEMM kernel itself (csrc/gemm8p.hip)
+            qkv2, _ = ext.linear_fwd(h1.view(-1, H), wqkv, bqkv, "", None, False)
+        else:
+            qkv2 = torch.addmm(bqkv, h1.view(-1, H), wqkv.t())   # (M, 3H) fused bias
         qkv = qkv2.view(B, L, 3, num_heads, H // num_heads)
         q = qkv[:, :, 0].transpose(1, 2)
         k = qkv[:, :, 1].transpose(1, 2)
         v = qkv[:, :, 2].transpose(1, 2)
         o, lse = ext.attn_fwd(q, k, v, causal, scale)            # (B,nh,L,d), (B,L,nh,d) storage
         o2 = o.transpose(1, 2).reshape(-1, H)                    # free view
-        a2 = torch.matmul(o2, wproj.t())
-        a = ext.bias_act_fwd(a2, bproj, "", x2)                  # + bias + residual (one pass)
-        a3 = a.view(B, L, H)
-        h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)
-        z1 = torch.matmul(h2.view(-1, H), w1.t())
-        f = ext.bias_act_fwd(z1, b1, act, None)                  # z1 -> pre-activation in place
-        y2 = torch.matmul(f, w2.t())
-        y = ext.bias_act_fwd(y2, b2, "", a)                      # + bias + residual
+        if hip:
+            a, _ = ext.linear_fwd(o2, wproj, bproj, "", x2, False)
+            a3 = a.view(B, L, H)
+            h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)
+            f, z1 = ext.linear_fwd(h2.view(-1, H), w1, b1, act, None, True)
+            y, _ = ext.linear_fwd(f, w2, b2, "", a, False)
+        else:
+            a2 = torch.matmul(o2, wproj.t())
+            a = ext.bias_act_fwd(a2, bproj, "", x2)              # + bias + residual (one pass)
+            a3 = a.view(B, L, H)
+            h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)
+            z1 = torch.matmul(h2.view(-1, H), w1.t())
+            f = ext.bias_act_fwd(z1, b1, act, None)              # z1 -> pre-activation in place
+            y2 = torch.matmul(f, w2.t())
+            y = ext.bias_act_fwd(y2, b2, "", a)                  # + bias + residual
         ctx.save_for_backward(x, ln1w, wqkv, wproj, ln2w, w1, w2,
                               h1, qkv, o, lse, a, mean1, rstd1, mean2, rstd2, h2, z1, f)
         ctx.dims = (B, L, H, num_heads)
@@ -79,14 +99,20 @@ class EncoderBlockFn(torch.autograd.Function):
         dy2 = dy.contiguous().view(-1, H)
         h1_2 = h1.view(-1, H)
         h2_2 = h2.view(-1, H)
+        hip = _hip_gemms(dy2.shape[0], H)
 
         # MLP fc2 (+residual into a)
-        df = torch.matmul(dy2, w2)
-        dw2 = torch.matmul(dy2.t(), f)
+        if hip:
+            # dX of fc2 with the activation backward fused into the GEMM
+            # epilogue: dz1 = (dy @ W2) * act'(z1) — no separate act_bwd pass
+            dz1 = ext.gemm_nt_8p_gradact(dy2, w2.t().contiguous(), z1, act)
+        else:
+            df = torch.matmul(dy2, w2)
+            dz1 = ext.act_bwd(df, z1, act)
+        dw2 = _dw_gemm(ext, dy2, f, w2.dtype)
         db2 = _colsum(ext, dy2)
-        dz1 = ext.act_bwd(df, z1, act)
-        dh2 = torch.matmul(dz1, w1)
-        dw1 = torch.matmul(dz1.t(), h2_2)
+        dh2 = _dx_gemm(ext, dz1, w1)
+        dw1 = _dw_gemm(ext, dz1, h2_2, w1.dtype)
         db1 = _colsum(ext, dz1)
         # LN2 backward with the MLP residual grad (dy) fused into dx
         da3, dln2w, dln2b = ext.layernorm_bwd(
@@ -95,8 +121,8 @@ class EncoderBlockFn(torch.autograd.Function):
         da2 = da3.view(-1, H)
 
         # attention out-projection
-        do2 = torch.matmul(da2, wproj)
-        dwproj = torch.matmul(da2.t(), o.transpose(1, 2).reshape(-1, H))
+        do2 = _dx_gemm(ext, da2, wproj)
+        dwproj = _dw_gemm(ext, da2, o.transpose(1, 2).reshape(-1, H), wproj.dtype)
         dbproj = _colsum(ext, da2)
         do_v = do2.view(B, L, nh, H // nh).permute(0, 2, 1, 3)   # (B,nh,L,d) strided view
 
@@ -113,8 +139,8 @@ class EncoderBlockFn(torch.autograd.Function):
         dqkv2 = dqkv.view(-1, 3 * H)
 
         # QKV projection
-        dh1 = torch.matmul(dqkv2, wqkv)
-        dwqkv = torch.matmul(dqkv2.t(), h1_2)
+        dh1 = _dx_gemm(ext, dqkv2, wqkv)
+        dwqkv = _dw_gemm(ext, dqkv2, h1_2, wqkv.dtype)
         dbqkv = _colsum(ext, dqkv2)
         # LN1 backward with the attention residual grad (da) fused into dx
         dx, dln1w, dln1b = ext.layernorm_bwd(

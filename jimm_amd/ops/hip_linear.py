@@ -71,12 +71,41 @@ def _gemm_nt_fp8(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
 
 
 def _gemm_mode() -> str:
-    # Default "blas": measured on MI355X (benchmarks/gemm_bench.py, r03): the
-    # in-house 256-tile MFMA GEMM reaches 542-868 TF/s vs rocBLAS 845-1089 on
-    # the model-zoo shapes, so plain GEMMs run through rocBLAS with the fused
-    # HIP bias+act(+residual) epilogue kernel. JIMM_AMD_GEMM=hip re-enables
-    # the in-house path (kernels stay built and tested).
-    return os.environ.get("JIMM_AMD_GEMM", "blas")
+    # Default "hip" (round 2): the 8-phase 256-tile MFMA GEMM (csrc/gemm8p)
+    # with epilogues (bias/act/residual/act-bwd) fused into the GEMM itself,
+    # plus the split-M TN dW kernel (csrc/gemm_tn8p). JIMM_AMD_GEMM=blas
+    # switches back to hipBLASLt/rocBLAS + separate elementwise passes.
+    return os.environ.get("JIMM_AMD_GEMM", "hip")
+
+
+def _deterministic() -> bool:
+    return os.environ.get("JIMM_AMD_DETERMINISTIC", "0") == "1"
+
+
+def _dw_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype) -> torch.Tensor:
+    """dW = dz^T @ x2 — in-house split-M TN kernel when supported (fp32
+    accumulate, atomic combine -> rocBLAS fallback in deterministic mode)."""
+    if (
+        _gemm_mode() == "hip"
+        and not _deterministic()
+        and dz.dtype == torch.bfloat16
+        and ext.gemm_tn8p_supported(dz.shape[0], dz.shape[1], x2.shape[1])
+    ):
+        return ext.gemm_tn_8p(dz, x2).to(out_dtype)
+    return torch.matmul(dz.t(), x2)
+
+
+def _dx_gemm(ext, dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy."""
+    if (
+        _gemm_mode() == "hip"
+        and dz.dtype == torch.bfloat16
+        and ext.gemm8p_supported(dz.shape[0], w.shape[1], w.shape[0])
+    ):
+        wt = w.t().contiguous()
+        y, _ = ext.linear_fwd(dz, wt, None, "", None, False)
+        return y
+    return torch.matmul(dz, w)
 
 
 def _gemm_nt(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
@@ -129,22 +158,8 @@ class _LinearActFn(torch.autograd.Function):
         else:
             dz = dy2
         ext = _backend.ext()
-        dx = torch.matmul(dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
-        if ctx.needs_input_grad[1]:
-            # dW = dz^T @ x. rocBLAS picks weak non-split kernels here
-            # (0.23-0.6 PF/s) but still beats the in-house split-K TN kernel
-            # (0.25-0.28 PF/s, r03 measurements) — rocBLAS by default, the
-            # in-house kernel opt-in via JIMM_AMD_DW=hip while it is tuned.
-            if (
-                os.environ.get("JIMM_AMD_DW", "blas") == "hip"
-                and dz.dtype == torch.bfloat16
-                and ext.gemm_dw_supported(dz.shape[0], dz.shape[1], x2.shape[1])
-            ):
-                dw = ext.gemm_tn_splitk(dz, x2)
-            else:
-                dw = torch.matmul(dz.t(), x2)
-        else:
-            dw = None
+        dx = _dx_gemm(ext, dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
+        dw = _dw_gemm(ext, dz, x2, w.dtype) if ctx.needs_input_grad[1] else None
         if ctx.has_bias and ctx.needs_input_grad[2]:
             if dz.shape[-1] % 8 == 0:
                 db = _backend.ext().colsum(dz).to(dz.dtype)
@@ -175,9 +190,16 @@ class _PatchEmbedFn(torch.autograd.Function):
         h, wn = H // patch, W // patch
         ext = _backend.ext()
         cols = ext.im2col_patch(img.contiguous(), patch)  # (B*h*w, C*P*P)
-        w2 = w.reshape(w.shape[0], -1)  # (hidden, C*P*P)
-        z = _gemm_nt(cols, w2)
-        y = ext.bias_act_fwd(z, b, "", None)
+        w2 = w.reshape(w.shape[0], -1).contiguous()  # (hidden, C*P*P)
+        if (
+            _gemm_mode() == "hip"
+            and cols.dtype == torch.bfloat16
+            and ext.gemm8p_supported(cols.shape[0], w2.shape[0], w2.shape[1])
+        ):
+            y, _ = ext.linear_fwd(cols, w2, b, "", None, False)
+        else:
+            z = _gemm_nt(cols, w2)
+            y = ext.bias_act_fwd(z, b, "", None)
         ctx.save_for_backward(cols, w2)
         ctx.img_shape = img.shape
         ctx.patch = patch
@@ -189,10 +211,11 @@ class _PatchEmbedFn(torch.autograd.Function):
     def backward(ctx, dy):
         cols, w2 = ctx.saved_tensors
         patch = ctx.patch
+        ext = _backend.ext()
         dy2 = dy.contiguous().reshape(-1, dy.shape[-1])
-        dcols = torch.matmul(dy2, w2)
-        dimg = _backend.ext().col2im_patch(dcols, list(ctx.img_shape), patch) if ctx.needs_input_grad[0] else None
-        dw = torch.matmul(dy2.t(), cols).view(ctx.w_shape) if ctx.needs_input_grad[1] else None
+        dcols = _dx_gemm(ext, dy2, w2)
+        dimg = ext.col2im_patch(dcols, list(ctx.img_shape), patch) if ctx.needs_input_grad[0] else None
+        dw = _dw_gemm(ext, dy2, cols, w2.dtype).view(ctx.w_shape) if ctx.needs_input_grad[1] else None
         if ctx.has_bias and ctx.needs_input_grad[2]:
             if dy2.shape[-1] % 8 == 0:
                 db = _backend.ext().colsum(dy2).to(dy2.dtype)
