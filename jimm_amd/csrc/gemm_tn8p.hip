@@ -48,11 +48,33 @@ constexpr int BN = 256, BKW = 256, BMS = 64;  // n-tile, k-tile, m-step
 constexpr int NTHREADS = 512;
 constexpr int IMG_BYTES = 32768;  // one 64x256 bf16 image
 
+__device__ __forceinline__ void raw_barrier_tn() {
+  asm volatile("s_barrier" ::: "memory");
+}
+
 // image byte offset of element (m_local 0..63, col 0..255)
 __device__ __forceinline__ int img_off(int m, int col) {
   const int kh = m >> 5, q = (m >> 2) & 7, jm = m & 3;
   const int qpos = (q & 1) * 4 + (q >> 1);
   return kh * 16384 + (col >> 4) * 1024 + qpos * 128 + jm * 32 + ((col >> 3) & 1) * 16;
+}
+
+// inverse: image byte offset o (16-B granular) -> (m_local, col)
+__device__ __forceinline__ void img_inv(int o, int& m, int& col) {
+  const int kh = o >> 14;
+  const int nt = (o >> 10) & 15;
+  const int qpos = (o >> 7) & 7;
+  const int jm = (o >> 5) & 3;
+  const int u = (o >> 4) & 1;
+  const int q = ((qpos >> 2) & 1) + (qpos & 3) * 2;
+  m = kh * 32 + q * 4 + jm;
+  col = nt * 16 + u * 8;
+}
+
+__device__ __forceinline__ void glds16t(const bf16* g, char* lds_dst) {
+  typedef const __attribute__((address_space(1))) unsigned int* gp_t;
+  typedef __attribute__((address_space(3))) unsigned int* lp_t;
+  __builtin_amdgcn_global_load_lds((gp_t)(const void*)g, (lp_t)(void*)lds_dst, 16, 0, 0);
 }
 
 // 4 tr reads of one frag-quad column block: returns the two m-octet halves
@@ -93,42 +115,52 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
 
   f32x4_t acc[2][4][4] = {};  // [pr][mi][ni]
 
-  // Staging decode for this thread: 4 rounds, round r covers global chunk
-  // g = r*512 + tid -> m_local = g>>5, col = (g&31)*8.
-  int m_loc[4], col[4], io[4];
+  // Staging decode for this thread.  Fast path (full 64-row tiles): glds
+  // with the image permutation on the per-lane SOURCE address — round r
+  // covers image bytes [r*8192 + tid*16], whose content is the global chunk
+  // img_inv() decodes.  Tail tile: coalesced register loads (zero-filled
+  // out of range) + scattered ds_write_b128.
+  int m_loc[4], col[4], io[4];     // register-staging (tail) decode
+  int gm[4], gc[4];                // glds source decode per round
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int g = r * NTHREADS + tid;
     m_loc[r] = g >> 5;
     col[r] = (g & 31) * 8;
     io[r] = img_off(m_loc[r], col[r]);
+    img_inv(r * 8192 + tid * 16, gm[r], gc[r]);
   }
+  const int ldst16 = tid * 16;
+  auto stage_glds = [&](int t, int buf) {
+    const int64_t mt0 = m_begin + (int64_t)t * BMS;
+    char* dzi = DZs(buf);
+    char* xi = Xs(buf);
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      glds16t(DZ + (mt0 + gm[r]) * N + n0 + gc[r], dzi + r * 8192 + ldst16);
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      glds16t(X + (mt0 + gm[r]) * Kw + k0 + gc[r], xi + r * 8192 + ldst16);
+  };
 
   typedef short s8 __attribute__((ext_vector_type(8)));
-  s8 ldz[4], lx[4];
-  auto issue_loads = [&](int t) {
+  // Tail-tile staging (coalesced guarded loads + scattered LDS writes;
+  // round-by-round to keep register pressure low — the tail runs once).
+  auto stage_reg_tail = [&](int t, int buf) {
     const int64_t mt0 = m_begin + (int64_t)t * BMS;
+    char* dzi = DZs(buf);
+    char* xi = Xs(buf);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int64_t m = mt0 + m_loc[r];
+      s8 a = s8{}, b = s8{};
       if (m < m_end) {
-        ldz[r] = *reinterpret_cast<const s8*>(DZ + m * N + n0 + col[r]);
-        lx[r] = *reinterpret_cast<const s8*>(X + m * Kw + k0 + col[r]);
-      } else {
-        ldz[r] = s8{};
-        lx[r] = s8{};
+        a = *reinterpret_cast<const s8*>(DZ + m * N + n0 + col[r]);
+        b = *reinterpret_cast<const s8*>(X + m * Kw + k0 + col[r]);
       }
+      *reinterpret_cast<s8*>(dzi + io[r]) = a;
+      *reinterpret_cast<s8*>(xi + io[r]) = b;
     }
-  };
-  auto write_dz = [&](int buf) {
-    char* img = DZs(buf);
-#pragma unroll
-    for (int r = 0; r < 4; ++r) *reinterpret_cast<s8*>(img + io[r]) = ldz[r];
-  };
-  auto write_x = [&](int buf) {
-    char* img = Xs(buf);
-#pragma unroll
-    for (int r = 0; r < 4; ++r) *reinterpret_cast<s8*>(img + io[r]) = lx[r];
   };
 
   // Per-lane tr-read address term: lane*8 B — each lane points at its own
@@ -137,21 +169,30 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
   // (l>>4)*64 + (l&15) + j*16 relative to lane 0's address).
   const int lane_term = lane * 8;
 
-  // prologue: stage tile 0
-  issue_loads(0);
-  write_dz(0);
-  write_x(0);
-  __syncthreads();
+  // prologue: stage tile 0 (glds when the tile has all 64 rows)
+  const bool tile0_full = m_begin + BMS <= m_end;
+  if (tile0_full) {
+    stage_glds(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    stage_reg_tail(0, 0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  }
+  raw_barrier_tn();
 
   for (int t = 0; t < ntile; ++t) {
     const int buf = t & 1;
     const bool more = t + 1 < ntile;
+    const bool next_full = m_begin + (int64_t)(t + 2) * BMS <= m_end;
     bf16x8_t xb[4];  // X-side fragments, reused across the pr pair
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
       // ---------- phase (kh, pr=0) ----------
-      if (kh == 0 && more) issue_loads(t + 1);
       {
+        // staging for t+1: full tiles go by glds at the kh=0 pr=1 phase;
+        // the ragged tail tile goes by registers (loads here, writes in
+        // the kh=1 phases)
+        if (kh == 1 && more && !next_full) stage_reg_tail(t + 1, buf ^ 1);
         // B (X image) frags ni=0..3 at nt = wn*4+ni, plus A (dz) frags
         // mi=0..3 at nt = wm*4+mi (pr=0).  8+8 tr reads + lgkmcnt(0) in one
         // asm (form i).
@@ -193,7 +234,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
         ad[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
-        __syncthreads();
+        raw_barrier_tn();
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -204,8 +245,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
       }
       // ---------- phase (kh, pr=1) ----------
       {
-        if (kh == 0 && more) write_dz(buf ^ 1);
-        if (kh == 1 && more) write_x(buf ^ 1);
+        if (kh == 0 && more && next_full) stage_glds(t + 1, buf ^ 1);
         typedef const __attribute__((address_space(3))) char* lds_p;
         const lds_p adi = (lds_p)(const void*)(DZs(buf) + kh * 16384 + 8192 + wm * 4096 + lane_term);
         bf16x4_t a0l, a0h, a1l, a1h, a2l, a2h, a3l, a3h;
@@ -229,7 +269,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
         ad[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
-        __syncthreads();
+        // kh==1 (phase p3): drain the glds staged at p1 before the barrier
+        // that gates the next tile's reads
+        if (kh == 1 && more && next_full) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        raw_barrier_tn();
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -284,7 +327,7 @@ torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x) {
   }();
   int splitm = splitm_env > 0
                    ? splitm_env
-                   : (int)std::min<int64_t>((640 + tiles - 1) / tiles, (M + BMS - 1) / BMS);
+                   : (int)std::min<int64_t>((1280 + tiles - 1) / tiles, (M + BMS - 1) / BMS);
   if ((int64_t)splitm > (M + BMS - 1) / BMS) splitm = (int)((M + BMS - 1) / BMS);
   if (splitm < 1) splitm = 1;
   int64_t chunk = ((M + splitm - 1) / splitm + BMS - 1) / BMS * BMS;
