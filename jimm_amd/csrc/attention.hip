@@ -39,13 +39,21 @@ constexpr int QBLK = 64;    // q rows per strip (16 per wave)
 // staging passes, 2 waves/SIMD; 2 = 3 waves/SIMD, twice the staging
 // (JIMM_AMD_ATTN_NSTRIP selects; default 4)
 constexpr int KVBLK = 64;   // keys per LDS tile
-constexpr int D = 64;       // head_dim (checked host-side)
-constexpr int LDS_PITCH = D + 8;  // +8 shorts: bank-conflict pad for b128 reads
 
-template <bool CAUSAL, int NSTRIP>
+// Head dims beyond 64: template on the PADDED head dim
+// DP in {64, 96, 128} (d-fragments need multiples of 32); the runtime D
+// (72, 80 pad to 96) guards ragged loads/stores — 8-element chunks are
+// all-in or all-out since every supported D is a multiple of 8.
+template <int DP>
+__device__ __forceinline__ bf16x8_t ld8g(const bf16* p, int off, int Dr) {
+  if (DP == 64 || off + 8 <= Dr) return *reinterpret_cast<const bf16x8_t*>(p + off);
+  return bf16x8_t{};
+}
+
+template <bool CAUSAL, int NSTRIP, int DP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
-    bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale, int H,
+    bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale, int H, int Dr,
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t o_sb, int64_t o_sh, int64_t o_sl) {
   // (sb, sh, sl) = element strides of the (B,H,L,64) view; innermost dim is
@@ -57,10 +65,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   // them once per 64-q workgroup — 4x redundant global traffic at L=197,
   // ~125 of its 213 us bandwidth-bound). T14 split staging + single LDS
   // buffer; each wave keeps NSTRIP online-softmax states.
+  constexpr int LDS_PITCH = DP + 8;  // bank-conflict pad for b128 reads
+  constexpr int NS = DP / 32;        // d-steps per fragment contraction
+  constexpr int NT = DP / 16;        // d-tiles of the output
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds = reinterpret_cast<short*>(smem);                       // 64*72
-  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // 64*72
-  short* p_lds = vt_lds + D * LDS_PITCH;                               // 4*16*72
+  short* k_lds = reinterpret_cast<short*>(smem);                       // KVBLK rows
+  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // DP rows
+  short* p_lds = vt_lds + DP * LDS_PITCH;                              // 4*16 rows
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -82,7 +93,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   // ---- load Q fragments for every strip (B-operand of the swapped QK^T) --
   // strip t covers q rows [q_base + t*64, +64); this wave's rows:
   // q0(t) = q_base + t*64 + wave*16; B[d][q] = Q[q0+lo][32*s + hi*8 + j]
-  bf16x8_t qb[NSTRIP][2];
+  bf16x8_t qb[NSTRIP][NS];
   int nactive = 0;
 #pragma unroll
   for (int t = 0; t < NSTRIP; ++t) {
@@ -90,13 +101,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     if (ti < ntq) nactive = t + 1;
     const int qrow = min(min(ti, ntq - 1) * QBLK + wave * 16 + lo, Lq - 1);
 #pragma unroll
-    for (int s = 0; s < 2; ++s)
-      qb[t][s] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * s + hi * 8);
+    for (int s = 0; s < NS; ++s)
+      qb[t][s] = ld8g<DP>(qp + (int64_t)qrow * q_sl, 32 * s + hi * 8, Dr);
   }
 
   short* my_p = p_lds + wave * 16 * LDS_PITCH;
 
-  f32x4_t acc_o[NSTRIP][4] = {};  // O tiles: rows q = hi*4+r, cols d = 16*dt+lo
+  f32x4_t acc_o[NSTRIP][NT] = {};  // O tiles: rows q = hi*4+r, cols d = 16*dt+lo
   float m_run[NSTRIP], l_run[NSTRIP];
 #pragma unroll
   for (int t = 0; t < NSTRIP; ++t) {
@@ -108,30 +119,41 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int kv_end = CAUSAL ? min(Lk, (ti_max + 1) * QBLK) : Lk;
 
   // T14 staging state
+  constexpr int NCG = (DP + 63) / 64;  // 64-wide column groups per row
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
-  bf16x8_t kreg[2], vreg[2];
+  bf16x8_t kreg[NCG][2], vreg[NCG][2];
   bool st_valid;
   auto load_tile_regs = [&](int kv0) {
     const int key = kv0 + st_row;
     st_valid = key < Lk;
     const int krow = min(key, Lk - 1);
-    kreg[0] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)krow * k_sl + st_c0);
-    kreg[1] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)krow * k_sl + st_c0 + 8);
-    vreg[0] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)krow * v_sl + st_c0);
-    vreg[1] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)krow * v_sl + st_c0 + 8);
+#pragma unroll
+    for (int cg = 0; cg < NCG; ++cg) {
+      const int c0 = cg * 64 + st_c0;
+      if (c0 >= DP) continue;
+      kreg[cg][0] = ld8g<DP>(kp + (int64_t)krow * k_sl, c0, Dr);
+      kreg[cg][1] = ld8g<DP>(kp + (int64_t)krow * k_sl, c0 + 8, Dr);
+      vreg[cg][0] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0, Dr);
+      vreg[cg][1] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0 + 8, Dr);
+    }
   };
   auto write_tile = [&]() {
-    if (st_valid) {
-      *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + st_c0) = kreg[0];
-      *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + st_c0 + 8) = kreg[1];
 #pragma unroll
-      for (int i = 0; i < 8; ++i) vt_lds[(st_c0 + i) * LDS_PITCH + st_row] = vreg[0][i];
+    for (int cg = 0; cg < NCG; ++cg) {
+      const int c0 = cg * 64 + st_c0;
+      if (c0 >= DP) continue;
+      if (st_valid) {
+        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0) = kreg[cg][0];
+        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0 + 8) = kreg[cg][1];
 #pragma unroll
-      for (int i = 0; i < 8; ++i) vt_lds[(st_c0 + 8 + i) * LDS_PITCH + st_row] = vreg[1][i];
-    } else {
-      for (int i = 0; i < 16; ++i) k_lds[st_row * LDS_PITCH + st_c0 + i] = 0;
-      for (int i = 0; i < 16; ++i) vt_lds[(st_c0 + i) * LDS_PITCH + st_row] = 0;
+        for (int i = 0; i < 8; ++i) vt_lds[(c0 + i) * LDS_PITCH + st_row] = vreg[cg][0][i];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) vt_lds[(c0 + 8 + i) * LDS_PITCH + st_row] = vreg[cg][1][i];
+      } else {
+        for (int i = 0; i < 16; ++i) k_lds[st_row * LDS_PITCH + c0 + i] = 0;
+        for (int i = 0; i < 16; ++i) vt_lds[(c0 + i) * LDS_PITCH + st_row] = 0;
+      }
     }
   };
 
@@ -155,7 +177,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
 #pragma unroll
-        for (int s = 0; s < 2; ++s) {
+        for (int s = 0; s < NS; ++s) {
           const bf16x8_t ka =
               *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * t + lo) * LDS_PITCH + 32 * s + hi * 8);
           sc[t] = MFMA16(ka, qb[st][s], sc[t]);
@@ -212,7 +234,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) alpha_r[r] = __shfl(alpha, hi * 4 + r, WAVE);
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt)
+      for (int dt = 0; dt < NT; ++dt)
 #pragma unroll
         for (int r = 0; r < 4; ++r) acc_o[st][dt][r] *= alpha_r[r];
 
@@ -223,7 +245,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         const bf16x8_t pa =
             *reinterpret_cast<const bf16x8_t*>(my_p + lo * LDS_PITCH + 32 * s + hi * 8);
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
+        for (int dt = 0; dt < NT; ++dt) {
           const bf16x8_t vb =
               *reinterpret_cast<const bf16x8_t*>(vt_lds + (16 * dt + lo) * LDS_PITCH + 32 * s + hi * 8);
           acc_o[st][dt] = MFMA16(pa, vb, acc_o[st][dt]);
@@ -255,8 +277,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const int qrow = q0 + hi * 4 + r;
       if (qrow >= Lq) continue;
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt)
-        op[(int64_t)qrow * o_sl + 16 * dt + lo] = f2bf(acc_o[st][dt][r] * invl_r[r]);
+      for (int dt = 0; dt < NT; ++dt)
+        if (DP == 64 || 16 * dt + lo < Dr)
+          op[(int64_t)qrow * o_sl + 16 * dt + lo] = f2bf(acc_o[st][dt][r] * invl_r[r]);
     }
     if (hi == 0 && q0 + lo < Lq)
       lse[bh * Lq + q0 + lo] = m_run[st] + __logf(l_run[st]);
@@ -272,47 +295,53 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   // un-copied (B,L,H,64) / fused-qkv (B,L,3,H,64) layouts.
   TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only, got ", q.scalar_type());
-  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "attn_fwd: (B,H,L,64) expected");
+  TORCH_CHECK(q.dim() == 4, "attn_fwd: (B,H,L,D) expected");
+  const int Dr = q.size(3);
+  TORCH_CHECK(Dr == 64 || Dr == 72 || Dr == 80 || Dr == 96 || Dr == 128,
+              "attn_fwd: head_dim must be one of {64,72,80,96,128}, got ", Dr);
+  const int DP = Dr <= 64 ? 64 : (Dr <= 96 ? 96 : 128);
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
   TORCH_CHECK(k.size(0) == B && k.size(1) == H && v.size(2) == Lk);
   for (auto* t : {&q, &k, &v})
     TORCH_CHECK(t->stride(3) == 1, "attn_fwd: innermost dim must be contiguous");
   // O written in (B,L,H,64) memory order so the model's (B,L,H*64) reshape is
   // a free view (no permute copy)
-  auto o_storage = torch::empty({B, Lq, H, 64}, q.options());
+  auto o_storage = torch::empty({B, Lq, H, Dr}, q.options());
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
   // 4 strips only when the q tiles divide evenly (measured: 4 wins at
   // L=197 — one staging pass; 2 wins elsewhere via 3 waves/SIMD occupancy:
   // L=577 261->298 TF/s, L=257 184->216, L=77 46->60)
   const int ntq = (Lq + QBLK - 1) / QBLK;
-  int nstrip = (ntq % 4 == 0) ? 4 : 2;
-  if (const char* e = getenv("JIMM_AMD_ATTN_NSTRIP")) nstrip = (std::string(e) == "2") ? 2 : 4;
+  int nstrip = (DP == 64 && ntq % 4 == 0) ? 4 : 2;
+  if (DP == 64) {
+    if (const char* e = getenv("JIMM_AMD_ATTN_NSTRIP")) nstrip = (std::string(e) == "2") ? 2 : 4;
+  }
   const dim3 grid((ntq + nstrip - 1) / nstrip, (unsigned)((int64_t)B * H));
-  const size_t shmem = (2 * KVBLK * LDS_PITCH + 2 * D * LDS_PITCH + 4 * 16 * LDS_PITCH) * sizeof(short);
+  const size_t shmem = ((KVBLK + DP + 4 * 16) * (DP + 8)) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
+#define ATTN_ARGS                                                                          \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                          \
+                     reinterpret_cast<bf16*>(o_storage.data_ptr()), lse.data_ptr<float>(), \
+                     Lq, Lk, (float)scale, H, Dr, q.stride(0), q.stride(1), q.stride(2),   \
+                     k.stride(0), k.stride(1), k.stride(2),                                \
+                     v.stride(0), v.stride(1), v.stride(2),                                \
+                     o.stride(0), o.stride(1), o.stride(2)
 #define ATTN_LAUNCH(C)                                                                     \
-  if (nstrip == 2)                                                                         \
-    hipLaunchKernelGGL((attn_fwd_kernel<C, 2>), grid, dim3(256), shmem, stream,            \
-                     reinterpret_cast<const bf16*>(q.data_ptr()),                          \
-                     reinterpret_cast<const bf16*>(k.data_ptr()),                          \
-                     reinterpret_cast<const bf16*>(v.data_ptr()),                          \
-                     reinterpret_cast<bf16*>(o_storage.data_ptr()), lse.data_ptr<float>(), \
-                     Lq, Lk, (float)scale, H, q.stride(0), q.stride(1), q.stride(2),       \
-                     k.stride(0), k.stride(1), k.stride(2),                                \
-                     v.stride(0), v.stride(1), v.stride(2),                                \
-                     o.stride(0), o.stride(1), o.stride(2));                               \
-  else                                                                                     \
-  hipLaunchKernelGGL((attn_fwd_kernel<C, 4>), grid, dim3(256), shmem, stream,              \
-                     reinterpret_cast<const bf16*>(q.data_ptr()),                          \
-                     reinterpret_cast<const bf16*>(k.data_ptr()),                          \
-                     reinterpret_cast<const bf16*>(v.data_ptr()),                          \
-                     reinterpret_cast<bf16*>(o_storage.data_ptr()), lse.data_ptr<float>(), \
-                     Lq, Lk, (float)scale, H, q.stride(0), q.stride(1), q.stride(2),       \
-                     k.stride(0), k.stride(1), k.stride(2),                                \
-                     v.stride(0), v.stride(1), v.stride(2),                                \
-                     o.stride(0), o.stride(1), o.stride(2))
+  do {                                                                                     \
+    if (DP == 128)                                                                         \
+      hipLaunchKernelGGL((attn_fwd_kernel<C, 2, 128>), grid, dim3(256), shmem, stream, ATTN_ARGS); \
+    else if (DP == 96)                                                                     \
+      hipLaunchKernelGGL((attn_fwd_kernel<C, 2, 96>), grid, dim3(256), shmem, stream, ATTN_ARGS);  \
+    else if (nstrip == 2)                                                                  \
+      hipLaunchKernelGGL((attn_fwd_kernel<C, 2, 64>), grid, dim3(256), shmem, stream, ATTN_ARGS);  \
+    else                                                                                   \
+      hipLaunchKernelGGL((attn_fwd_kernel<C, 4, 64>), grid, dim3(256), shmem, stream, ATTN_ARGS);  \
+  } while (0)
   if (causal) ATTN_LAUNCH(true); else ATTN_LAUNCH(false);
 #undef ATTN_LAUNCH
+#undef ATTN_ARGS
   return {o, lse};
 }

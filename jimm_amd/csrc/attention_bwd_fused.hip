@@ -38,18 +38,24 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
 constexpr int BLK = 64;            // q rows / keys per workgroup tile
-// NKV (kv tiles per dkv workgroup) is a template param: 2 = half the Q/dO
-// staging passes at 2 waves/SIMD; 1 = 3 waves/SIMD occupancy
-constexpr int NQS = 2;             // q tiles per dq workgroup
-constexpr int D = 64;              // head_dim (checked host-side)
-constexpr int PITCH = D + 8;       // bank-conflict pad for b128 reads
+
+// Head dims beyond 64: padded-D template DP in {64,96,128}; runtime Dr
+// guards ragged chunks (all supported D are multiples of 8, so 8-element
+// chunks are all-in or all-out). DP=64 keeps the round-1 fast path
+// (NKV=2 / NQS=2 with cached fragments); DP>64 uses 1 tile per workgroup
+// with inline fragment reads to stay inside the register budget.
+template <int DP>
+__device__ __forceinline__ bf16x8_t ld8gb(const bf16* p, int off, int Dr) {
+  if (DP == 64 || off + 8 <= Dr) return *reinterpret_cast<const bf16x8_t*>(p + off);
+  return bf16x8_t{};
+}
 
 // ---------------------------------------------------------------------------
 // D = rowsum(dO * O), stride-aware
 // ---------------------------------------------------------------------------
 
 __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
-                               float* __restrict__ Dv, int H, int L, int64_t nrows,
+                               float* __restrict__ Dv, int H, int L, int Dr, int64_t nrows,
                                int64_t do_sb, int64_t do_sh, int64_t do_sl,
                                int64_t o_sb, int64_t o_sh, int64_t o_sl) {
   // 8 rows per wave (all 64 lanes loading): lane -> (sub-row l>>3, 8 cols)
@@ -57,7 +63,7 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
   const int lane = threadIdx.x % WAVE;
   const int wpb = blockDim.x / WAVE;
   const int sub = lane >> 3;          // 0..7 row within the wave's group
-  const int c0 = (lane & 7) * 8;      // 8 elements per lane
+  const int c0 = (lane & 7) * 8;      // 8 elements per lane per 64-chunk
   for (int64_t base = ((int64_t)blockIdx.x * wpb + wave) * 8; base < nrows;
        base += (int64_t)gridDim.x * wpb * 8) {
     const int64_t row = base + sub;
@@ -67,12 +73,15 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
     const int l = (int)(rc % L);
     const bf16* a = dO + b * do_sb + h * do_sh + (int64_t)l * do_sl;
     const bf16* o = O + b * o_sb + h * o_sh + (int64_t)l * o_sl;
-    float av[8], ov[8];
-    vload_f32<8>(a + c0, av);
-    vload_f32<8>(o + c0, ov);
     float acc = 0.f;
+    for (int cg = 0; cg < Dr; cg += 64) {
+      if (cg + c0 + 8 > Dr) continue;
+      float av[8], ov[8];
+      vload_f32<8>(a + cg + c0, av);
+      vload_f32<8>(o + cg + c0, ov);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) acc += av[j] * ov[j];
+      for (int j = 0; j < 8; ++j) acc += av[j] * ov[j];
+    }
     // reduce across the 8 lanes of this sub-row
 #pragma unroll
     for (int off = 1; off < 8; off <<= 1) acc += __shfl_xor(acc, off, WAVE);
@@ -84,11 +93,12 @@ __global__ void attn_d2_kernel(const bf16* __restrict__ dO, const bf16* __restri
 // dK/dV kernel: workgroup owns keys [kv0, kv0+64); loops q tiles
 // ---------------------------------------------------------------------------
 
-template <bool CAUSAL, int NKV>
+template <bool CAUSAL, int NKV, int DP>
 __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int Lq, int Lk, float scale, int H,
+    int Dr,
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dk_sb, int64_t dk_sh, int64_t dk_sl, int64_t dv_sb, int64_t dv_sh, int64_t dv_sl) {
@@ -98,11 +108,15 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   // K/V live as per-strip A-FRAGMENTS in registers — no K/V LDS at all.
   // LDS: Q^T/dO^T (transposed) + Q/dO (row) images of the current q tile,
   // per-wave P/dS tile. ~46 KiB.
+  constexpr int PITCH = DP + 8;
+  constexpr int NS = DP / 32;
+  constexpr int NT = DP / 16;
+  constexpr int NCG = (DP + 63) / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* qt_lds = reinterpret_cast<short*>(smem);   // Q^T  [64 d][72]
-  short* dot_lds = qt_lds + D * PITCH;              // dO^T [64 d][72]
-  short* qr_lds = dot_lds + D * PITCH;              // Q    [64 q][72]
-  short* dor_lds = qr_lds + BLK * PITCH;            // dO   [64 q][72]
+  short* qt_lds = reinterpret_cast<short*>(smem);   // Q^T  [DP d][PITCH]
+  short* dot_lds = qt_lds + DP * PITCH;             // dO^T [DP d][PITCH]
+  short* qr_lds = dot_lds + DP * PITCH;             // Q    [64 q][PITCH]
+  short* dor_lds = qr_lds + BLK * PITCH;            // dO   [64 q][PITCH]
   short* p_lds = dor_lds + BLK * PITCH;
 
   const int tid = threadIdx.x;
@@ -124,7 +138,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   // strip s covers kv tile ti(s) = blockIdx.x + s*gridDim.x; this wave's 16
   // keys of that tile: rows 16*wave + lo. Invalid rows clamp; their P/dS
   // contributions are masked to zero and their stores are guarded.
-  bf16x8_t ka[NKV][2], va[NKV][2];
+  bf16x8_t ka[NKV][NS], va[NKV][NS];
   int kvbase[NKV];
   int nactive = 0;
 #pragma unroll
@@ -134,16 +148,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     if (ti < ntk) nactive = s + 1;
     const int key = min(min(ti, ntk - 1) * BLK + 16 * wave + lo, Lk - 1);
 #pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      ka[s][t] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + 32 * t + hi * 8);
-      va[s][t] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + 32 * t + hi * 8);
+    for (int t = 0; t < NS; ++t) {
+      ka[s][t] = ld8gb<DP>(kp + (int64_t)key * k_sl, 32 * t + hi * 8, Dr);
+      va[s][t] = ld8gb<DP>(vp + (int64_t)key * v_sl, 32 * t + hi * 8, Dr);
     }
   }
 
   short* my_p = p_lds + wave * 16 * PITCH;
 
-  f32x4_t acc_dk[NKV][4] = {};  // rows key = 16*wave + hi*4+r, cols d = 16*dt+lo
-  f32x4_t acc_dv[NKV][4] = {};
+  f32x4_t acc_dk[NKV][NT] = {};  // rows key = 16*wave + hi*4+r, cols d = 16*dt+lo
+  f32x4_t acc_dv[NKV][NT] = {};
 
   // causal: the earliest kv tile of this WG bounds the first useful q tile
   const int q_start = CAUSAL ? (kvbase[0] / BLK) * BLK : 0;
@@ -152,35 +166,44 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   // per-thread staging slot for the Q/dO images
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
-  bf16x8_t qreg[2], doreg[2];
+  bf16x8_t qreg[NCG][2], doreg[NCG][2];
   bool st_valid;
   auto load_stage_regs = [&](int q0) {
     const int qi = q0 + st_row;
     st_valid = qi < Lq;
     const int qr = min(qi, Lq - 1);
 #pragma unroll
-    for (int hh = 0; hh < 2; ++hh) {
-      qreg[hh] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qr * q_sl + st_c0 + hh * 8);
-      doreg[hh] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qr * do_sl + st_c0 + hh * 8);
+    for (int cg = 0; cg < NCG; ++cg) {
+      if (cg * 64 + st_c0 >= DP) continue;
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+        qreg[cg][hh] = ld8gb<DP>(qp + (int64_t)qr * q_sl, cg * 64 + st_c0 + hh * 8, Dr);
+        doreg[cg][hh] = ld8gb<DP>(dop + (int64_t)qr * do_sl, cg * 64 + st_c0 + hh * 8, Dr);
+      }
     }
   };
   auto write_stage = [&]() {
-    if (st_valid) {
 #pragma unroll
-      for (int hh = 0; hh < 2; ++hh) {
-        *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + st_c0 + hh * 8) = qreg[hh];
-        *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + st_c0 + hh * 8) = doreg[hh];
+    for (int cg = 0; cg < NCG; ++cg) {
+      const int c0 = cg * 64 + st_c0;
+      if (c0 >= DP) continue;
+      if (st_valid) {
 #pragma unroll
-        for (int i = 0; i < 8; ++i) qt_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = qreg[hh][i];
+        for (int hh = 0; hh < 2; ++hh) {
+          *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + c0 + hh * 8) = qreg[cg][hh];
+          *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + c0 + hh * 8) = doreg[cg][hh];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) dot_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = doreg[hh][i];
-      }
-    } else {
-      for (int i = 0; i < 16; ++i) {
-        qr_lds[st_row * PITCH + st_c0 + i] = 0;
-        dor_lds[st_row * PITCH + st_c0 + i] = 0;
-        qt_lds[(st_c0 + i) * PITCH + st_row] = 0;
-        dot_lds[(st_c0 + i) * PITCH + st_row] = 0;
+          for (int i = 0; i < 8; ++i) qt_lds[(c0 + hh * 8 + i) * PITCH + st_row] = qreg[cg][hh][i];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) dot_lds[(c0 + hh * 8 + i) * PITCH + st_row] = doreg[cg][hh][i];
+        }
+      } else {
+        for (int i = 0; i < 16; ++i) {
+          qr_lds[st_row * PITCH + c0 + i] = 0;
+          dor_lds[st_row * PITCH + c0 + i] = 0;
+          qt_lds[(c0 + i) * PITCH + st_row] = 0;
+          dot_lds[(c0 + i) * PITCH + st_row] = 0;
+        }
       }
     }
   };
@@ -194,13 +217,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const int q0 = q_start + it * BLK;
     // ---- B-fragments of Q^T and dO^T from the LDS row images (shared by
     // every kv strip) ------------------------------------------------------
-    bf16x8_t qb[4][2], dob[4][2];
+    // DP==64 caches the q-tile fragments across the NKV strips; larger DP
+    // reads them inline per use (the 4xNS cache would spill at NS=4)
+    bf16x8_t qb[DP == 64 ? 4 : 1][NS], dob[DP == 64 ? 4 : 1][NS];
+    if constexpr (DP == 64) {
 #pragma unroll
-    for (int qt = 0; qt < 4; ++qt) {
+      for (int qt = 0; qt < 4; ++qt) {
 #pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qr_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
-        dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dor_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+        for (int s = 0; s < NS; ++s) {
+          qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qr_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+          dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dor_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+        }
       }
     }
 
@@ -218,9 +245,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         f32x4_t sc = {};
         f32x4_t dpc = {};
 #pragma unroll
-        for (int s = 0; s < 2; ++s) {
-          sc = MFMA16(ka[sidx][s], qb[qt][s], sc);
-          dpc = MFMA16(va[sidx][s], dob[qt][s], dpc);
+        for (int s = 0; s < NS; ++s) {
+          bf16x8_t qf, dof;
+          if constexpr (DP == 64) {
+            qf = qb[qt][s];
+            dof = dob[qt][s];
+          } else {
+            qf = *reinterpret_cast<const bf16x8_t*>(qr_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+            dof = *reinterpret_cast<const bf16x8_t*>(dor_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+          }
+          sc = MFMA16(ka[sidx][s], qf, sc);
+          dpc = MFMA16(va[sidx][s], dof, dpc);
         }
         __builtin_amdgcn_s_setprio(0);
         const int qi = q0 + 16 * qt + lo;
@@ -246,7 +281,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       for (int s = 0; s < 2; ++s) {
         const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
+        for (int dt = 0; dt < NT; ++dt) {
           const bf16x8_t bfrag =
               *reinterpret_cast<const bf16x8_t*>(dot_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
           acc_dv[sidx][dt] = MFMA16(pa, bfrag, acc_dv[sidx][dt]);
@@ -265,7 +300,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       for (int s = 0; s < 2; ++s) {
         const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
+        for (int dt = 0; dt < NT; ++dt) {
           const bf16x8_t bfrag =
               *reinterpret_cast<const bf16x8_t*>(qt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
           acc_dk[sidx][dt] = MFMA16(dsa, bfrag, acc_dk[sidx][dt]);
@@ -293,7 +328,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       const int key = kvbase[sidx] + 16 * wave + hi * 4 + r;
       if (key >= Lk) continue;
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
+      for (int dt = 0; dt < NT; ++dt) {
+        if (DP != 64 && 16 * dt + lo >= Dr) continue;
         dkp[(int64_t)key * dk_sl + 16 * dt + lo] = f2bf(acc_dk[sidx][dt][r]);
         dvp[(int64_t)key * dv_sl + 16 * dt + lo] = f2bf(acc_dv[sidx][dt][r]);
       }
@@ -305,11 +341,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 // dQ kernel: workgroup owns q rows [q0, q0+64); loops kv tiles
 // ---------------------------------------------------------------------------
 
-template <bool CAUSAL>
+template <bool CAUSAL, int NQS, int DP>
 __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     const bf16* __restrict__ dO, const float* __restrict__ lse, const float* __restrict__ Dv,
-    bf16* __restrict__ dq, int Lq, int Lk, float scale, int H,
+    bf16* __restrict__ dq, int Lq, int Lk, float scale, int H, int Dr,
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t dq_sb, int64_t dq_sh, int64_t dq_sl) {
@@ -317,10 +353,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   // images are staged ONCE for all of them; Q/dO A-fragments live in
   // registers per strip. LDS: K^T + K/V row images of the current kv tile,
   // per-wave dS tile.
+  constexpr int PITCH = DP + 8;
+  constexpr int NS = DP / 32;
+  constexpr int NT = DP / 16;
+  constexpr int NCG = (DP + 63) / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* kt_lds = reinterpret_cast<short*>(smem);     // K^T [64 d][72]
-  short* kr_lds = kt_lds + D * PITCH;                 // K   [64 key][72]
-  short* vr_lds = kr_lds + BLK * PITCH;               // V   [64 key][72]
+  short* kt_lds = reinterpret_cast<short*>(smem);     // K^T [DP d][PITCH]
+  short* kr_lds = kt_lds + DP * PITCH;                // K   [64 key][PITCH]
+  short* vr_lds = kr_lds + BLK * PITCH;               // V   [64 key][PITCH]
   short* ds_lds = vr_lds + BLK * PITCH;
 
   const int tid = threadIdx.x;
@@ -341,7 +381,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   // ---- per-strip Q/dO A-fragments + per-row lse/D -------------------------
   // strip s covers q tile ti(s) = blockIdx.x + s*gridDim.x; this wave's 16
   // q rows of that tile: rows 16*wave + ...
-  bf16x8_t qa[NQS][2], doa[NQS][2];
+  bf16x8_t qa[NQS][NS], doa[NQS][NS];
   float lse_r[NQS][4], d_r[NQS][4];
   int qbase[NQS];
   int nactive = 0;
@@ -353,9 +393,9 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const int q0 = min(ti, ntq - 1) * BLK + wave * 16;
     const int qrow = min(q0 + lo, Lq - 1);
 #pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      qa[s][t] = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 * t + hi * 8);
-      doa[s][t] = *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qrow * do_sl + 32 * t + hi * 8);
+    for (int t = 0; t < NS; ++t) {
+      qa[s][t] = ld8gb<DP>(qp + (int64_t)qrow * q_sl, 32 * t + hi * 8, Dr);
+      doa[s][t] = ld8gb<DP>(dop + (int64_t)qrow * do_sl, 32 * t + hi * 8, Dr);
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -366,7 +406,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   }
 
   short* my_ds = ds_lds + wave * 16 * PITCH;
-  f32x4_t acc_dq[NQS][4] = {};  // rows q = hi*4+r, cols d = 16*dt + lo
+  f32x4_t acc_dq[NQS][NT] = {};  // rows q = hi*4+r, cols d = 16*dt + lo
 
   // causal: only kv tiles up to the LAST active strip's diagonal are needed
   const int ti_max = blockIdx.x + (nactive - 1) * gridDim.x;
@@ -375,32 +415,41 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
-  bf16x8_t kreg[2], vreg[2];
+  bf16x8_t kreg[NCG][2], vreg[NCG][2];
   bool st_valid;
   auto load_stage_regs = [&](int kv0) {
     const int key = kv0 + st_row;
     st_valid = key < Lk;
     const int kr = min(key, Lk - 1);
 #pragma unroll
-    for (int hh = 0; hh < 2; ++hh) {
-      kreg[hh] = *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)kr * k_sl + st_c0 + hh * 8);
-      vreg[hh] = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)kr * v_sl + st_c0 + hh * 8);
+    for (int cg = 0; cg < NCG; ++cg) {
+      if (cg * 64 + st_c0 >= DP) continue;
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+        kreg[cg][hh] = ld8gb<DP>(kp + (int64_t)kr * k_sl, cg * 64 + st_c0 + hh * 8, Dr);
+        vreg[cg][hh] = ld8gb<DP>(vp + (int64_t)kr * v_sl, cg * 64 + st_c0 + hh * 8, Dr);
+      }
     }
   };
   auto write_stage = [&]() {
-    if (st_valid) {
 #pragma unroll
-      for (int hh = 0; hh < 2; ++hh) {
-        *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + st_c0 + hh * 8) = kreg[hh];
-        *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + st_c0 + hh * 8) = vreg[hh];
+    for (int cg = 0; cg < NCG; ++cg) {
+      const int c0 = cg * 64 + st_c0;
+      if (c0 >= DP) continue;
+      if (st_valid) {
 #pragma unroll
-        for (int i = 0; i < 8; ++i) kt_lds[(st_c0 + hh * 8 + i) * PITCH + st_row] = kreg[hh][i];
-      }
-    } else {
-      for (int i = 0; i < 16; ++i) {
-        kr_lds[st_row * PITCH + st_c0 + i] = 0;
-        vr_lds[st_row * PITCH + st_c0 + i] = 0;
-        kt_lds[(st_c0 + i) * PITCH + st_row] = 0;
+        for (int hh = 0; hh < 2; ++hh) {
+          *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + c0 + hh * 8) = kreg[cg][hh];
+          *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + c0 + hh * 8) = vreg[cg][hh];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kt_lds[(c0 + hh * 8 + i) * PITCH + st_row] = kreg[cg][hh][i];
+        }
+      } else {
+        for (int i = 0; i < 16; ++i) {
+          kr_lds[st_row * PITCH + c0 + i] = 0;
+          vr_lds[st_row * PITCH + c0 + i] = 0;
+          kt_lds[(c0 + i) * PITCH + st_row] = 0;
+        }
       }
     }
   };
@@ -413,13 +462,15 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * BLK;
     // ---- B-fragments of K^T and V^T from the LDS row images (shared) -----
-    bf16x8_t kb[4][2], vb[4][2];
+    bf16x8_t kb[DP == 64 ? 4 : 1][NS], vb[DP == 64 ? 4 : 1][NS];
+    if constexpr (DP == 64) {
 #pragma unroll
-    for (int kt = 0; kt < 4; ++kt) {
+      for (int kt = 0; kt < 4; ++kt) {
 #pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
-        vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+        for (int s = 0; s < NS; ++s) {
+          kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+          vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+        }
       }
     }
 
@@ -436,9 +487,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
         f32x4_t sc = {};
         f32x4_t dpc = {};
 #pragma unroll
-        for (int s = 0; s < 2; ++s) {
-          sc = MFMA16(qa[sidx][s], kb[kt][s], sc);
-          dpc = MFMA16(doa[sidx][s], vb[kt][s], dpc);
+        for (int s = 0; s < NS; ++s) {
+          bf16x8_t kf, vf;
+          if constexpr (DP == 64) {
+            kf = kb[kt][s];
+            vf = vb[kt][s];
+          } else {
+            kf = *reinterpret_cast<const bf16x8_t*>(kr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+            vf = *reinterpret_cast<const bf16x8_t*>(vr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+          }
+          sc = MFMA16(qa[sidx][s], kf, sc);
+          dpc = MFMA16(doa[sidx][s], vf, dpc);
         }
         __builtin_amdgcn_s_setprio(0);
         const int key = kv0 + 16 * kt + lo;
@@ -460,7 +519,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       for (int s = 0; s < 2; ++s) {
         const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_ds + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
+        for (int dt = 0; dt < NT; ++dt) {
           const bf16x8_t bfrag =
               *reinterpret_cast<const bf16x8_t*>(kt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
           acc_dq[sidx][dt] = MFMA16(dsa, bfrag, acc_dq[sidx][dt]);
@@ -488,8 +547,9 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       const int qi = qbase[sidx] + wave * 16 + hi * 4 + r;
       if (qi >= Lq) continue;
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt)
-        dqp[(int64_t)qi * dq_sl + 16 * dt + lo] = f2bf(acc_dq[sidx][dt][r]);
+      for (int dt = 0; dt < NT; ++dt)
+        if (DP == 64 || 16 * dt + lo < Dr)
+          dqp[(int64_t)qi * dq_sl + 16 * dt + lo] = f2bf(acc_dq[sidx][dt][r]);
     }
   }
 }
@@ -501,7 +561,11 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
                     torch::Tensor dv, bool causal, double scale) {
   // all tensors (B,H,L,64) views, innermost contiguous, arbitrary b/h/l strides
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
-  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "attn_bwd_fused: (B,H,L,64) expected");
+  TORCH_CHECK(q.dim() == 4, "attn_bwd_fused: (B,H,L,D) expected");
+  const int Dr = q.size(3);
+  TORCH_CHECK(Dr == 64 || Dr == 72 || Dr == 80 || Dr == 96 || Dr == 128,
+              "attn_bwd_fused: head_dim must be one of {64,72,80,96,128}, got ", Dr);
+  const int DP = Dr <= 64 ? 64 : (Dr <= 96 ? 96 : 128);
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), Lk = k.size(2);
   for (auto* t : {&q, &k, &v, &o, &dO, &dq, &dk, &dv})
     TORCH_CHECK(t->stride(3) == 1, "attn_bwd_fused: innermost dim must be contiguous");
@@ -516,58 +580,73 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
     hipLaunchKernelGGL(attn_d2_kernel, grid, dim3(256), 0, stream,
                        reinterpret_cast<const bf16*>(dO.data_ptr()),
                        reinterpret_cast<const bf16*>(o.data_ptr()), Dv.data_ptr<float>(), H, Lq,
-                       nrows, dO.stride(0), dO.stride(1), dO.stride(2),
+                       Dr, nrows, dO.stride(0), dO.stride(1), dO.stride(2),
                        o.stride(0), o.stride(1), o.stride(2));
   }
 
-  const size_t shmem_dkv = (4 * BLK * PITCH + 4 * 16 * PITCH) * sizeof(short);
-  const size_t shmem_dq = (3 * D * PITCH + 4 * 16 * PITCH) * sizeof(short);
+  const int pitch = DP + 8;
+  const size_t shmem_dkv = ((2 * DP + 2 * BLK + 4 * 16) * pitch) * sizeof(short);
+  const size_t shmem_dq = ((DP + 2 * BLK + 4 * 16) * pitch) * sizeof(short);
   const int ntk = (Lk + BLK - 1) / BLK;
   static const int nkv_env = [] {
     const char* e = getenv("JIMM_AMD_ATTN_NKV");
     return e ? atoi(e) : 0;
   }();
-  const int nkv = nkv_env ? nkv_env : 2;
+  // DP > 64: one tile per workgroup (register budget)
+  const int nkv = DP != 64 ? 1 : (nkv_env ? nkv_env : 2);
+  const int nqs = DP != 64 ? 1 : 2;
   const dim3 grid_dkv((ntk + nkv - 1) / nkv, (unsigned)((int64_t)B * H));
   const int ntq = (Lq + BLK - 1) / BLK;
-  const dim3 grid_dq((ntq + NQS - 1) / NQS, (unsigned)((int64_t)B * H));
+  const dim3 grid_dq((ntq + nqs - 1) / nqs, (unsigned)((int64_t)B * H));
 
+#define DKV_ARGS                                                                             \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                            \
+                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
+                     Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dk.data_ptr()),           \
+                     reinterpret_cast<bf16*>(dv.data_ptr()), Lq, Lk, (float)scale, H, Dr,    \
+                     q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),        \
+                     k.stride(2), v.stride(0), v.stride(1), v.stride(2), dO.stride(0),       \
+                     dO.stride(1), dO.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),   \
+                     dv.stride(0), dv.stride(1), dv.stride(2)
 #define DKV_LAUNCH(C)                                                                        \
-  if (nkv == 1)                                                                              \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 1>), grid_dkv, dim3(256), shmem_dkv, stream,  \
-                     reinterpret_cast<const bf16*>(q.data_ptr()),                            \
-                     reinterpret_cast<const bf16*>(k.data_ptr()),                            \
-                     reinterpret_cast<const bf16*>(v.data_ptr()),                            \
-                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
-                     Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dk.data_ptr()),           \
-                     reinterpret_cast<bf16*>(dv.data_ptr()), Lq, Lk, (float)scale, H,        \
-                     q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),        \
-                     k.stride(2), v.stride(0), v.stride(1), v.stride(2), dO.stride(0),       \
-                     dO.stride(1), dO.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),   \
-                     dv.stride(0), dv.stride(1), dv.stride(2));                              \
-  else                                                                                       \
-  hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 2>), grid_dkv, dim3(256), shmem_dkv, stream,    \
-                     reinterpret_cast<const bf16*>(q.data_ptr()),                            \
-                     reinterpret_cast<const bf16*>(k.data_ptr()),                            \
-                     reinterpret_cast<const bf16*>(v.data_ptr()),                            \
-                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
-                     Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dk.data_ptr()),           \
-                     reinterpret_cast<bf16*>(dv.data_ptr()), Lq, Lk, (float)scale, H,        \
-                     q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),        \
-                     k.stride(2), v.stride(0), v.stride(1), v.stride(2), dO.stride(0),       \
-                     dO.stride(1), dO.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),   \
-                     dv.stride(0), dv.stride(1), dv.stride(2))
-#define DQ_LAUNCH(C)                                                                         \
-  hipLaunchKernelGGL((attn_bwd_dq_kernel<C>), grid_dq, dim3(256), shmem_dq, stream,          \
+  do {                                                                                       \
+    if (DP == 128)                                                                           \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 1, 128>), grid_dkv, dim3(256), shmem_dkv,   \
+                         stream, DKV_ARGS);                                                  \
+    else if (DP == 96)                                                                       \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 1, 96>), grid_dkv, dim3(256), shmem_dkv,    \
+                         stream, DKV_ARGS);                                                  \
+    else if (nkv == 1)                                                                       \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 1, 64>), grid_dkv, dim3(256), shmem_dkv,    \
+                         stream, DKV_ARGS);                                                  \
+    else                                                                                     \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<C, 2, 64>), grid_dkv, dim3(256), shmem_dkv,    \
+                         stream, DKV_ARGS);                                                  \
+  } while (0)
+#define DQ_ARGS                                                                              \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                            \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                            \
                      reinterpret_cast<const bf16*>(v.data_ptr()),                            \
                      reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),    \
                      Dv.data_ptr<float>(), reinterpret_cast<bf16*>(dq.data_ptr()),           \
-                     Lq, Lk, (float)scale, H, q.stride(0), q.stride(1), q.stride(2),         \
+                     Lq, Lk, (float)scale, H, Dr, q.stride(0), q.stride(1), q.stride(2),     \
                      k.stride(0), k.stride(1), k.stride(2), v.stride(0), v.stride(1),        \
                      v.stride(2), dO.stride(0), dO.stride(1), dO.stride(2),                  \
-                     dq.stride(0), dq.stride(1), dq.stride(2))
+                     dq.stride(0), dq.stride(1), dq.stride(2)
+#define DQ_LAUNCH(C)                                                                         \
+  do {                                                                                       \
+    if (DP == 128)                                                                           \
+      hipLaunchKernelGGL((attn_bwd_dq_kernel<C, 1, 128>), grid_dq, dim3(256), shmem_dq,      \
+                         stream, DQ_ARGS);                                                   \
+    else if (DP == 96)                                                                       \
+      hipLaunchKernelGGL((attn_bwd_dq_kernel<C, 1, 96>), grid_dq, dim3(256), shmem_dq,       \
+                         stream, DQ_ARGS);                                                   \
+    else                                                                                     \
+      hipLaunchKernelGGL((attn_bwd_dq_kernel<C, 2, 64>), grid_dq, dim3(256), shmem_dq,       \
+                         stream, DQ_ARGS);                                                   \
+  } while (0)
   if (causal) {
     DKV_LAUNCH(true);
     DQ_LAUNCH(true);
@@ -577,4 +656,6 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   }
 #undef DKV_LAUNCH
 #undef DQ_LAUNCH
+#undef DKV_ARGS
+#undef DQ_ARGS
 }

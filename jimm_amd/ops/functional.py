@@ -190,11 +190,12 @@ class _AttentionQKVFn(torch.autograd.Function):
 def attention_qkv(qkv: torch.Tensor, *, causal: bool = False, scale: float | None = None) -> torch.Tensor:
     """qkv (B, L, 3, H, D) -> (B, H, L, D). GPU fast path for self-attention.
 
-    The flash kernels are bf16/head_dim-64 by design; other dtypes/head dims
+    The flash kernels are bf16 with head_dim in {64, 72, 80, 96, 128}
+    (DP-padded MFMA templates); other dtypes/head dims
     run the composite torch math (on GPU or CPU alike)."""
     if scale is None:
         scale = 1.0 / math.sqrt(qkv.shape[-1])
-    if _backend.use_hip(qkv) and qkv.dtype == torch.bfloat16 and qkv.shape[-1] == 64:
+    if _backend.use_hip(qkv) and qkv.dtype == torch.bfloat16 and qkv.shape[-1] in (64, 72, 80, 96, 128):
         return _AttentionQKVFn.apply(qkv, causal, scale)
     q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
     return attention(q, k, v, causal=causal, scale=scale)
@@ -210,7 +211,7 @@ def attention(
 ) -> torch.Tensor:
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _backend.use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] == 64:
+    if _backend.use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 72, 80, 96, 128):
         return _AttentionFn.apply(q, k, v, causal, scale)
     # composite reference (fp32 math; float64 preserved for gradcheck)
     ref_dtype = torch.float64 if q.dtype == torch.float64 else torch.float32
