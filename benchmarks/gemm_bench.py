@@ -61,9 +61,9 @@ def main_dw():
         dz = torch.randn(M, N, device=dev).bfloat16()
         x = torch.randn(M, K, device=dev).bfloat16() / K**0.5
         tf = 2 * M * N * K / 1e12
-        t_mine = bench(lambda: ext.gemm_tn_splitk(dz, x))
+        t_mine = bench(lambda: ext.gemm_tn_8p(dz, x))
         t_blas = bench(lambda: torch.matmul(dz.t(), x))
-        dw = ext.gemm_tn_splitk(dz, x)
+        dw = ext.gemm_tn_8p(dz, x)
         ref = torch.matmul(dz.t().float(), x.float())
         err = (dw.float() - ref).abs().max().item() / ref.abs().max().item()
         print(f"dW M={M:6d} N={N:5d} K={K:5d} mine {tf/t_mine:7.1f} TF/s ({t_mine*1e3:6.2f} ms)  "

@@ -171,11 +171,17 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const int q0 = ti * QBLK + wave * 16;
       if (ti >= ntq || q0 >= Lq) continue;
       if (CAUSAL && kv0 >= (ti + 1) * QBLK) continue;  // fully masked strip
-      // ---- S^T = K . Q^T : 4 key tiles x 2 d-steps ------------------------
+      // keys beyond kv_hi are padding (>= Lk) or above this wave-strip's
+      // causal diagonal (> q0+15): their sub-tiles are skipped outright —
+      // at L=77 causal ~3/4 of the key sub-tiles are pure waste otherwise
+      //  — and masked to -inf / P=0 below.
+      const int kv_hi = CAUSAL ? min(Lk, q0 + 16) : Lk;
+      // ---- S^T = K . Q^T : 4 key tiles x NS d-steps -----------------------
       __builtin_amdgcn_s_setprio(1);
       f32x4_t sc[4] = {};
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
+        if (kv0 + 16 * t >= kv_hi) continue;
 #pragma unroll
         for (int s = 0; s < NS; ++s) {
           const bf16x8_t ka =
@@ -242,6 +248,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
+        if (kv0 + 32 * s >= kv_hi) continue;  // P = 0 for the whole step
         const bf16x8_t pa =
             *reinterpret_cast<const bf16x8_t*>(my_p + lo * LDS_PITCH + 32 * s + hi * 8);
 #pragma unroll

@@ -238,9 +238,22 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       if (CAUSAL && q0 + BLK - 1 < kv0) continue;  // whole tile above diagonal
 
       // ---- S^T = K . Q^T ; P^T = exp(scale*S^T - lse[q]) -----------------
+      // this wave's keys start at key_min; q sub-tiles fully above the
+      // causal diagonal (or beyond Lq) contribute P = dS = 0 — write zeros
+      // and skip their MFMAs (at L=77 causal most sub-tiles are waste)
+      const int key_min = kv0 + 16 * wave;
       bf16x4 ds_stash[4];
 #pragma unroll
       for (int qt = 0; qt < 4; ++qt) {
+        const bool skip_qt =
+            (CAUSAL && q0 + 16 * qt + 15 < key_min) || (q0 + 16 * qt >= Lq);
+        if (skip_qt) {
+          const bf16x4 z = {};
+#pragma unroll
+          for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = 0;
+          ds_stash[qt] = z;
+          continue;
+        }
         __builtin_amdgcn_s_setprio(1);
         f32x4_t sc = {};
         f32x4_t dpc = {};
@@ -279,6 +292,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
+        if ((CAUSAL && q0 + 32 * s + 31 < key_min) || q0 + 32 * s >= Lq) continue;
         const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
         for (int dt = 0; dt < NT; ++dt) {
@@ -298,6 +312,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
+        if ((CAUSAL && q0 + 32 * s + 31 < key_min) || q0 + 32 * s >= Lq) continue;
         const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
         for (int dt = 0; dt < NT; ++dt) {
@@ -481,8 +496,18 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       if (CAUSAL && kv0 >= qbase[sidx] + BLK) continue;  // tile above diagonal
 
       // ---- S, P, dP, dS per 16-key tile ----------------------------------
+      // key sub-tiles above this strip's causal diagonal or beyond Lk give
+      // dS = 0: write zeros, skip the MFMAs
+      const int q_max = q0s + 15;
 #pragma unroll
       for (int kt = 0; kt < 4; ++kt) {
+        const bool skip_kt =
+            (CAUSAL && kv0 + 16 * kt > q_max) || (kv0 + 16 * kt >= Lk);
+        if (skip_kt) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) my_ds[(hi * 4 + r) * PITCH + 16 * kt + lo] = 0;
+          continue;
+        }
         __builtin_amdgcn_s_setprio(1);
         f32x4_t sc = {};
         f32x4_t dpc = {};
@@ -517,6 +542,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
+        if ((CAUSAL && kv0 + 32 * s > q_max) || kv0 + 32 * s >= Lk) continue;
         const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_ds + lo * PITCH + 32 * s + hi * 8);
 #pragma unroll
         for (int dt = 0; dt < NT; ++dt) {

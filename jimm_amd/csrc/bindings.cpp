@@ -42,13 +42,11 @@ std::vector<torch::Tensor> xent_rows_fwd(torch::Tensor logits, torch::Tensor lab
 torch::Tensor xent_rows_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
                             double gscale);
 std::vector<torch::Tensor> sigmoid_loss_ew(torch::Tensor logits, int64_t diag0);
-bool gemm_dw_supported(int64_t M, int64_t N, int64_t K);
 bool gemm8p_supported(int64_t M, int64_t N, int64_t K);
 bool gemm_tn8p_supported(int64_t M, int64_t N, int64_t K);
 torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x);
 torch::Tensor gemm_nt_8p_gradact(torch::Tensor dy, torch::Tensor wt, torch::Tensor z, std::string act);
 torch::Tensor tr16_probe(torch::Tensor src, int64_t mode);
-torch::Tensor gemm_tn_splitk(torch::Tensor dz, torch::Tensor x);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias, std::string act,
                                       c10::optional<torch::Tensor> residual, bool save_z);
@@ -76,7 +74,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("mfma_probe32", &mfma_probe32, "MFMA 32x32x16 layout probe");
   m.def("gemm_supported", &gemm_supported, "MFMA GEMM shape support check");
-  m.def("gemm_dw_supported", &gemm_dw_supported, "split-K dW GEMM shape check");
   m.def("gemm8p_supported", &gemm8p_supported, "8-phase 256-tile GEMM shape check");
   m.def("gemm_tn8p_supported", &gemm_tn8p_supported, "TN dW GEMM shape check");
   m.def("gemm_tn_8p", &gemm_tn_8p, "TN weight-grad GEMM, split-M, tr_b16 (K15)");
@@ -87,6 +84,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("xent_rows_fwd", &xent_rows_fwd, "fused softmax-CE forward (K13)");
   m.def("xent_rows_bwd", &xent_rows_bwd, "fused softmax-CE backward (K13)");
   m.def("sigmoid_loss_ew", &sigmoid_loss_ew, "SigLIP sigmoid loss + dLogits (K13)");
-  m.def("gemm_tn_splitk", &gemm_tn_splitk, "split-K TN GEMM for weight grads (K15)");
   m.def("linear_fwd", &linear_fwd, "MFMA GEMM + fused epilogue (K4/K6/K7/K8)");
 }

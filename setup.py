@@ -33,7 +33,6 @@ setup(
                 "jimm_amd/csrc/gemm256.hip",
                 "jimm_amd/csrc/gemm8p.hip",
                 "jimm_amd/csrc/gemm_tn8p.hip",
-                "jimm_amd/csrc/gemm_dw.hip",
                 "jimm_amd/csrc/losses.hip",
                 "jimm_amd/csrc/probe.hip",
             ],

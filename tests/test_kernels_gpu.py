@@ -402,17 +402,17 @@ def test_gemm256(M, N, K, act):
     [
         (50432, 2304, 768),   # ViT-B qkv dW
         (1024, 768, 3072),    # fc2 dW small-M
-        (64, 128, 128),       # single tile, S=M/64 clamp
+        (64, 256, 256),       # single tile, splitm clamp
         (19712, 1536, 512),   # CLIP text
     ],
 )
-def test_gemm_dw_splitk(M, N, K):
-    """Split-K TN dW kernel vs fp32 reference."""
+def test_gemm_dw_tn8p(M, N, K):
+    """Split-M TN dW kernel (tr_b16 + glds staging) vs fp32 reference."""
     torch.manual_seed(0)
     dz = torch.randn(M, N, device=dev()).bfloat16()
     x = (torch.randn(M, K, device=dev()) / math.sqrt(K)).bfloat16()
-    assert EXT.gemm_dw_supported(M, N, K)
-    dw = EXT.gemm_tn_splitk(dz, x)
+    assert EXT.gemm_tn8p_supported(M, N, K)
+    dw = EXT.gemm_tn_8p(dz, x)
     ref = dz.float().t() @ x.float()
     assert rel_err(dw, ref) < 2e-2, rel_err(dw, ref)
 
