@@ -81,7 +81,8 @@ def main() -> None:
     p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
     p.add_argument("--ckpt", action="store_true", help="gradient checkpointing (huge batches)")
     p.add_argument("--graph", choices=["auto", "1", "0"], default="auto",
-                   help="hipGraph-capture the train step (auto: single-GPU only)")
+                   help="hipGraph-capture the train step (any world size; capture "
+                        "failure falls back to eager COORDINATED across ranks)")
     args = p.parse_args()
 
     from jimm_amd.ops._backend import maybe_enable_tunableop

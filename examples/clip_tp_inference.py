@@ -3,9 +3,11 @@ examples/clip_inference.py, which runs with mesh (1, n_devices) =
 ("batch","model"), i.e. PURE tensor sharding of the weights across all
 devices (/root/reference/examples/clip_inference.py:17-18).
 
-Here the sharding is explicit (parallel/tp.py): every encoder block keeps
-num_heads/N heads and mlp_dim/N of the MLP per rank, with all-reduces after
-the row-parallel projections.
+Here the sharding is explicit (parallel/tp.py shard_clip) and covers the
+WHOLE model: every encoder block keeps num_heads/N heads and mlp_dim/N of
+the MLP per rank, the token embedding is vocab-parallel, and the
+visual/text projections are row-parallel — with all-reduces after each
+row-parallel product.
 
     torchrun --standalone --nproc-per-node 8 examples/clip_tp_inference.py
     (also runs on CPU with gloo for any world size that divides the heads)
@@ -19,7 +21,7 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 
 import jimm_amd
-from jimm_amd.parallel.tp import shard_encoder
+from jimm_amd.parallel.tp import shard_clip
 from jimm_amd.train.trainer import init_distributed
 
 PROMPTS = [
@@ -47,8 +49,10 @@ def main():
         model = jimm_amd.CLIP()
     model = model.to(device, dtype).eval()
     if world > 1:
-        shard_encoder(model.vision_model.encoder, None)
-        shard_encoder(model.text_model.encoder, None)
+        # FULL-model sharding: encoders + vocab-parallel token embedding +
+        # row-parallel projections (the reference's (1, n) mesh shards every
+        # parameter — clip.py:89,112-131)
+        shard_clip(model, None)
 
     img = torch.randn(1, 3, 224, 224, device=device, dtype=dtype)
     ids = simple_tokenize(PROMPTS).to(device)

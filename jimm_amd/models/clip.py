@@ -78,10 +78,14 @@ class CLIP(nn.Module):
         self.text_model.encoder.gradient_checkpointing = True
 
     def encode_image(self, images: torch.Tensor) -> torch.Tensor:
-        return ops.linear(self.vision_model(images), self.visual_projection.weight)
+        from jimm_amd.parallel.tp import row_parallel_linear
+
+        return row_parallel_linear(self.vision_model(images), self.visual_projection)
 
     def encode_text(self, input_ids: torch.Tensor) -> torch.Tensor:
-        return ops.linear(self.text_model(input_ids), self.text_projection.weight)
+        from jimm_amd.parallel.tp import row_parallel_linear
+
+        return row_parallel_linear(self.text_model(input_ids), self.text_projection)
 
     def forward(self, images: torch.Tensor, input_ids: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         """Returns (logits_per_image, logits_per_text) — clip.py:169-188."""

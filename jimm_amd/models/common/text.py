@@ -60,7 +60,25 @@ class TextTransformer(nn.Module):
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         """input_ids (B, L) -> pooled (B, H)."""
-        x = ops.embed_pos(input_ids, self.token_embedding.weight, self.pos_embedding)  # K10 fused
+        if getattr(self, "_tp_vocab", None) is not None:
+            # vocab-parallel embedding (parallel/tp.py): this rank holds
+            # token rows [start, end); other ranks' tokens contribute a zero
+            # vector; the all-reduce reassembles the full lookup. The
+            # position embedding is added AFTER the reduce (it is replicated
+            # and must be counted once).
+            import torch.nn.functional as F
+
+            from jimm_amd.parallel.tp import reduce_from_tp
+
+            start, end = self._tp_vocab
+            local = (input_ids >= start) & (input_ids < end)
+            ids_l = (input_ids - start).clamp(0, end - start - 1)
+            x = F.embedding(ids_l, self.token_embedding.weight)
+            x = x * local.unsqueeze(-1).to(x.dtype)
+            x = reduce_from_tp(x, self._tp_group)
+            x = x + self.pos_embedding[:, : input_ids.shape[1]].to(x.dtype)
+        else:
+            x = ops.embed_pos(input_ids, self.token_embedding.weight, self.pos_embedding)  # K10 fused
         x = self.encoder(x)
         x = ops.layer_norm(x, self.ln_final.weight, self.ln_final.bias, self.eps)
         if self.pooling == "EOT":

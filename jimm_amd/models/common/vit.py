@@ -54,11 +54,33 @@ class MAPHead(nn.Module):
         wq, wk, wv = self.in_proj_weight.chunk(3, dim=0)
         bq, bk, bv = self.in_proj_bias.chunk(3, dim=0)
         probe = self.probe.expand(B, -1, -1).to(hidden.dtype)
+        if getattr(self, "_tp_group", None) is not None:
+            # column-parallel q/k/v consume the full (replicated) input; the
+            # identity-fwd/all-reduce-bwd makes dhidden/dprobe exact under TP
+            from jimm_amd.parallel.tp import copy_to_tp
+
+            hidden = copy_to_tp(hidden, self._tp_group)
+            probe = copy_to_tp(probe, self._tp_group)
         q = ops.linear(probe, wq, bq).view(B, 1, nh, d).transpose(1, 2)  # (B, nh, 1, d)
         k = ops.linear(hidden, wk, bk).view(B, L, nh, d).transpose(1, 2)
         v = ops.linear(hidden, wv, bv).view(B, L, nh, d).transpose(1, 2)
         x = ops.attention(q, k, v)  # (B, nh, 1, d) — degenerate flash, Lq=1 (K9)
-        x = x.transpose(1, 2).reshape(B, 1, H)
+        x = x.transpose(1, 2).reshape(B, 1, nh * d)
+        if getattr(self, "_tp_group", None) is not None:
+            # TP path (parallel/tp.py shard_map_head): local heads, partial
+            # out-proj / fc2 all-reduced, biases and residual counted once
+            from jimm_amd.parallel.tp import copy_to_tp, reduce_from_tp
+
+            g = self._tp_group
+            part = ops.linear(x, self.out_proj.weight)
+            x = reduce_from_tp(part, g) + self.out_proj.bias
+            residual = x
+            x = ops.layer_norm(x, self.layernorm.weight, self.layernorm.bias, self.eps)
+            x = copy_to_tp(x, g)
+            x = ops.linear(x, self.fc1.weight, self.fc1.bias, act=self.act)
+            part = ops.linear(x, self.fc2.weight)
+            x = reduce_from_tp(part, g) + self.fc2.bias + residual
+            return x[:, 0]
         x = ops.linear(x, self.out_proj.weight, self.out_proj.bias)
         residual = x
         x = ops.layer_norm(x, self.layernorm.weight, self.layernorm.bias, self.eps)

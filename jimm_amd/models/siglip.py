@@ -79,7 +79,9 @@ class SigLIP(nn.Module):
         return self.vision_model(images)  # no visual projection (siglip.py:123-133)
 
     def encode_text(self, input_ids: torch.Tensor) -> torch.Tensor:
-        return ops.linear(self.text_model(input_ids), self.text_projection.weight, self.text_projection.bias)
+        from jimm_amd.parallel.tp import row_parallel_linear
+
+        return row_parallel_linear(self.text_model(input_ids), self.text_projection)
 
     def forward(self, images: torch.Tensor, input_ids: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         """Returns (logits_per_image, logits_per_text) — siglip.py:155-174."""

@@ -38,6 +38,13 @@ class DataParallelGrads:
         process_group=None,
         broadcast_params: bool = True,
     ) -> None:
+        import os
+
+        if "JIMM_AMD_BUCKET_MB" in os.environ:
+            # xGMI tuning knob: ring all-reduce is per-link bound (7 x
+            # ~153 GB/s p2p links), so bucket size trades pipeline depth
+            # against per-launch overhead; sweep on the first 8-GPU node
+            bucket_bytes = int(float(os.environ["JIMM_AMD_BUCKET_MB"]) * 1024 * 1024)
         self.model = model
         self.group = process_group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
@@ -94,6 +101,7 @@ class DataParallelGrads:
         for b in self.buckets:
             b["pending"] = len(b["params"])
             b["work"] = None
+        self.last_step_bytes = 0
 
     def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
         if not self.enabled:
@@ -101,6 +109,7 @@ class DataParallelGrads:
         b = self.buckets[self._param_bucket[p]]
         b["pending"] -= 1
         if b["pending"] == 0:
+            self.last_step_bytes += b["flat"].numel() * b["flat"].element_size()
             b["work"] = dist.all_reduce(b["flat"], op=dist.ReduceOp.SUM, group=self.group, async_op=True)
 
     def zero_grad(self) -> None:
@@ -121,7 +130,9 @@ class DataParallelGrads:
             else:
                 continue  # bucket untouched this step (e.g. frozen tower)
             b["flat"].mul_(inv)
+        bytes_reduced = self.last_step_bytes
         self._reset_counters()
+        self.last_step_bytes = bytes_reduced  # keep for comm_stats logging
 
     def grad_bytes(self) -> int:
         return sum(b["flat"].numel() * b["flat"].element_size() for b in self.buckets)

@@ -15,6 +15,8 @@ from __future__ import annotations
 import torch
 import torch.distributed as dist
 
+_last_gather_bytes = 0  # per-call payload, surfaced by Trainer.comm_stats()
+
 
 class _AllGatherFn(torch.autograd.Function):
     @staticmethod
@@ -25,6 +27,8 @@ class _AllGatherFn(torch.autograd.Function):
         ctx.rank = dist.get_rank(group)
         xc = x.contiguous()
         out = torch.empty(world * x.shape[0], *x.shape[1:], dtype=x.dtype, device=x.device)
+        global _last_gather_bytes
+        _last_gather_bytes = out.numel() * out.element_size()
         dist.all_gather_into_tensor(out, xc, group=group)
         return out
 

@@ -93,6 +93,12 @@ class Adam(torch.optim.Optimizer):
                 self._ref_step(ps, gs, ms, vs, masters, lr, b1, b2, eps, wd, step_t)
         return loss
 
+    def set_device_lr(self, lr: float) -> None:
+        """Refresh the device-side lr used by graph-replayed adam_apply."""
+        for gi, prep in self._prepared.items():
+            prep[2].fill_(lr)
+            self._last_lr[gi] = lr
+
     def load_state_dict(self, state_dict) -> None:
         super().load_state_dict(state_dict)
         # The base class casts loaded state to the param dtype; for bf16

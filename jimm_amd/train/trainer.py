@@ -79,8 +79,7 @@ class Trainer:
         # under graph replay the captured Adam reads lr from device memory:
         # refresh it OUTSIDE the graph
         if self._graph is not None and getattr(self.opt, "_prepared", None):
-            self.opt._prepared[2].fill_(lr)
-            self.opt._last_lr = lr
+            self.opt.set_device_lr(lr)
 
     def train_step(self, batch) -> dict[str, torch.Tensor]:
         if self.cfg.lr_schedule is not None:
@@ -147,6 +146,22 @@ class Trainer:
         self.step_idx += 1
         out["loss"] = loss.detach()
         return out
+
+    def comm_stats(self) -> dict:
+        """Per-step collective traffic (per-collective bytes).
+
+        allreduce_bytes counts the DP gradient payload actually reduced in
+        the latest step; gather_bytes the contrastive-loss embedding
+        all-gather (clip/siglip tasks)."""
+        stats = {
+            "allreduce_bytes": self.ddp.last_step_bytes,
+            "allreduce_buckets": len(self.ddp.buckets),
+            "grad_bytes_total": self.ddp.grad_bytes(),
+        }
+        from jimm_amd.parallel import gather as _g
+
+        stats["gather_bytes"] = getattr(_g, "_last_gather_bytes", 0)
+        return stats
 
     # -- checkpoint / resume (SURVEY §5: the reference is load-only; we add
     # full trainer state so long runs are resumable) ------------------------
