@@ -55,8 +55,16 @@ class EncoderBlockFn(torch.autograd.Function):
         x = x.contiguous()
         x2 = x.view(-1, H)
         hip = _hip_gemms(B * L, H)
+        from jimm_amd.ops.hip_linear import _FP8_STATE, _fp8_ok, _gemm_nt_fp8
+
+        fp8 = _FP8_STATE["enabled"] and x.dtype == torch.bfloat16
         h1, mean1, rstd1 = ext.layernorm_fwd(x, ln1w, ln1b, eps)
-        if hip:
+        if fp8 and _fp8_ok(h1.view(-1, H), wqkv):
+            # BASELINE config 5 path: forward GEMMs on the fp8 MFMA pipe
+            # (e4m3 per-tensor dynamic scales, device-side — graph-safe);
+            # LN/softmax/losses and the whole backward stay bf16/fp32
+            qkv2 = ext.bias_act_fwd(_gemm_nt_fp8(h1.view(-1, H), wqkv), bqkv, "", None)
+        elif hip:
             # in-house MFMA GEMMs with the bias / act / residual epilogues
             # fused into the GEMM kernel itself (csrc/gemm8p.hip)
             qkv2, _ = ext.linear_fwd(h1.view(-1, H), wqkv, bqkv, "", None, False)
@@ -68,7 +76,14 @@ class EncoderBlockFn(torch.autograd.Function):
         v = qkv[:, :, 2].transpose(1, 2)
         o, lse = ext.attn_fwd(q, k, v, causal, scale)            # (B,nh,L,d), (B,L,nh,d) storage
         o2 = o.transpose(1, 2).reshape(-1, H)                    # free view
-        if hip:
+        if fp8 and _fp8_ok(o2, wproj):
+            a = ext.bias_act_fwd(_gemm_nt_fp8(o2, wproj), bproj, "", x2)
+            a3 = a.view(B, L, H)
+            h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)
+            z1 = _gemm_nt_fp8(h2.view(-1, H), w1)
+            f = ext.bias_act_fwd(z1, b1, act, None)  # z1 -> pre-act in place
+            y = ext.bias_act_fwd(_gemm_nt_fp8(f, w2), b2, "", a)
+        elif hip:
             a, _ = ext.linear_fwd(o2, wproj, bproj, "", x2, False)
             a3 = a.view(B, L, H)
             h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)

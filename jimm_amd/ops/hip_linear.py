@@ -1,20 +1,18 @@
 """GPU linear / patch-embed path: GEMM + fused bias/activation/residual epilogue.
 
 GEMM engine selection (env ``JIMM_AMD_GEMM``):
-  * ``blas`` (default) — hipBLASLt/rocBLAS (plain library GEMMs, with the
-               committed MI355X TunableOp table; bias-only linears use the
-               hipBLASLt bias epilogue via ``torch.addmm``), followed by the
-               fused HIP bias+act(+residual) elementwise kernel where needed.
-               Measured faster than the in-house GEMMs on the model-zoo
-               shapes (benchmarks/gemm_bench.py).
-  * ``hip``  — in-house MFMA bf16 GEMM kernels (csrc/gemm.hip, 128x128 tile;
-               csrc/gemm256.hip, 256x256 tile) with the epilogue fused into
-               the GEMM itself.
-  * fp8      — ``jimm_amd.ops.set_fp8(True)``: e4m3 scaled_mm forward path.
-
-Backward GEMMs: dX runs through rocBLAS; dW defaults to rocBLAS (TunableOp
-split-K picks) with the in-house split-K TN kernel opt-in (JIMM_AMD_DW=hip);
-dz comes from the fused HIP ``act_bwd`` kernel and db from ``colsum``.
+  * ``hip`` (default) — in-house 8-phase 256-tile MFMA bf16 GEMM
+               (csrc/gemm8p.hip) with bias/act/residual/act-bwd epilogues
+               fused into the GEMM itself; dW on the split-M TN kernel
+               (csrc/gemm_tn8p.hip, fp32-exact); dX on the NT kernel over a
+               pre-transposed weight copy.
+  * ``blas`` — hipBLASLt/rocBLAS (with the committed MI355X TunableOp
+               table) + the separate fused bias+act(+residual) elementwise
+               kernel. Kept for A/B and for shapes the in-house kernels
+               do not cover (N % 256 != 0 or K % 64 != 0).
+  * fp8      — ``jimm_amd.ops.set_fp8(True)``: e4m3 scaled_mm forward and
+               plain-dX GEMMs (per-tensor dynamic device-side scales,
+               graph-capturable); gradact/dW/LN/losses stay bf16/fp32.
 """
 
 from __future__ import annotations
@@ -70,6 +68,18 @@ def _gemm_nt_fp8(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     )
 
 
+def _gemm_nn_fp8(g: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """dX = g @ w in fp8 (e4m3 per-tensor dynamic scales, bf16 out).
+
+    b operand must be column-major for _scaled_mm: quantize w.t() row-major
+    and pass its transpose view."""
+    g8, sg = _quant_e4m3(g)
+    wt8, sw = _quant_e4m3(w.t().contiguous())
+    return torch._scaled_mm(
+        g8, wt8.t(), scale_a=sg.view(1, 1), scale_b=sw.view(1, 1), out_dtype=torch.bfloat16
+    )
+
+
 def _gemm_mode() -> str:
     # Default "hip" (round 2): the 8-phase 256-tile MFMA GEMM (csrc/gemm8p)
     # with epilogues (bias/act/residual/act-bwd) fused into the GEMM itself,
@@ -96,7 +106,11 @@ def _dw_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype) -> torch.Tensor
 
 
 def _dx_gemm(ext, dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy."""
+    """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy;
+    fp8 scaled_mm when the fp8 mode is on (forward-quality grads are not
+    needed for dX of linear layers at config-5 tolerances)."""
+    if _FP8_STATE["enabled"] and _fp8_ok(dz, w) and w.shape[1] % 16 == 0:
+        return _gemm_nn_fp8(dz, w)
     if (
         _gemm_mode() == "hip"
         and dz.dtype == torch.bfloat16

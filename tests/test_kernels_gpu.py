@@ -652,3 +652,40 @@ def test_mfma_probe32():
     C = EXT.mfma_probe32(A, B)
     ref = A.float() @ B.float()
     assert (C - ref).abs().max().item() < 1e-2, (C - ref).abs().max().item()
+
+
+# ---------------------------------------------------------------------------
+def test_fp32_model_gpu_vs_cpu_oracle():
+    """an fp32 forward on GPU (fp32 HIP layernorm + rocBLAS
+    GEMMs + composite attention) must match the CPU oracle to <= 1e-4 —
+    a tight check that separates kernel/plumbing bugs from bf16 noise,
+    which the 0.15-0.25 bf16 tolerances would mask."""
+    torch.manual_seed(0)
+    import jimm_amd as J
+
+    model = J.VisionTransformer(num_classes=10, img_size=32, patch_size=16,
+                                num_layers=2, num_heads=4, mlp_dim=128, hidden_size=64)
+    model = model.float().eval()
+    x = torch.randn(3, 3, 32, 32)
+    with torch.no_grad():
+        ref = model(x)
+        got = model.to(dev())(x.to(dev())).cpu()
+    err = (got - ref).abs().max().item()
+    assert err < 1e-4, err
+
+
+def test_fp32_clip_gpu_vs_cpu_oracle():
+    torch.manual_seed(1)
+    import jimm_amd as J
+
+    model = J.CLIP(embed_dim=16, image_resolution=32, vision_layers=1, vision_width=64,
+                   vision_patch_size=16, context_length=8, vocab_size=100,
+                   transformer_width=64, transformer_heads=2, transformer_layers=1).float().eval()
+    imgs = torch.randn(2, 3, 32, 32)
+    ids = torch.randint(0, 99, (2, 8))
+    ids[:, -1] = 99
+    with torch.no_grad():
+        ref, _ = model(imgs, ids)
+        got, _ = model.to(dev())(imgs.to(dev()), ids.to(dev()))
+    err = (got.cpu() - ref).abs().max().item()
+    assert err < 1e-4, err
