@@ -106,11 +106,11 @@ def _dw_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype) -> torch.Tensor
 
 
 def _dx_gemm(ext, dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy;
-    fp8 scaled_mm when the fp8 mode is on (forward-quality grads are not
-    needed for dX of linear layers at config-5 tolerances)."""
-    if _FP8_STATE["enabled"] and _fp8_ok(dz, w) and w.shape[1] % 16 == 0:
-        return _gemm_nn_fp8(dz, w)
+    """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy.
+
+    (fp8 dX via _gemm_nn_fp8 was measured NET-NEGATIVE at config 5: the
+    per-call w.t().contiguous() + two quantization passes cost more than
+    the fp8 GEMM saves — backward stays bf16.)"""
     if (
         _gemm_mode() == "hip"
         and dz.dtype == torch.bfloat16
