@@ -50,6 +50,60 @@ __device__ __forceinline__ bf16x8_t ld8g(const bf16* p, int off, int Dr) {
   return bf16x8_t{};
 }
 
+// V block image (the gemm_tn8p.hip recipe, see attention_bwd_fused.hip):
+// [64 key][DP d] stored as [key-half][d16][8 key-quads evens-first][4][16],
+// written with vector ds_write_b128 from the row-major staging registers
+// and read as PV B-fragments with batched ds_read_b64_tr_b16 at per-lane
+// address base + lane*8 B.
+template <int ND>
+__device__ __forceinline__ int boff_f(int q, int d) {
+  const int qp = (q >> 2) & 7;
+  const int qpos = (qp & 1) * 4 + (qp >> 1);
+  return (q >> 5) * (32 * ND) + (d >> 4) * 512 + qpos * 64 + (q & 3) * 16 + (d & 15);
+}
+
+typedef short bf16x4_trf __attribute__((ext_vector_type(4)));
+typedef const __attribute__((address_space(3))) char* lds_cpf;
+
+__device__ __forceinline__ void trf_x4(lds_cpf base, bf16x8_t (&out)[4]) {
+  bf16x4_trf a0l, a0h, a1l, a1h, a2l, a2h, a3l, a3h;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8 offset:0\n\t"
+      "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+      "ds_read_b64_tr_b16 %2, %8 offset:1024\n\t"
+      "ds_read_b64_tr_b16 %3, %8 offset:1536\n\t"
+      "ds_read_b64_tr_b16 %4, %8 offset:2048\n\t"
+      "ds_read_b64_tr_b16 %5, %8 offset:2560\n\t"
+      "ds_read_b64_tr_b16 %6, %8 offset:3072\n\t"
+      "ds_read_b64_tr_b16 %7, %8 offset:3584\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a0l), "=&v"(a0h), "=&v"(a1l), "=&v"(a1h), "=&v"(a2l), "=&v"(a2h),
+        "=&v"(a3l), "=&v"(a3h)
+      : "v"(base)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  out[0] = __builtin_shufflevector(a0l, a0h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
+__device__ __forceinline__ void trf_x2(lds_cpf base, bf16x8_t (&out)[2]) {
+  bf16x4_trf a0l, a0h, a1l, a1h;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+      "ds_read_b64_tr_b16 %1, %4 offset:512\n\t"
+      "ds_read_b64_tr_b16 %2, %4 offset:1024\n\t"
+      "ds_read_b64_tr_b16 %3, %4 offset:1536\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a0l), "=&v"(a0h), "=&v"(a1l), "=&v"(a1h)
+      : "v"(base)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  out[0] = __builtin_shufflevector(a0l, a0h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
 template <bool CAUSAL, int NSTRIP, int DP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
@@ -70,8 +124,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   constexpr int NT = DP / 16;        // d-tiles of the output
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);                       // KVBLK rows
-  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // DP rows
-  short* p_lds = vt_lds + DP * LDS_PITCH;                              // 4*16 rows
+  short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // V block image [KVBLK][DP]
+  short* p_lds = vt_lds + KVBLK * DP;                                  // 4*16 rows
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -146,13 +200,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       if (st_valid) {
         *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0) = kreg[cg][0];
         *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0 + 8) = kreg[cg][1];
-#pragma unroll
-        for (int i = 0; i < 8; ++i) vt_lds[(c0 + i) * LDS_PITCH + st_row] = vreg[cg][0][i];
-#pragma unroll
-        for (int i = 0; i < 8; ++i) vt_lds[(c0 + 8 + i) * LDS_PITCH + st_row] = vreg[cg][1][i];
+        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0)) = vreg[cg][0];
+        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0 + 8)) = vreg[cg][1];
       } else {
-        for (int i = 0; i < 16; ++i) k_lds[st_row * LDS_PITCH + c0 + i] = 0;
-        for (int i = 0; i < 16; ++i) vt_lds[(c0 + i) * LDS_PITCH + st_row] = 0;
+        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0) = bf16x8_t{};
+        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0 + 8) = bf16x8_t{};
+        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0)) = bf16x8_t{};
+        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0 + 8)) = bf16x8_t{};
       }
     }
   };
@@ -244,21 +298,22 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) acc_o[st][dt][r] *= alpha_r[r];
 
-      // ---- O += P . V : A = P (from LDS), B = V^T reads ------------------
-      __builtin_amdgcn_s_setprio(1);
+      // ---- O += P . V : A = P (from LDS), B = V^T via tr reads -----------
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
         if (kv0 + 32 * s >= kv_hi) continue;  // P = 0 for the whole step
+        bf16x8_t vb[NT];
+        const lds_cpf bbase = (lds_cpf)(const void*)(vt_lds + s * 32 * DP) + lane * 8;
+        trf_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&vb[0]));
+        if constexpr (NT == 6) trf_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&vb[4]));
+        if constexpr (NT == 8) trf_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&vb[4]));
         const bf16x8_t pa =
             *reinterpret_cast<const bf16x8_t*>(my_p + lo * LDS_PITCH + 32 * s + hi * 8);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int dt = 0; dt < NT; ++dt) {
-          const bf16x8_t vb =
-              *reinterpret_cast<const bf16x8_t*>(vt_lds + (16 * dt + lo) * LDS_PITCH + 32 * s + hi * 8);
-          acc_o[st][dt] = MFMA16(pa, vb, acc_o[st][dt]);
-        }
+        for (int dt = 0; dt < NT; ++dt) acc_o[st][dt] = MFMA16(pa, vb[dt], acc_o[st][dt]);
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     // T14: overwrite the K/V tiles with tile it+1 after everyone is done
     __syncthreads();
@@ -325,7 +380,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
     if (const char* e = getenv("JIMM_AMD_ATTN_NSTRIP")) nstrip = (std::string(e) == "2") ? 2 : 4;
   }
   const dim3 grid((ntq + nstrip - 1) / nstrip, (unsigned)((int64_t)B * H));
-  const size_t shmem = ((KVBLK + DP + 4 * 16) * (DP + 8)) * sizeof(short);
+  const size_t shmem = (KVBLK * DP + (KVBLK + 4 * 16) * (DP + 8)) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
 #define ATTN_ARGS                                                                          \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \

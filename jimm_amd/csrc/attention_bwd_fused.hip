@@ -50,6 +50,64 @@ __device__ __forceinline__ bf16x8_t ld8gb(const bf16* p, int off, int Dr) {
   return bf16x8_t{};
 }
 
+// Block-format image for transposed fragment reads (the gemm_tn8p.hip
+// recipe): a [64 q][ND d] tile stored as [q-half (q>>5)][d16 (d>>4)]
+// [8 q-quads, evens first][4][16] — written from row-major registers with
+// ds_write_b128 (two vector writes per 16-d chunk instead of 16 scalar
+// transpose stores, which were 8-way bank-conflicted), and read as MFMA
+// B-fragments with batched ds_read_b64_tr_b16 at per-lane address
+// base + lane*8 B (mapping verified by ext.tr16_probe).
+template <int ND>
+__device__ __forceinline__ int boff(int q, int d) {
+  const int qp = (q >> 2) & 7;
+  const int qpos = (qp & 1) * 4 + (qp >> 1);
+  return (q >> 5) * (32 * ND) + (d >> 4) * 512 + qpos * 64 + (q & 3) * 16 + (d & 15);
+}
+
+typedef short bf16x4_tr __attribute__((ext_vector_type(4)));
+typedef const __attribute__((address_space(3))) char* lds_cp;
+
+// 4 B-fragments (dt, dt+1, dt+2, dt+3) of one 32-contraction step from a
+// block image: 8 tr reads + lgkmcnt(0) in one asm (guide §5.7 form i).
+__device__ __forceinline__ void tr_frag_x4(lds_cp base, bf16x8_t (&out)[4]) {
+  bf16x4_tr a0l, a0h, a1l, a1h, a2l, a2h, a3l, a3h;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8 offset:0\n\t"
+      "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+      "ds_read_b64_tr_b16 %2, %8 offset:1024\n\t"
+      "ds_read_b64_tr_b16 %3, %8 offset:1536\n\t"
+      "ds_read_b64_tr_b16 %4, %8 offset:2048\n\t"
+      "ds_read_b64_tr_b16 %5, %8 offset:2560\n\t"
+      "ds_read_b64_tr_b16 %6, %8 offset:3072\n\t"
+      "ds_read_b64_tr_b16 %7, %8 offset:3584\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a0l), "=&v"(a0h), "=&v"(a1l), "=&v"(a1h), "=&v"(a2l), "=&v"(a2h),
+        "=&v"(a3l), "=&v"(a3h)
+      : "v"(base)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  out[0] = __builtin_shufflevector(a0l, a0h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
+__device__ __forceinline__ void tr_frag_x2(lds_cp base, bf16x8_t (&out)[2]) {
+  bf16x4_tr a0l, a0h, a1l, a1h;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+      "ds_read_b64_tr_b16 %1, %4 offset:512\n\t"
+      "ds_read_b64_tr_b16 %2, %4 offset:1024\n\t"
+      "ds_read_b64_tr_b16 %3, %4 offset:1536\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a0l), "=&v"(a0h), "=&v"(a1l), "=&v"(a1h)
+      : "v"(base)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  out[0] = __builtin_shufflevector(a0l, a0h, 0, 1, 2, 3, 4, 5, 6, 7);
+  out[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
 // ---------------------------------------------------------------------------
 // D = rowsum(dO * O), stride-aware
 // ---------------------------------------------------------------------------
@@ -113,9 +171,9 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int NT = DP / 16;
   constexpr int NCG = (DP + 63) / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* qt_lds = reinterpret_cast<short*>(smem);   // Q^T  [DP d][PITCH]
-  short* dot_lds = qt_lds + DP * PITCH;             // dO^T [DP d][PITCH]
-  short* qr_lds = dot_lds + DP * PITCH;             // Q    [64 q][PITCH]
+  short* qb_lds = reinterpret_cast<short*>(smem);   // Q  block image [64 q][DP d]
+  short* dob_lds = qb_lds + BLK * DP;               // dO block image
+  short* qr_lds = dob_lds + BLK * DP;               // Q    [64 q][PITCH] row image
   short* dor_lds = qr_lds + BLK * PITCH;            // dO   [64 q][PITCH]
   short* p_lds = dor_lds + BLK * PITCH;
 
@@ -192,17 +250,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         for (int hh = 0; hh < 2; ++hh) {
           *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + c0 + hh * 8) = qreg[cg][hh];
           *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + c0 + hh * 8) = doreg[cg][hh];
-#pragma unroll
-          for (int i = 0; i < 8; ++i) qt_lds[(c0 + hh * 8 + i) * PITCH + st_row] = qreg[cg][hh][i];
-#pragma unroll
-          for (int i = 0; i < 8; ++i) dot_lds[(c0 + hh * 8 + i) * PITCH + st_row] = doreg[cg][hh][i];
+          *reinterpret_cast<bf16x8_t*>(qb_lds + boff<DP>(st_row, c0 + hh * 8)) = qreg[cg][hh];
+          *reinterpret_cast<bf16x8_t*>(dob_lds + boff<DP>(st_row, c0 + hh * 8)) = doreg[cg][hh];
         }
       } else {
-        for (int i = 0; i < 16; ++i) {
-          qr_lds[st_row * PITCH + c0 + i] = 0;
-          dor_lds[st_row * PITCH + c0 + i] = 0;
-          qt_lds[(c0 + i) * PITCH + st_row] = 0;
-          dot_lds[(c0 + i) * PITCH + st_row] = 0;
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) {
+          *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
+          *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
+          *reinterpret_cast<bf16x8_t*>(qb_lds + boff<DP>(st_row, c0 + hh * 8)) = bf16x8_t{};
+          *reinterpret_cast<bf16x8_t*>(dob_lds + boff<DP>(st_row, c0 + hh * 8)) = bf16x8_t{};
         }
       }
     }
@@ -288,20 +345,21 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         ds_stash[qt] = dsk;
       }
 
-      // ---- dV += P^T . dO  (A = P^T via LDS, B = dO^T rows) --------------
-      __builtin_amdgcn_s_setprio(1);
+      // ---- dV += P^T . dO  (A = P^T via LDS, B = dO^T via tr reads) ------
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
         if ((CAUSAL && q0 + 32 * s + 31 < key_min) || q0 + 32 * s >= Lq) continue;
+        bf16x8_t bfr[NT];
+        const lds_cp bbase = (lds_cp)(const void*)(dob_lds + s * 32 * DP) + lane * 8;
+        tr_frag_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[0]));
+        if constexpr (NT == 6) tr_frag_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&bfr[4]));
+        if constexpr (NT == 8) tr_frag_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[4]));
         const bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int dt = 0; dt < NT; ++dt) {
-          const bf16x8_t bfrag =
-              *reinterpret_cast<const bf16x8_t*>(dot_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
-          acc_dv[sidx][dt] = MFMA16(pa, bfrag, acc_dv[sidx][dt]);
-        }
+        for (int dt = 0; dt < NT; ++dt) acc_dv[sidx][dt] = MFMA16(pa, bfr[dt], acc_dv[sidx][dt]);
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
 
       // ---- overwrite the wave tile with dS^T, then dK += dS^T . Q --------
 #pragma unroll
@@ -309,19 +367,20 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) my_p[(hi * 4 + r) * PITCH + 16 * qt + lo] = ds_stash[qt][r];
       }
-      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
         if ((CAUSAL && q0 + 32 * s + 31 < key_min) || q0 + 32 * s >= Lq) continue;
+        bf16x8_t bfr[NT];
+        const lds_cp bbase = (lds_cp)(const void*)(qb_lds + s * 32 * DP) + lane * 8;
+        tr_frag_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[0]));
+        if constexpr (NT == 6) tr_frag_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&bfr[4]));
+        if constexpr (NT == 8) tr_frag_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[4]));
         const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_p + lo * PITCH + 32 * s + hi * 8);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int dt = 0; dt < NT; ++dt) {
-          const bf16x8_t bfrag =
-              *reinterpret_cast<const bf16x8_t*>(qt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
-          acc_dk[sidx][dt] = MFMA16(dsa, bfrag, acc_dk[sidx][dt]);
-        }
+        for (int dt = 0; dt < NT; ++dt) acc_dk[sidx][dt] = MFMA16(dsa, bfr[dt], acc_dk[sidx][dt]);
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     // T14: all waves done reading this q tile -> overwrite with tile it+1
     __syncthreads();
@@ -373,8 +432,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   constexpr int NT = DP / 16;
   constexpr int NCG = (DP + 63) / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* kt_lds = reinterpret_cast<short*>(smem);     // K^T [DP d][PITCH]
-  short* kr_lds = kt_lds + DP * PITCH;                // K   [64 key][PITCH]
+  short* kb_lds = reinterpret_cast<short*>(smem);     // K block image [64 key][DP d]
+  short* kr_lds = kb_lds + BLK * DP;                  // K   [64 key][PITCH]
   short* vr_lds = kr_lds + BLK * PITCH;               // V   [64 key][PITCH]
   short* ds_lds = vr_lds + BLK * PITCH;
 
@@ -456,14 +515,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
         for (int hh = 0; hh < 2; ++hh) {
           *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + c0 + hh * 8) = kreg[cg][hh];
           *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + c0 + hh * 8) = vreg[cg][hh];
-#pragma unroll
-          for (int i = 0; i < 8; ++i) kt_lds[(c0 + hh * 8 + i) * PITCH + st_row] = kreg[cg][hh][i];
+          *reinterpret_cast<bf16x8_t*>(kb_lds + boff<DP>(st_row, c0 + hh * 8)) = kreg[cg][hh];
         }
       } else {
-        for (int i = 0; i < 16; ++i) {
-          kr_lds[st_row * PITCH + c0 + i] = 0;
-          vr_lds[st_row * PITCH + c0 + i] = 0;
-          kt_lds[(c0 + i) * PITCH + st_row] = 0;
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) {
+          *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
+          *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
+          *reinterpret_cast<bf16x8_t*>(kb_lds + boff<DP>(st_row, c0 + hh * 8)) = bf16x8_t{};
         }
       }
     }
@@ -538,20 +597,21 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
         for (int r = 0; r < 4; ++r) my_ds[(hi * 4 + r) * PITCH + 16 * kt + lo] = dsk[r];
       }
 
-      // ---- dQ += dS . K (A = dS via LDS, B = K from K^T rows) ------------
-      __builtin_amdgcn_s_setprio(1);
+      // ---- dQ += dS . K (A = dS via LDS, B = K^T via tr reads) -----------
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
         if ((CAUSAL && kv0 + 32 * s > q_max) || kv0 + 32 * s >= Lk) continue;
+        bf16x8_t bfr[NT];
+        const lds_cp bbase = (lds_cp)(const void*)(kb_lds + s * 32 * DP) + lane * 8;
+        tr_frag_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[0]));
+        if constexpr (NT == 6) tr_frag_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&bfr[4]));
+        if constexpr (NT == 8) tr_frag_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[4]));
         const bf16x8_t dsa = *reinterpret_cast<const bf16x8_t*>(my_ds + lo * PITCH + 32 * s + hi * 8);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int dt = 0; dt < NT; ++dt) {
-          const bf16x8_t bfrag =
-              *reinterpret_cast<const bf16x8_t*>(kt_lds + (16 * dt + lo) * PITCH + 32 * s + hi * 8);
-          acc_dq[sidx][dt] = MFMA16(dsa, bfrag, acc_dq[sidx][dt]);
-        }
+        for (int dt = 0; dt < NT; ++dt) acc_dq[sidx][dt] = MFMA16(dsa, bfr[dt], acc_dq[sidx][dt]);
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     // T14: overwrite the single-buffer images with tile it+1 after everyone
     // is done reading, then issue tile it+2's loads
@@ -611,8 +671,9 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   }
 
   const int pitch = DP + 8;
-  const size_t shmem_dkv = ((2 * DP + 2 * BLK + 4 * 16) * pitch) * sizeof(short);
-  const size_t shmem_dq = ((DP + 2 * BLK + 4 * 16) * pitch) * sizeof(short);
+  const size_t shmem_dkv =
+      (2 * BLK * DP + (2 * BLK + 4 * 16) * pitch) * sizeof(short);
+  const size_t shmem_dq = (BLK * DP + (2 * BLK + 4 * 16) * pitch) * sizeof(short);
   const int ntk = (Lk + BLK - 1) / BLK;
   static const int nkv_env = [] {
     const char* e = getenv("JIMM_AMD_ATTN_NKV");
