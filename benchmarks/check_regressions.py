@@ -17,14 +17,26 @@ import jimm_amd  # noqa: F401
 from jimm_amd.ops import _backend
 from jimm_amd.ops._backend import maybe_enable_tunableop
 
-# round-1 recorded values (TF/s), with a 10% regression margin applied
+# round-2 recorded values (TF/s), with a ~10% regression margin applied
+# (r02: block-image attention + small-L kernel + in-house GEMM default)
 FLOORS_ATTN = {  # (B, H, L, causal): (fwd_tf, bwd_tf)
-    (256, 12, 197, False): (160, 140),
-    (64, 16, 577, False): (255, 225),
-    (256, 8, 77, True): (55, 38),
-    (256, 12, 257, False): (195, 150),
+    (256, 12, 197, False): (145, 145),
+    (64, 16, 577, False): (255, 220),
+    (256, 8, 77, True): (60, 42),
+    (256, 12, 257, False): (185, 155),
 }
-FLOOR_STEP_VIT_B1024 = 5100  # img/s
+FLOOR_STEP_VIT_B1024 = 5250  # img/s (r02: 5430 with in-house GEMM default)
+# in-house GEMM floors on the b1024 model shapes (median TF/s, -10%)
+FLOORS_GEMM = {  # (M, N, K, act): fwd_tf
+    (201728, 2304, 768, ""): 790,
+    (201728, 3072, 768, "gelu"): 600,
+    (201728, 768, 3072, ""): 890,
+}
+FLOORS_DW = {  # (M, N, K): tf
+    (201728, 2304, 768): 620,
+    (201728, 3072, 768): 650,
+    (201728, 768, 3072): 650,
+}
 
 
 def bench(fn, iters=20, warmup=6):
@@ -59,6 +71,25 @@ def main():
               f"bwd {r_bwd:6.1f} TF/s [{status_b}]", flush=True)
         if "REGRESSION" in (status_f, status_b):
             failures.append((B, H, L, causal))
+    for (M, N, K, act), floor in FLOORS_GEMM.items():
+        x = (torch.rand(M, K, device=dev) * 2 - 1).bfloat16()
+        w = ((torch.rand(N, K, device=dev) * 2 - 1) / K**0.5).bfloat16()
+        b = torch.randn(N, device=dev).bfloat16()
+        tf = 2 * M * N * K / 1e12
+        r = tf / bench(lambda: ext.linear_fwd(x, w, b, act, None, bool(act)), iters=10)
+        st = "OK" if r > 0.9 * floor else "REGRESSION"
+        print(f"gemm fwd M={M} N={N:5d} K={K:5d} {act or '-':5s}: {r:6.1f} TF/s [{st}]", flush=True)
+        if st == "REGRESSION":
+            failures.append((M, N, K, act))
+    for (M, N, K), floor in FLOORS_DW.items():
+        dz = (torch.rand(M, N, device=dev) * 2 - 1).bfloat16()
+        x = ((torch.rand(M, K, device=dev) * 2 - 1) / 8).bfloat16()
+        tf = 2 * M * N * K / 1e12
+        r = tf / bench(lambda: ext.gemm_tn_8p(dz, x), iters=10)
+        st = "OK" if r > 0.9 * floor else "REGRESSION"
+        print(f"gemm dW  M={M} N={N:5d} K={K:5d}: {r:6.1f} TF/s [{st}]", flush=True)
+        if st == "REGRESSION":
+            failures.append((M, N, K))
     if failures:
         print("REGRESSIONS:", failures)
         sys.exit(1)
