@@ -171,11 +171,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int NT = DP / 16;
   constexpr int NCG = (DP + 63) / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // staging images are DOUBLE-buffered: the write pass for tile it+1 runs
+  // before/under tile it's MFMAs into the other buffer, so each q-tile
+  // needs ONE __syncthreads instead of two (the single-buffer version
+  // parked 41-44% of wave time at its write-drain barriers)
+  constexpr int IMGS = 2 * BLK * DP + 2 * BLK * PITCH;  // per buffer
   short* qb_lds = reinterpret_cast<short*>(smem);   // Q  block image [64 q][DP d]
   short* dob_lds = qb_lds + BLK * DP;               // dO block image
   short* qr_lds = dob_lds + BLK * DP;               // Q    [64 q][PITCH] row image
   short* dor_lds = qr_lds + BLK * PITCH;            // dO   [64 q][PITCH]
-  short* p_lds = dor_lds + BLK * PITCH;
+  short* p_lds = reinterpret_cast<short*>(smem) + 2 * IMGS;
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -240,38 +245,36 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       }
     }
   };
-  auto write_stage = [&]() {
+  auto write_stage = [&](int buf) {
+    const int bo = buf * IMGS;
 #pragma unroll
     for (int cg = 0; cg < NCG; ++cg) {
       const int c0 = cg * 64 + st_c0;
       if (c0 >= DP) continue;
-      if (st_valid) {
 #pragma unroll
-        for (int hh = 0; hh < 2; ++hh) {
-          *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + c0 + hh * 8) = qreg[cg][hh];
-          *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + c0 + hh * 8) = doreg[cg][hh];
-          *reinterpret_cast<bf16x8_t*>(qb_lds + boff<DP>(st_row, c0 + hh * 8)) = qreg[cg][hh];
-          *reinterpret_cast<bf16x8_t*>(dob_lds + boff<DP>(st_row, c0 + hh * 8)) = doreg[cg][hh];
-        }
-      } else {
-#pragma unroll
-        for (int hh = 0; hh < 2; ++hh) {
-          *reinterpret_cast<bf16x8_t*>(qr_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
-          *reinterpret_cast<bf16x8_t*>(dor_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
-          *reinterpret_cast<bf16x8_t*>(qb_lds + boff<DP>(st_row, c0 + hh * 8)) = bf16x8_t{};
-          *reinterpret_cast<bf16x8_t*>(dob_lds + boff<DP>(st_row, c0 + hh * 8)) = bf16x8_t{};
-        }
+      for (int hh = 0; hh < 2; ++hh) {
+        const bf16x8_t qv = st_valid ? qreg[cg][hh] : bf16x8_t{};
+        const bf16x8_t dv2 = st_valid ? doreg[cg][hh] : bf16x8_t{};
+        *reinterpret_cast<bf16x8_t*>(qr_lds + bo + st_row * PITCH + c0 + hh * 8) = qv;
+        *reinterpret_cast<bf16x8_t*>(dor_lds + bo + st_row * PITCH + c0 + hh * 8) = dv2;
+        *reinterpret_cast<bf16x8_t*>(qb_lds + bo + boff<DP>(st_row, c0 + hh * 8)) = qv;
+        *reinterpret_cast<bf16x8_t*>(dob_lds + bo + boff<DP>(st_row, c0 + hh * 8)) = dv2;
       }
     }
   };
 
   load_stage_regs(q_start);
-  write_stage();
+  write_stage(0);
   if (ntiles > 1) load_stage_regs(q_start + BLK);
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int q0 = q_start + it * BLK;
+    const int sbo = (it & 1) * IMGS;  // this tile's staging buffer offset
+    if (it + 1 < ntiles) {
+      write_stage((it + 1) & 1);
+      if (it + 2 < ntiles) load_stage_regs(q_start + (it + 2) * BLK);
+    }
     // ---- B-fragments of Q^T and dO^T from the LDS row images (shared by
     // every kv strip) ------------------------------------------------------
     // DP==64 caches the q-tile fragments across the NKV strips; larger DP
@@ -282,8 +285,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       for (int qt = 0; qt < 4; ++qt) {
 #pragma unroll
         for (int s = 0; s < NS; ++s) {
-          qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qr_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
-          dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dor_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+          qb[qt][s] = *reinterpret_cast<const bf16x8_t*>(qr_lds + sbo + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+          dob[qt][s] = *reinterpret_cast<const bf16x8_t*>(dor_lds + sbo + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
         }
       }
     }
@@ -321,8 +324,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
             qf = qb[qt][s];
             dof = dob[qt][s];
           } else {
-            qf = *reinterpret_cast<const bf16x8_t*>(qr_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
-            dof = *reinterpret_cast<const bf16x8_t*>(dor_lds + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+            qf = *reinterpret_cast<const bf16x8_t*>(qr_lds + sbo + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
+            dof = *reinterpret_cast<const bf16x8_t*>(dor_lds + sbo + (16 * qt + lo) * PITCH + 32 * s + hi * 8);
           }
           sc = MFMA16(ka[sidx][s], qf, sc);
           dpc = MFMA16(va[sidx][s], dof, dpc);
@@ -350,7 +353,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       for (int s = 0; s < 2; ++s) {
         if ((CAUSAL && q0 + 32 * s + 31 < key_min) || q0 + 32 * s >= Lq) continue;
         bf16x8_t bfr[NT];
-        const lds_cp bbase = (lds_cp)(const void*)(dob_lds + s * 32 * DP) + lane * 8;
+        const lds_cp bbase = (lds_cp)(const void*)(dob_lds + sbo + s * 32 * DP) + lane * 8;
         tr_frag_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[0]));
         if constexpr (NT == 6) tr_frag_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&bfr[4]));
         if constexpr (NT == 8) tr_frag_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[4]));
@@ -371,7 +374,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       for (int s = 0; s < 2; ++s) {
         if ((CAUSAL && q0 + 32 * s + 31 < key_min) || q0 + 32 * s >= Lq) continue;
         bf16x8_t bfr[NT];
-        const lds_cp bbase = (lds_cp)(const void*)(qb_lds + s * 32 * DP) + lane * 8;
+        const lds_cp bbase = (lds_cp)(const void*)(qb_lds + sbo + s * 32 * DP) + lane * 8;
         tr_frag_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[0]));
         if constexpr (NT == 6) tr_frag_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&bfr[4]));
         if constexpr (NT == 8) tr_frag_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[4]));
@@ -382,13 +385,10 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
     }
-    // T14: all waves done reading this q tile -> overwrite with tile it+1
+    // double-buffered: tile it+1's write pass ran into the other buffer
+    // before the MFMAs above; ONE barrier publishes it and retires this
+    // tile's reads
     __syncthreads();
-    if (it + 1 < ntiles) {
-      write_stage();
-      if (it + 2 < ntiles) load_stage_regs(q_start + (it + 2) * BLK);
-      __syncthreads();
-    }
   }
 
   // ---- store dK, dV (strided, bf16) ---------------------------------------
@@ -672,7 +672,7 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
 
   const int pitch = DP + 8;
   const size_t shmem_dkv =
-      (2 * BLK * DP + (2 * BLK + 4 * 16) * pitch) * sizeof(short);
+      (2 * (2 * BLK * DP + 2 * BLK * pitch) + 4 * 16 * pitch) * sizeof(short);
   const size_t shmem_dq = (BLK * DP + (2 * BLK + 4 * 16) * pitch) * sizeof(short);
   const int ntk = (Lk + BLK - 1) / BLK;
   static const int nkv_env = [] {
