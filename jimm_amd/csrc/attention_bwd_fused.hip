@@ -432,10 +432,12 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   constexpr int NT = DP / 16;
   constexpr int NCG = (DP + 63) / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // double-buffered staging (as in the dkv kernel): one barrier per tile
+  constexpr int IMGSQ = BLK * DP + 2 * BLK * PITCH;  // per buffer
   short* kb_lds = reinterpret_cast<short*>(smem);     // K block image [64 key][DP d]
   short* kr_lds = kb_lds + BLK * DP;                  // K   [64 key][PITCH]
   short* vr_lds = kr_lds + BLK * PITCH;               // V   [64 key][PITCH]
-  short* ds_lds = vr_lds + BLK * PITCH;
+  short* ds_lds = reinterpret_cast<short*>(smem) + 2 * IMGSQ;
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -505,36 +507,35 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       }
     }
   };
-  auto write_stage = [&]() {
+  auto write_stage = [&](int buf) {
+    const int bo = buf * IMGSQ;
 #pragma unroll
     for (int cg = 0; cg < NCG; ++cg) {
       const int c0 = cg * 64 + st_c0;
       if (c0 >= DP) continue;
-      if (st_valid) {
 #pragma unroll
-        for (int hh = 0; hh < 2; ++hh) {
-          *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + c0 + hh * 8) = kreg[cg][hh];
-          *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + c0 + hh * 8) = vreg[cg][hh];
-          *reinterpret_cast<bf16x8_t*>(kb_lds + boff<DP>(st_row, c0 + hh * 8)) = kreg[cg][hh];
-        }
-      } else {
-#pragma unroll
-        for (int hh = 0; hh < 2; ++hh) {
-          *reinterpret_cast<bf16x8_t*>(kr_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
-          *reinterpret_cast<bf16x8_t*>(vr_lds + st_row * PITCH + c0 + hh * 8) = bf16x8_t{};
-          *reinterpret_cast<bf16x8_t*>(kb_lds + boff<DP>(st_row, c0 + hh * 8)) = bf16x8_t{};
-        }
+      for (int hh = 0; hh < 2; ++hh) {
+        const bf16x8_t kv2 = st_valid ? kreg[cg][hh] : bf16x8_t{};
+        const bf16x8_t vv2 = st_valid ? vreg[cg][hh] : bf16x8_t{};
+        *reinterpret_cast<bf16x8_t*>(kr_lds + bo + st_row * PITCH + c0 + hh * 8) = kv2;
+        *reinterpret_cast<bf16x8_t*>(vr_lds + bo + st_row * PITCH + c0 + hh * 8) = vv2;
+        *reinterpret_cast<bf16x8_t*>(kb_lds + bo + boff<DP>(st_row, c0 + hh * 8)) = kv2;
       }
     }
   };
 
   load_stage_regs(0);
-  write_stage();
+  write_stage(0);
   if (ntiles > 1) load_stage_regs(BLK);
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * BLK;
+    const int sbo = (it & 1) * IMGSQ;
+    if (it + 1 < ntiles) {
+      write_stage((it + 1) & 1);
+      if (it + 2 < ntiles) load_stage_regs(kv0 + 2 * BLK);
+    }
     // ---- B-fragments of K^T and V^T from the LDS row images (shared) -----
     bf16x8_t kb[DP == 64 ? 4 : 1][NS], vb[DP == 64 ? 4 : 1][NS];
     if constexpr (DP == 64) {
@@ -542,8 +543,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       for (int kt = 0; kt < 4; ++kt) {
 #pragma unroll
         for (int s = 0; s < NS; ++s) {
-          kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
-          vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+          kb[kt][s] = *reinterpret_cast<const bf16x8_t*>(kr_lds + sbo + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+          vb[kt][s] = *reinterpret_cast<const bf16x8_t*>(vr_lds + sbo + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
         }
       }
     }
@@ -577,8 +578,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
             kf = kb[kt][s];
             vf = vb[kt][s];
           } else {
-            kf = *reinterpret_cast<const bf16x8_t*>(kr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
-            vf = *reinterpret_cast<const bf16x8_t*>(vr_lds + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+            kf = *reinterpret_cast<const bf16x8_t*>(kr_lds + sbo + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
+            vf = *reinterpret_cast<const bf16x8_t*>(vr_lds + sbo + (16 * kt + lo) * PITCH + 32 * s + hi * 8);
           }
           sc = MFMA16(qa[sidx][s], kf, sc);
           dpc = MFMA16(doa[sidx][s], vf, dpc);
@@ -602,7 +603,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       for (int s = 0; s < 2; ++s) {
         if ((CAUSAL && kv0 + 32 * s > q_max) || kv0 + 32 * s >= Lk) continue;
         bf16x8_t bfr[NT];
-        const lds_cp bbase = (lds_cp)(const void*)(kb_lds + s * 32 * DP) + lane * 8;
+        const lds_cp bbase = (lds_cp)(const void*)(kb_lds + sbo + s * 32 * DP) + lane * 8;
         tr_frag_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[0]));
         if constexpr (NT == 6) tr_frag_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&bfr[4]));
         if constexpr (NT == 8) tr_frag_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&bfr[4]));
@@ -613,14 +614,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
     }
-    // T14: overwrite the single-buffer images with tile it+1 after everyone
-    // is done reading, then issue tile it+2's loads
+    // double-buffered: ONE barrier publishes tile it+1 and retires it
     __syncthreads();
-    if (it + 1 < ntiles) {
-      write_stage();
-      if (it + 2 < ntiles) load_stage_regs(kv0 + 2 * BLK);
-      __syncthreads();
-    }
   }
 
   // ---- store dQ (strided, bf16) -------------------------------------------
@@ -673,7 +668,7 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   const int pitch = DP + 8;
   const size_t shmem_dkv =
       (2 * (2 * BLK * DP + 2 * BLK * pitch) + 4 * 16 * pitch) * sizeof(short);
-  const size_t shmem_dq = (BLK * DP + (2 * BLK + 4 * 16) * pitch) * sizeof(short);
+  const size_t shmem_dq = (2 * (BLK * DP + 2 * BLK * pitch) + 4 * 16 * pitch) * sizeof(short);
   const int ntk = (Lk + BLK - 1) / BLK;
   static const int nkv_env = [] {
     const char* e = getenv("JIMM_AMD_ATTN_NKV");
