@@ -122,10 +122,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   constexpr int LDS_PITCH = DP + 8;  // bank-conflict pad for b128 reads
   constexpr int NS = DP / 32;        // d-steps per fragment contraction
   constexpr int NT = DP / 16;        // d-tiles of the output
+  // K/V staging double-buffered: tile it+1's write pass runs under tile
+  // it's compute into the other buffer; ONE barrier per kv tile
+  constexpr int IMGSF = KVBLK * LDS_PITCH + KVBLK * DP;  // per buffer
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);                       // KVBLK rows
   short* vt_lds = k_lds + KVBLK * LDS_PITCH;                           // V block image [KVBLK][DP]
-  short* p_lds = vt_lds + KVBLK * DP;                                  // 4*16 rows
+  short* p_lds = reinterpret_cast<short*>(smem) + 2 * IMGSF;           // 4*16 rows
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
@@ -192,33 +195,36 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       vreg[cg][1] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0 + 8, Dr);
     }
   };
-  auto write_tile = [&]() {
+  auto write_tile = [&](int buf) {
+    const int bo = buf * IMGSF;
 #pragma unroll
     for (int cg = 0; cg < NCG; ++cg) {
       const int c0 = cg * 64 + st_c0;
       if (c0 >= DP) continue;
-      if (st_valid) {
-        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0) = kreg[cg][0];
-        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0 + 8) = kreg[cg][1];
-        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0)) = vreg[cg][0];
-        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0 + 8)) = vreg[cg][1];
-      } else {
-        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0) = bf16x8_t{};
-        *reinterpret_cast<bf16x8_t*>(k_lds + st_row * LDS_PITCH + c0 + 8) = bf16x8_t{};
-        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0)) = bf16x8_t{};
-        *reinterpret_cast<bf16x8_t*>(vt_lds + boff_f<DP>(st_row, c0 + 8)) = bf16x8_t{};
-      }
+      const bf16x8_t k0 = st_valid ? kreg[cg][0] : bf16x8_t{};
+      const bf16x8_t k1 = st_valid ? kreg[cg][1] : bf16x8_t{};
+      const bf16x8_t v0 = st_valid ? vreg[cg][0] : bf16x8_t{};
+      const bf16x8_t v1 = st_valid ? vreg[cg][1] : bf16x8_t{};
+      *reinterpret_cast<bf16x8_t*>(k_lds + bo + st_row * LDS_PITCH + c0) = k0;
+      *reinterpret_cast<bf16x8_t*>(k_lds + bo + st_row * LDS_PITCH + c0 + 8) = k1;
+      *reinterpret_cast<bf16x8_t*>(vt_lds + bo + boff_f<DP>(st_row, c0)) = v0;
+      *reinterpret_cast<bf16x8_t*>(vt_lds + bo + boff_f<DP>(st_row, c0 + 8)) = v1;
     }
   };
 
   const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
   load_tile_regs(0);
-  write_tile();
+  write_tile(0);
   if (ntiles > 1) load_tile_regs(KVBLK);
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * KVBLK;
+    const int sbo = (it & 1) * IMGSF;
+    if (it + 1 < ntiles) {
+      write_tile((it + 1) & 1);
+      if (it + 2 < ntiles) load_tile_regs(kv0 + 2 * KVBLK);
+    }
 #pragma unroll
     for (int st = 0; st < NSTRIP; ++st) {
       const int ti = blockIdx.x + st * gridDim.x;
@@ -239,7 +245,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
         for (int s = 0; s < NS; ++s) {
           const bf16x8_t ka =
-              *reinterpret_cast<const bf16x8_t*>(k_lds + (16 * t + lo) * LDS_PITCH + 32 * s + hi * 8);
+              *reinterpret_cast<const bf16x8_t*>(k_lds + sbo + (16 * t + lo) * LDS_PITCH + 32 * s + hi * 8);
           sc[t] = MFMA16(ka, qb[st][s], sc[t]);
         }
       }
@@ -303,7 +309,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int s = 0; s < 2; ++s) {
         if (kv0 + 32 * s >= kv_hi) continue;  // P = 0 for the whole step
         bf16x8_t vb[NT];
-        const lds_cpf bbase = (lds_cpf)(const void*)(vt_lds + s * 32 * DP) + lane * 8;
+        const lds_cpf bbase = (lds_cpf)(const void*)(vt_lds + sbo + s * 32 * DP) + lane * 8;
         trf_x4(bbase, *reinterpret_cast<bf16x8_t(*)[4]>(&vb[0]));
         if constexpr (NT == 6) trf_x2(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[2]>(&vb[4]));
         if constexpr (NT == 8) trf_x4(bbase + 4096, *reinterpret_cast<bf16x8_t(*)[4]>(&vb[4]));
@@ -315,13 +321,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
     }
-    // T14: overwrite the K/V tiles with tile it+1 after everyone is done
+    // double-buffered: ONE barrier publishes tile it+1 and retires it
     __syncthreads();
-    if (it + 1 < ntiles) {
-      write_tile();
-      if (it + 2 < ntiles) load_tile_regs(kv0 + 2 * KVBLK);
-      __syncthreads();
-    }
   }
 
   // ---- epilogue: O /= l, store O and lse ----------------------------------
@@ -537,7 +538,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
     if (const char* e = getenv("JIMM_AMD_ATTN_NSTRIP")) nstrip = (std::string(e) == "2") ? 2 : 4;
   }
   const dim3 grid((ntq + nstrip - 1) / nstrip, (unsigned)((int64_t)B * H));
-  const size_t shmem = (KVBLK * DP + (KVBLK + 4 * 16) * (DP + 8)) * sizeof(short);
+  const size_t shmem =
+      (2 * (KVBLK * DP + KVBLK * (DP + 8)) + 4 * 16 * (DP + 8)) * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
 #define ATTN_ARGS                                                                          \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \
