@@ -689,3 +689,39 @@ def test_fp32_clip_gpu_vs_cpu_oracle():
         got, _ = model.to(dev())(imgs.to(dev()), ids.to(dev()))
     err = (got.cpu() - ref).abs().max().item()
     assert err < 1e-4, err
+
+
+# ---------------------------------------------------------------------------
+def test_bf16_checkpoint_resume_step(tmp_path):
+    """bf16 save -> load -> step. The base-class
+    load_state_dict casts fp32 moments/masters to the param dtype; the Adam
+    override must restore them to fp32 or the fused kernel's
+    data_ptr<float>() fails on the next GPU step."""
+    import jimm_amd as J
+    from jimm_amd.train import TrainConfig, Trainer
+    from jimm_amd.train import SyntheticImages
+
+    torch.manual_seed(0)
+    m = J.VisionTransformer(num_classes=10, img_size=32, patch_size=16,
+                            num_layers=1, num_heads=4, mlp_dim=256, hidden_size=64)
+    m = m.to(dev(), torch.bfloat16)
+    tr = Trainer(m, TrainConfig(task="vit", lr=1e-3))
+    data = iter(SyntheticImages(8, 32, 10, dev(), dtype=torch.bfloat16, seed=1))
+    for _ in range(2):
+        tr.train_step(next(data))
+    ck = str(tmp_path / "ck.pt")
+    tr.save_checkpoint(ck)
+
+    m2 = J.VisionTransformer(num_classes=10, img_size=32, patch_size=16,
+                             num_layers=1, num_heads=4, mlp_dim=256, hidden_size=64)
+    m2 = m2.to(dev(), torch.bfloat16)
+    tr2 = Trainer(m2, TrainConfig(task="vit", lr=1e-3))
+    tr2.load_checkpoint(ck)
+    # moments and masters must be fp32 after the load
+    for st in tr2.opt.state.values():
+        for kk in ("m", "v", "master"):
+            if kk in st and st[kk] is not None:
+                assert st[kk].dtype == torch.float32, (kk, st[kk].dtype)
+    # and the fused GPU step must run (would raise data_ptr<float> before)
+    out = tr2.train_step(next(data))
+    assert torch.isfinite(out["loss"]).item()
