@@ -373,7 +373,9 @@ __global__ __launch_bounds__(256) void attn_fwd_small_kernel(
     int64_t BH) {
   constexpr int D = 64;
   constexpr int LP = 96;      // padded key count (block images need 32-multiples)
-  constexpr int PPITCH = 88;  // P-tile pitch (16-B-aligned rows)
+  constexpr int PPITCH = 96;  // P-tile pitch: the PV A-frag reads key
+                              // columns up to 95 (s=2 covers keys 64..95),
+                              // so the row must span the full padded range
   // K lives entirely in REGISTERS (5 kt x 2 s fragments = 40 VGPR/lane);
   // only the V block image + the P tile are in LDS: ~15 KiB per wave ->
   // two 4-wave workgroups per CU (the v1 with a K image was 1 wave/SIMD
@@ -388,6 +390,12 @@ __global__ __launch_bounds__(256) void attn_fwd_small_kernel(
 
   const int64_t bh = (int64_t)blockIdx.x * 4 + wave;
   if (bh >= BH) return;  // wave-uniform; no barriers anywhere in this kernel
+  // zero the P tile once: the PV step reads key columns beyond Lk whose P
+  // is never written (their V rows are zero-padded, but garbage LDS can
+  // hold NaN bits and NaN * 0 = NaN — seen as an intermittent test failure)
+#pragma unroll
+  for (int i = 0; i < 16 * PPITCH / (WAVE * 8); ++i)
+    *reinterpret_cast<bf16x8_t*>(pt + (i * WAVE + lane) * 8) = bf16x8_t{};
   const int64_t b = bh / H, h = bh % H;
   const bf16* qp = q + b * q_sb + h * q_sh;
   const bf16* kp = k + b * k_sb + h * k_sh;
@@ -554,7 +562,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   if (Dr == 64 && Lq == Lk && Lk <= 80) {
     const int64_t BH = (int64_t)B * H;
     const dim3 sgrid((unsigned)((BH + 3) / 4));
-    const size_t sshmem = 4 * (96 * 64 + 16 * 88) * sizeof(short);
+    const size_t sshmem = 4 * (96 * 64 + 16 * 96) * sizeof(short);
 #define SMALL_ARGS                                                                         \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                          \

@@ -84,7 +84,7 @@ class EncoderBlock(nn.Module):
             return self._forward_tp(x)
         from jimm_amd.ops import block as fused
 
-        if fused.fused_block_enabled(x, self.dropout.p):
+        if fused.fused_block_enabled(x, self.dropout.p) and self.head_dim in (64, 72, 80, 96, 128):
             import math
 
             return fused.encoder_block(
