@@ -8,6 +8,12 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
                                          c10::optional<torch::Tensor> addend);
 torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, std::string act,
                            c10::optional<torch::Tensor> residual);
+std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, torch::Tensor bias,
+                                            std::string act, torch::Tensor scale8,
+                                            torch::Tensor amax);
+std::vector<torch::Tensor> layernorm_fwd_fp8(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                                             double eps, torch::Tensor scale8,
+                                             torch::Tensor amax);
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor z, std::string act);
 torch::Tensor colsum(torch::Tensor dz);
 torch::Tensor im2col_patch(torch::Tensor img, int64_t patch);
@@ -57,6 +63,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dy"), py::arg("x"), py::arg("w"), py::arg("mean"), py::arg("rstd"),
         py::arg("addend") = py::none());
   m.def("bias_act_fwd", &bias_act_fwd, "fused bias+activation(+residual) forward");
+  m.def("bias_act_fwd_fp8", &bias_act_fwd_fp8, "bias+act forward with fused e4m3 emit");
+  m.def("layernorm_fwd_fp8", &layernorm_fwd_fp8, "LayerNorm forward with fused e4m3 emit");
   m.def("act_bwd", &act_bwd, "activation backward: dy * act'(z)");
   m.def("colsum", &colsum, "column sum -> fp32 (bias gradient, K15)");
   m.def("im2col_patch", &im2col_patch, "patch-embed unfold (K1)");

@@ -9,6 +9,7 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <hip/hip_fp8.h>
 
 #include "common.h"
 
@@ -31,12 +32,18 @@ __global__ void bias_act_kernel(T* __restrict__ z, const T* __restrict__ bias,
   }
 }
 
-template <typename T, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE>
+template <typename T, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool FP8O = false>
 __global__ void bias_act_vec_kernel(T* __restrict__ z, const T* __restrict__ bias,
                                     const T* __restrict__ res, T* __restrict__ y,
-                                    int64_t nv, int Mv, int act) {
-  // vectorized x8 variant; requires M % 8 == 0 (Mv = M/8, nv = n/8)
+                                    int64_t nv, int Mv, int act,
+                                    unsigned char* __restrict__ y8 = nullptr,
+                                    const float* __restrict__ scale8 = nullptr,
+                                    unsigned int* __restrict__ amax_bits = nullptr) {
+  // vectorized x8 variant; requires M % 8 == 0 (Mv = M/8, nv = n/8).
+  // FP8O: also emit the delayed-scaled e4m3 copy of y (see layernorm.hip).
   constexpr int V = 8;
+  float rs8 = 1.f, tmax = 0.f;
+  if (FP8O) rs8 = 1.f / scale8[0];
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
        i += (int64_t)gridDim.x * blockDim.x) {
     float zv[V], o[V];
@@ -57,6 +64,19 @@ __global__ void bias_act_vec_kernel(T* __restrict__ z, const T* __restrict__ bia
       for (int j = 0; j < V; ++j) o[j] += rv[j];
     }
     vstore_f32<V>(y + i * V, o);
+    if (FP8O) {
+      unsigned char q8[V];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        tmax = fmaxf(tmax, fabsf(o[j]));
+        q8[j] = (unsigned char)__hip_cvt_float_to_fp8(o[j] * rs8, __HIP_SATFINITE, __HIP_E4M3);
+      }
+      *reinterpret_cast<uint2*>(y8 + i * V) = *reinterpret_cast<uint2*>(q8);
+    }
+  }
+  if (FP8O) {
+    tmax = wave_reduce_max(tmax);
+    if ((threadIdx.x % WAVE) == 0) atomicMax(amax_bits, __float_as_uint(tmax));
   }
 }
 
@@ -332,6 +352,30 @@ torch::Tensor colsum(torch::Tensor dz) {
     TORCH_CHECK(false, "colsum: unsupported dtype");
   }
   return ws.sum(0);  // (gy*rstep, N) fp32 reduce — tiny
+}
+
+std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, torch::Tensor bias,
+                                            std::string act, torch::Tensor scale8,
+                                            torch::Tensor amax) {
+  // bf16 z (+bias, act, save-pre in place) -> (y bf16, y8 e4m3 bytes)
+  TORCH_CHECK(z.is_cuda() && z.is_contiguous() && z.scalar_type() == torch::kBFloat16);
+  const int M = z.size(-1);
+  TORCH_CHECK(M % 8 == 0);
+  const int code = act_code(act);
+  auto b = bias.contiguous().to(torch::kBFloat16);
+  auto y = torch::empty_like(z);
+  auto y8 = torch::empty(z.sizes(), z.options().dtype(torch::kUInt8));
+  const int64_t n = z.numel();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((n / 8 + block - 1) / block, kMaxGrid);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL((bias_act_vec_kernel<bf16, true, false, true, true>), dim3(grid),
+                     dim3(block), 0, stream, reinterpret_cast<bf16*>(z.data_ptr()),
+                     reinterpret_cast<const bf16*>(b.data_ptr()), nullptr,
+                     reinterpret_cast<bf16*>(y.data_ptr()), n / 8, M / 8, code,
+                     y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
+                     reinterpret_cast<unsigned int*>(amax.data_ptr()));
+  return {y, y8};
 }
 
 torch::Tensor cls_pos_fwd(torch::Tensor x, c10::optional<torch::Tensor> cls,

@@ -14,19 +14,30 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <hip/hip_fp8.h>
 
 #include "common.h"
 
 namespace {
 
-template <typename T, int VEC>
+// FP8O: also emit a delayed-scaled e4m3 copy of y (producer-fused fp8
+// quantization, BASELINE config 5): y8 = y * (1/scale8[0]) saturated, and
+// this call's |y| max accumulates into amax_out (uint-bits atomicMax, one
+// per wave) — the NEXT step's scale is amax/448.  No extra read passes:
+// the quantization rides the existing LN write.
+template <typename T, int VEC, bool FP8O = false>
 __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                               const float* __restrict__ b, T* __restrict__ y,
                               float* __restrict__ mean_out, float* __restrict__ rstd_out,
-                              int64_t nrows, int H, float eps) {
+                              int64_t nrows, int H, float eps,
+                              unsigned char* __restrict__ y8 = nullptr,
+                              const float* __restrict__ scale8 = nullptr,
+                              unsigned int* __restrict__ amax_bits = nullptr) {
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int waves_per_block = blockDim.x / WAVE;
+  float rs8 = 1.f, tmax = 0.f;
+  if (FP8O) rs8 = 1.f / scale8[0];
   for (int64_t row = (int64_t)blockIdx.x * waves_per_block + wave; row < nrows;
        row += (int64_t)gridDim.x * waves_per_block) {
     const T* xr = x + row * H;
@@ -58,7 +69,24 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
 #pragma unroll
       for (int k = 0; k < VEC; ++k) yv[k] = (xv[k] - mean) * rstd * wv[k] + bv[k];
       vstore_f32<VEC>(yr + i, yv);
+      if (FP8O) {
+        unsigned char q8[VEC];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          tmax = fmaxf(tmax, fabsf(yv[k]));
+          q8[k] = (unsigned char)__hip_cvt_float_to_fp8(yv[k] * rs8, __HIP_SATFINITE, __HIP_E4M3);
+        }
+        if (VEC == 8) *reinterpret_cast<uint2*>(y8 + row * H + i) = *reinterpret_cast<uint2*>(q8);
+        else {
+#pragma unroll
+          for (int k = 0; k < VEC; ++k) y8[row * H + i + k] = q8[k];
+        }
+      }
     }
+  }
+  if (FP8O) {
+    tmax = wave_reduce_max(tmax);
+    if (lane == 0) atomicMax(amax_bits, __float_as_uint(tmax));
   }
 }
 
@@ -270,6 +298,32 @@ void launch_ln_bwd(const T* dy, const T* x, const float* w, const float* mean,
 }
 
 }  // namespace
+
+std::vector<torch::Tensor> layernorm_fwd_fp8(torch::Tensor x, torch::Tensor w,
+                                             torch::Tensor b, double eps,
+                                             torch::Tensor scale8, torch::Tensor amax) {
+  // bf16 in, (y bf16, y8 e4m3-bytes, mean, rstd) out; H % 512 == 0 (VEC 8)
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == torch::kBFloat16);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % (WAVE * 8) == 0, "layernorm_fwd_fp8: H % 512 != 0");
+  const int64_t nrows = x.numel() / H;
+  auto wf = w.contiguous().to(torch::kFloat32);
+  auto bf = b.contiguous().to(torch::kFloat32);
+  auto y = torch::empty_like(x);
+  auto y8 = torch::empty({nrows, (int64_t)H}, x.options().dtype(torch::kUInt8));
+  auto mean = torch::empty({nrows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({nrows}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>((nrows + 3) / 4, 2048);
+  hipLaunchKernelGGL((ln_fwd_kernel<bf16, 8, true>), dim3(grid), dim3(block), 0, stream,
+                     reinterpret_cast<const bf16*>(x.data_ptr()), wf.data_ptr<float>(),
+                     bf.data_ptr<float>(), reinterpret_cast<bf16*>(y.data_ptr()),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), nrows, H, 0.f + (float)eps,
+                     y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
+                     reinterpret_cast<unsigned int*>(amax.data_ptr()));
+  return {y, y8, mean, rstd};
+}
 
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                                          double eps) {

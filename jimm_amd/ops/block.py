@@ -34,6 +34,32 @@ def _hip_gemms(M: int, H: int) -> bool:
     return _gemm_mode() == "hip" and H % 256 == 0
 
 
+def _fp8_block_state(mod: torch.nn.Module, device: torch.device):
+    """Per-block delayed-scaling state (3 activation sites: h1->qkv,
+    h2->fc1, f->fc2): scale8 used by this step's producers, amax collected
+    this step, scale8 <- amax/448 at block end (next step's scale)."""
+    st = getattr(mod, "_fp8_state", None)
+    if st is None or st[0].device != device:
+        st = (
+            torch.ones(3, device=device, dtype=torch.float32),
+            torch.zeros(3, device=device, dtype=torch.float32),
+        )
+        mod._fp8_state = st
+    return st
+
+
+def _mm_fp8q(y8: torch.Tensor, s: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """GEMM on a producer-emitted e4m3 activation (uint8 bytes + its scale)
+    against a per-call-quantized e4m3 weight; bf16 out."""
+    from jimm_amd.ops.hip_linear import _quant_e4m3
+
+    w8, sw = _quant_e4m3(w)
+    return torch._scaled_mm(
+        y8.view(torch.float8_e4m3fn), w8.t(), scale_a=s.view(1, 1),
+        scale_b=sw.view(1, 1), out_dtype=torch.bfloat16,
+    )
+
+
 def fused_block_enabled(x: torch.Tensor, dropout_p: float) -> bool:
     return (
         x.is_cuda
@@ -48,6 +74,7 @@ class EncoderBlockFn(torch.autograd.Function):
     @staticmethod
     def forward(
         ctx, x, ln1w, ln1b, wqkv, bqkv, wproj, bproj, ln2w, ln2b, w1, b1, w2, b2,
+        scale8, amax8,
         num_heads: int, act: str, eps: float, causal: bool, scale: float,
     ):
         ext = _backend.ext()
@@ -55,34 +82,57 @@ class EncoderBlockFn(torch.autograd.Function):
         x = x.contiguous()
         x2 = x.view(-1, H)
         hip = _hip_gemms(B * L, H)
-        from jimm_amd.ops.hip_linear import _FP8_STATE, _fp8_ok, _gemm_nt_fp8
 
-        fp8 = _FP8_STATE["enabled"] and x.dtype == torch.bfloat16
-        h1, mean1, rstd1 = ext.layernorm_fwd(x, ln1w, ln1b, eps)
-        if fp8 and _fp8_ok(h1.view(-1, H), wqkv):
-            # BASELINE config 5 path: forward GEMMs on the fp8 MFMA pipe
-            # (e4m3 per-tensor dynamic scales, device-side — graph-safe);
-            # LN/softmax/losses and the whole backward stay bf16/fp32
-            qkv2 = ext.bias_act_fwd(_gemm_nt_fp8(h1.view(-1, H), wqkv), bqkv, "", None)
-        elif hip:
-            # in-house MFMA GEMMs with the bias / act / residual epilogues
-            # fused into the GEMM kernel itself (csrc/gemm8p.hip)
-            qkv2, _ = ext.linear_fwd(h1.view(-1, H), wqkv, bqkv, "", None, False)
+        # Producer-fused fp8 (BASELINE config 5): LN / bias-act kernels emit
+        # the e4m3 copy of their output alongside bf16 (one extra store, no
+        # standalone quantization pass) using last step's per-site scale;
+        # this step's amax feeds next step's scale (delayed scaling, all
+        # device-side — graph-safe). Backward stays bf16 on the saved
+        # activations. Sites: h1->qkv, h2->fc1, f->fc2; proj keeps the bf16
+        # fused-residual GEMM (its input o comes from attention, not from a
+        # producer kernel we control cheaply).
+        fp8 = (
+            scale8 is not None
+            and x.dtype == torch.bfloat16
+            and H % 512 == 0
+            and hasattr(torch, "_scaled_mm")
+        )
+        if fp8:
+            amax8.zero_()
+            h1, h1q, mean1, rstd1 = ext.layernorm_fwd_fp8(
+                x, ln1w, ln1b, eps, scale8[0:1], amax8[0:1]
+            )
+            qkv2 = ext.bias_act_fwd(_mm_fp8q(h1q, scale8[0:1], wqkv), bqkv, "", None)
         else:
-            qkv2 = torch.addmm(bqkv, h1.view(-1, H), wqkv.t())   # (M, 3H) fused bias
+            h1, mean1, rstd1 = ext.layernorm_fwd(x, ln1w, ln1b, eps)
+            if hip:
+                # in-house MFMA GEMMs with the bias / act / residual epilogues
+                # fused into the GEMM kernel itself (csrc/gemm8p.hip)
+                qkv2, _ = ext.linear_fwd(h1.view(-1, H), wqkv, bqkv, "", None, False)
+            else:
+                qkv2 = torch.addmm(bqkv, h1.view(-1, H), wqkv.t())  # (M, 3H) fused bias
         qkv = qkv2.view(B, L, 3, num_heads, H // num_heads)
         q = qkv[:, :, 0].transpose(1, 2)
         k = qkv[:, :, 1].transpose(1, 2)
         v = qkv[:, :, 2].transpose(1, 2)
         o, lse = ext.attn_fwd(q, k, v, causal, scale)            # (B,nh,L,d), (B,L,nh,d) storage
         o2 = o.transpose(1, 2).reshape(-1, H)                    # free view
-        if fp8 and _fp8_ok(o2, wproj):
-            a = ext.bias_act_fwd(_gemm_nt_fp8(o2, wproj), bproj, "", x2)
+        if fp8:
+            if hip:
+                a, _ = ext.linear_fwd(o2, wproj, bproj, "", x2, False)
+            else:
+                a = ext.bias_act_fwd(torch.matmul(o2, wproj.t()), bproj, "", x2)
             a3 = a.view(B, L, H)
-            h2, mean2, rstd2 = ext.layernorm_fwd(a3, ln2w, ln2b, eps)
-            z1 = _gemm_nt_fp8(h2.view(-1, H), w1)
-            f = ext.bias_act_fwd(z1, b1, act, None)  # z1 -> pre-act in place
-            y = ext.bias_act_fwd(_gemm_nt_fp8(f, w2), b2, "", a)
+            h2, h2q, mean2, rstd2 = ext.layernorm_fwd_fp8(
+                a3, ln2w, ln2b, eps, scale8[1:2], amax8[1:2]
+            )
+            z1 = _mm_fp8q(h2q, scale8[1:2], w1)
+            # z1 -> pre-act in place; f8 = e4m3(f) fused into the same pass
+            f, f8 = ext.bias_act_fwd_fp8(z1, b1, act, scale8[2:3], amax8[2:3])
+            y = ext.bias_act_fwd(_mm_fp8q(f8, scale8[2:3], w2), b2, "", a)
+            # next step's scales (delayed): scale = amax / 448, floor for
+            # the first/degenerate steps
+            scale8.copy_(torch.clamp(amax8 / 448.0, min=1e-12))
         elif hip:
             a, _ = ext.linear_fwd(o2, wproj, bproj, "", x2, False)
             a3 = a.view(B, L, H)
@@ -162,12 +212,18 @@ class EncoderBlockFn(torch.autograd.Function):
             dh1.view(B, L, H), x, ln1w, mean1, rstd1, da3
         )
         return (dx, dln1w, dln1b, dwqkv, dbqkv, dwproj, dbproj, dln2w, dln2b,
-                dw1, db1, dw2, db2, None, None, None, None, None)
+                dw1, db1, dw2, db2, None, None, None, None, None, None, None)
 
 
 def encoder_block(x, norm1, qkv, proj, norm2, fc1, fc2, *, num_heads, act, eps, causal, scale):
+    from jimm_amd.ops.hip_linear import _FP8_STATE
+
+    scale8 = amax8 = None
+    if _FP8_STATE["enabled"] and x.is_cuda and x.dtype == torch.bfloat16:
+        scale8, amax8 = _fp8_block_state(norm1, x.device)
     return EncoderBlockFn.apply(
         x, norm1.weight, norm1.bias, qkv.weight, qkv.bias, proj.weight, proj.bias,
         norm2.weight, norm2.bias, fc1.weight, fc1.bias, fc2.weight, fc2.bias,
+        scale8, amax8,
         num_heads, act, eps, causal, scale,
     )
