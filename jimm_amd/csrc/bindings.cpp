@@ -8,7 +8,7 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
                                          c10::optional<torch::Tensor> addend);
 torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, std::string act,
                            c10::optional<torch::Tensor> residual);
-std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, torch::Tensor bias,
+std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, c10::optional<torch::Tensor> bias,
                                             std::string act, torch::Tensor scale8,
                                             torch::Tensor amax);
 std::vector<torch::Tensor> layernorm_fwd_fp8(torch::Tensor x, torch::Tensor w, torch::Tensor b,

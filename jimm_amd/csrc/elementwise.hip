@@ -354,27 +354,37 @@ torch::Tensor colsum(torch::Tensor dz) {
   return ws.sum(0);  // (gy*rstep, N) fp32 reduce — tiny
 }
 
-std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, torch::Tensor bias,
+std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, c10::optional<torch::Tensor> bias,
                                             std::string act, torch::Tensor scale8,
                                             torch::Tensor amax) {
-  // bf16 z (+bias, act, save-pre in place) -> (y bf16, y8 e4m3 bytes)
+  // bf16 z (+bias + save-pre in place when bias given; else z already IS the
+  // pre-activation, e.g. bias fused in the producing GEMM) -> act -> (y bf16,
+  // y8 e4m3 bytes)
   TORCH_CHECK(z.is_cuda() && z.is_contiguous() && z.scalar_type() == torch::kBFloat16);
   const int M = z.size(-1);
   TORCH_CHECK(M % 8 == 0);
   const int code = act_code(act);
-  auto b = bias.contiguous().to(torch::kBFloat16);
   auto y = torch::empty_like(z);
   auto y8 = torch::empty(z.sizes(), z.options().dtype(torch::kUInt8));
   const int64_t n = z.numel();
   const int block = 256;
   const int grid = (int)std::min<int64_t>((n / 8 + block - 1) / block, kMaxGrid);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL((bias_act_vec_kernel<bf16, true, false, true, true>), dim3(grid),
-                     dim3(block), 0, stream, reinterpret_cast<bf16*>(z.data_ptr()),
-                     reinterpret_cast<const bf16*>(b.data_ptr()), nullptr,
-                     reinterpret_cast<bf16*>(y.data_ptr()), n / 8, M / 8, code,
-                     y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
-                     reinterpret_cast<unsigned int*>(amax.data_ptr()));
+  if (bias) {
+    auto b = bias->contiguous().to(torch::kBFloat16);
+    hipLaunchKernelGGL((bias_act_vec_kernel<bf16, true, false, true, true>), dim3(grid),
+                       dim3(block), 0, stream, reinterpret_cast<bf16*>(z.data_ptr()),
+                       reinterpret_cast<const bf16*>(b.data_ptr()), nullptr,
+                       reinterpret_cast<bf16*>(y.data_ptr()), n / 8, M / 8, code,
+                       y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
+                       reinterpret_cast<unsigned int*>(amax.data_ptr()));
+  } else {
+    hipLaunchKernelGGL((bias_act_vec_kernel<bf16, false, false, false, true>), dim3(grid),
+                       dim3(block), 0, stream, reinterpret_cast<bf16*>(z.data_ptr()),
+                       nullptr, nullptr, reinterpret_cast<bf16*>(y.data_ptr()), n / 8,
+                       M / 8, code, y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
+                       reinterpret_cast<unsigned int*>(amax.data_ptr()));
+  }
   return {y, y8};
 }
 

@@ -48,15 +48,17 @@ def _fp8_block_state(mod: torch.nn.Module, device: torch.device):
     return st
 
 
-def _mm_fp8q(y8: torch.Tensor, s: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+def _mm_fp8q(y8: torch.Tensor, s: torch.Tensor, w: torch.Tensor,
+             bias: torch.Tensor | None = None) -> torch.Tensor:
     """GEMM on a producer-emitted e4m3 activation (uint8 bytes + its scale)
-    against a per-call-quantized e4m3 weight; bf16 out."""
+    against a per-call-quantized e4m3 weight; bias fused in the hipBLASLt
+    epilogue; bf16 out."""
     from jimm_amd.ops.hip_linear import _quant_e4m3
 
     w8, sw = _quant_e4m3(w)
     return torch._scaled_mm(
         y8.view(torch.float8_e4m3fn), w8.t(), scale_a=s.view(1, 1),
-        scale_b=sw.view(1, 1), out_dtype=torch.bfloat16,
+        scale_b=sw.view(1, 1), bias=bias, out_dtype=torch.bfloat16,
     )
 
 
@@ -102,7 +104,7 @@ class EncoderBlockFn(torch.autograd.Function):
             h1, h1q, mean1, rstd1 = ext.layernorm_fwd_fp8(
                 x, ln1w, ln1b, eps, scale8[0:1], amax8[0:1]
             )
-            qkv2 = ext.bias_act_fwd(_mm_fp8q(h1q, scale8[0:1], wqkv), bqkv, "", None)
+            qkv2 = _mm_fp8q(h1q, scale8[0:1], wqkv, bias=bqkv)
         else:
             h1, mean1, rstd1 = ext.layernorm_fwd(x, ln1w, ln1b, eps)
             if hip:
@@ -126,10 +128,9 @@ class EncoderBlockFn(torch.autograd.Function):
             h2, h2q, mean2, rstd2 = ext.layernorm_fwd_fp8(
                 a3, ln2w, ln2b, eps, scale8[1:2], amax8[1:2]
             )
-            z1 = _mm_fp8q(h2q, scale8[1:2], w1)
-            # z1 -> pre-act in place; f8 = e4m3(f) fused into the same pass
-            f, f8 = ext.bias_act_fwd_fp8(z1, b1, act, scale8[2:3], amax8[2:3])
-            y = ext.bias_act_fwd(_mm_fp8q(f8, scale8[2:3], w2), b2, "", a)
+            z1 = _mm_fp8q(h2q, scale8[1:2], w1, bias=b1)  # z1 = pre-act (bias fused)
+            f, f8 = ext.bias_act_fwd_fp8(z1, None, act, scale8[2:3], amax8[2:3])
+            y = ext.bias_act_fwd(_mm_fp8q(f8, scale8[2:3], w2, bias=b2), None, "", a)
             # next step's scales (delayed): scale = amax / 448, floor for
             # the first/degenerate steps
             scale8.copy_(torch.clamp(amax8 / 448.0, min=1e-12))
