@@ -35,8 +35,11 @@ def main():
     tun = torch.cuda.tunable
     tun.enable(True)
     tun.tuning_enable(True)
-    tun.set_max_tuning_duration_ms(500)
-    tun.set_max_tuning_iterations(200)
+    for name, val in (("set_max_tuning_duration", 500), ("set_max_tuning_iterations", 200)):
+        try:
+            getattr(tun, name)(val)
+        except AttributeError:
+            pass
 
     cases = []
     for n, k in SHAPES:
