@@ -32,7 +32,9 @@ def t_ms(fn, iters=30):
 
 
 def main():
+    os.makedirs("gpurun_out", exist_ok=True)
     tun = torch.cuda.tunable
+    tun.set_filename("gpurun_out/tunableop_fp8.csv")  # flushed at process exit
     tun.enable(True)
     tun.tuning_enable(True)
     for name, val in (("set_max_tuning_duration", 500), ("set_max_tuning_iterations", 200)):
@@ -56,8 +58,6 @@ def main():
         torch.cuda.synchronize()
         print(f"tuned ({n},{k}) bias={bias is not None}", flush=True)
 
-    os.makedirs("gpurun_out", exist_ok=True)
-    tun.write_file("gpurun_out/tunableop_fp8.csv")
     tun.tuning_enable(False)
 
     print(f"{'shape':16s} {'bias':5s} {'TF/s':>8s}")
