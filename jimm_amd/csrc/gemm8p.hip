@@ -45,6 +45,7 @@
 // N%256==0, K%64==0, K>=128 required (gemm8p_supported).
 
 #include <torch/extension.h>
+#include <hip/hip_fp8.h>
 #include <ATen/hip/HIPContext.h>
 
 #include "common.h"
@@ -94,11 +95,14 @@ __device__ __forceinline__ int frag_off(int row, int chunk) {
   return (row << 7) + ((chunk ^ (row & 7)) << 4);
 }
 
-template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool MGUARD, bool GRADM = false>
+template <int ACT, bool HAS_BIAS, bool HAS_RES, bool SAVE_PRE, bool MGUARD, bool GRADM = false,
+          bool FP8O = false>
 __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ W, const float* __restrict__ bias,
     const bf16* __restrict__ res, bf16* __restrict__ Y, bf16* __restrict__ Z,
-    int M, int N, int K) {
+    int M, int N, int K, unsigned char* __restrict__ Y8 = nullptr,
+    const float* __restrict__ scale8 = nullptr,
+    unsigned int* __restrict__ amax_bits = nullptr) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // A images: [buf 0..1][half 0..1] at (buf*2+half)*16 KiB;
   // B images: [slot 0..2][half 0..1] at 64 KiB + (slot*2+half)*16 KiB.
@@ -232,6 +236,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
 
   // Epilogue: acc[pr][mi][ni] -> rows m0 + pr*128 + wm*64 + 16mi + hi*4 + r,
   // cols n0 + wn*64 + 16ni + lo.
+  float rs8 = 1.f, tmax = 0.f;
+  if (FP8O) rs8 = 1.f / scale8[0];
 #pragma unroll
   for (int pr = 0; pr < 2; ++pr) {
 #pragma unroll
@@ -256,9 +262,19 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
             if (HAS_RES) vy += bf2f(res[(int64_t)m * N + n]);
           }
           Y[(int64_t)m * N + n] = f2bf(vy);
+          if (FP8O) {
+            // delayed-scaled e4m3 copy (consumed by the fp8 dX GEMM)
+            tmax = fmaxf(tmax, fabsf(vy));
+            Y8[(int64_t)m * N + n] =
+                (unsigned char)__hip_cvt_float_to_fp8(vy * rs8, __HIP_SATFINITE, __HIP_E4M3);
+          }
         }
       }
     }
+  }
+  if (FP8O) {
+    tmax = wave_reduce_max(tmax);
+    if (lane == 0) atomicMax(amax_bits, __float_as_uint(tmax));
   }
 }
 
@@ -300,7 +316,8 @@ torch::Tensor gemm_nt_8p_gradact(torch::Tensor dy, torch::Tensor wt, torch::Tens
                        reinterpret_cast<const bf16*>(dy.data_ptr()),                       \
                        reinterpret_cast<const bf16*>(wt.data_ptr()), nullptr,              \
                        reinterpret_cast<const bf16*>(z.data_ptr()),                        \
-                       reinterpret_cast<bf16*>(y.data_ptr()), nullptr, M, N, K);           \
+                       reinterpret_cast<bf16*>(y.data_ptr()), nullptr, M, N, K,            \
+                       nullptr, nullptr, nullptr);                                         \
   } while (0)
 #define DG(ACTC)                                                                           \
   do {                                                                                     \
@@ -315,6 +332,58 @@ torch::Tensor gemm_nt_8p_gradact(torch::Tensor dy, torch::Tensor wt, torch::Tens
 #undef DG
 #undef LAUNCH_GRAD
   return y;
+}
+
+std::vector<torch::Tensor> gemm_nt_8p_gradact_fp8(torch::Tensor dy, torch::Tensor wt,
+                                                  torch::Tensor z, std::string act,
+                                                  torch::Tensor scale8, torch::Tensor amax) {
+  // gemm_nt_8p_gradact + fused delayed-scaled e4m3 emission of dz (producer
+  // of the downstream fp8 dX GEMM); returns {dz bf16, dz8 e4m3 bytes}.
+  const int M = dy.size(0), K = dy.size(1), N = wt.size(0);
+  TORCH_CHECK(dy.is_contiguous() && wt.is_contiguous() && z.is_contiguous());
+  TORCH_CHECK(wt.size(1) == K && z.size(0) == M && z.size(1) == N);
+  TORCH_CHECK(gemm8p_supported(M, N, K));
+  int act_code = ACT_NONE;
+  if (act == "gelu") act_code = ACT_GELU;
+  else if (act == "gelu_tanh") act_code = ACT_GELU_TANH;
+  else if (act == "quickgelu") act_code = ACT_QUICKGELU;
+  else TORCH_CHECK(false, "gradact needs an activation, got ", act);
+  auto y = torch::empty({M, N}, dy.options());
+  auto y8 = torch::empty({M, N}, dy.options().dtype(torch::kUInt8));
+  auto stream = at::hip::getCurrentHIPStream();
+  const size_t shmem = 10 * HALF_BYTES;
+  const bool mguard = (M % BM) != 0;
+  const int mt = (M + BM - 1) / BM;
+#define LAUNCH_GRAD8(ACTC, MG)                                                             \
+  do {                                                                                     \
+    auto kfn = gemm_nt_8p_kernel<ACTC, false, true, false, MG, true, true>;                \
+    static bool attr_g8_##ACTC##MG = [&] {                                                 \
+      hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),                              \
+                          hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);         \
+      return true;                                                                         \
+    }();                                                                                   \
+    (void)attr_g8_##ACTC##MG;                                                              \
+    hipLaunchKernelGGL(kfn, dim3(mt * (N / BN)), dim3(NTHREADS), shmem, stream,            \
+                       reinterpret_cast<const bf16*>(dy.data_ptr()),                       \
+                       reinterpret_cast<const bf16*>(wt.data_ptr()), nullptr,              \
+                       reinterpret_cast<const bf16*>(z.data_ptr()),                        \
+                       reinterpret_cast<bf16*>(y.data_ptr()), nullptr, M, N, K,            \
+                       y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),             \
+                       reinterpret_cast<unsigned int*>(amax.data_ptr()));                  \
+  } while (0)
+#define DG8(ACTC)                                                                          \
+  do {                                                                                     \
+    if (mguard) LAUNCH_GRAD8(ACTC, true);                                                  \
+    else LAUNCH_GRAD8(ACTC, false);                                                        \
+  } while (0)
+  switch (act_code) {
+    case ACT_GELU: DG8(ACT_GELU); break;
+    case ACT_GELU_TANH: DG8(ACT_GELU_TANH); break;
+    case ACT_QUICKGELU: DG8(ACT_QUICKGELU); break;
+  }
+#undef DG8
+#undef LAUNCH_GRAD8
+  return {y, y8};
 }
 
 void gemm_nt_8p(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias_f32,
@@ -347,7 +416,8 @@ void gemm_nt_8p(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> b
     hipLaunchKernelGGL(kfn, dim3(mt * (N / BN)), dim3(NTHREADS), shmem, stream,            \
                        reinterpret_cast<const bf16*>(x.data_ptr()),                        \
                        reinterpret_cast<const bf16*>(w.data_ptr()), biasp, resp,           \
-                       reinterpret_cast<bf16*>(y.data_ptr()), zp, M, N, K);                \
+                       reinterpret_cast<bf16*>(y.data_ptr()), zp, M, N, K,                 \
+                       nullptr, nullptr, nullptr);                                         \
   } while (0)
 #define DISPATCH_MG(ACTC, HB, HR, SP)                                                      \
   do {                                                                                     \

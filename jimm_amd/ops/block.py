@@ -35,17 +35,27 @@ def _hip_gemms(M: int, H: int) -> bool:
 
 
 def _fp8_block_state(mod: torch.nn.Module, device: torch.device):
-    """Per-block delayed-scaling state (3 activation sites: h1->qkv,
-    h2->fc1, f->fc2): scale8 used by this step's producers, amax collected
-    this step, scale8 <- amax/448 at block end (next step's scale)."""
+    """Per-block delayed-scaling state, 5 sites: [0] h1->qkv, [1] h2->fc1,
+    [2] f->fc2 (forward), [3] dz1->fc1-dX, [4] dqkv->qkv-dX (backward).
+    scale8 is what this step's producers divide by; amax collected this step
+    becomes next step's scale (scale8[0:3] updated at forward end, [3:5] at
+    backward end). Grad sites start at 1/448 (assume amax~1; adapts in one
+    step) so first-step gradients are not flushed to zero."""
     st = getattr(mod, "_fp8_state", None)
     if st is None or st[0].device != device:
-        st = (
-            torch.ones(3, device=device, dtype=torch.float32),
-            torch.zeros(3, device=device, dtype=torch.float32),
-        )
+        scale = torch.ones(5, device=device, dtype=torch.float32)
+        scale[3:5] = 1.0 / 448.0
+        st = (scale, torch.zeros(5, device=device, dtype=torch.float32))
         mod._fp8_state = st
     return st
+
+
+def _quant_e4m3_t(w: torch.Tensor):
+    """Quantize W^T to e4m3 (row-major (K, N) bytes): the column-major b
+    operand a scaled_mm dX GEMM (dz @ W) needs is its .t() view."""
+    s = (w.abs().amax().float() / 448.0).clamp(min=1e-12)
+    w8t = (w.t() * s.reciprocal().to(w.dtype)).to(torch.float8_e4m3fn)
+    return w8t, s
 
 
 def _mm_fp8q(y8: torch.Tensor, s: torch.Tensor, w: torch.Tensor,
@@ -131,9 +141,9 @@ class EncoderBlockFn(torch.autograd.Function):
             z1 = _mm_fp8q(h2q, scale8[1:2], w1, bias=b1)  # z1 = pre-act (bias fused)
             f, f8 = ext.bias_act_fwd_fp8(z1, None, act, scale8[2:3], amax8[2:3])
             y = ext.bias_act_fwd(_mm_fp8q(f8, scale8[2:3], w2, bias=b2), None, "", a)
-            # next step's scales (delayed): scale = amax / 448, floor for
-            # the first/degenerate steps
-            scale8.copy_(torch.clamp(amax8 / 448.0, min=1e-12))
+            # next step's forward scales (delayed); grad sites [3:5] are
+            # updated at backward end, after their amax is collected
+            scale8[0:3].copy_(torch.clamp(amax8[0:3] / 448.0, min=1e-12))
         elif hip:
             a, _ = ext.linear_fwd(o2, wproj, bproj, "", x2, False)
             a3 = a.view(B, L, H)
@@ -153,6 +163,7 @@ class EncoderBlockFn(torch.autograd.Function):
                               h1, qkv, o, lse, a, mean1, rstd1, mean2, rstd2, h2, z1, f)
         ctx.dims = (B, L, H, num_heads)
         ctx.meta = (act, causal, scale)
+        ctx.fp8_state = (scale8, amax8) if fp8 else None
         return y.view(B, L, H)
 
     @staticmethod
@@ -166,9 +177,18 @@ class EncoderBlockFn(torch.autograd.Function):
         h1_2 = h1.view(-1, H)
         h2_2 = h2.view(-1, H)
         hip = _hip_gemms(dy2.shape[0], H)
+        fp8 = ctx.fp8_state is not None and hip
+        if fp8:
+            scale8, amax8 = ctx.fp8_state
 
         # MLP fc2 (+residual into a)
-        if hip:
+        if fp8:
+            # gradact dX with fused e4m3 emission of dz1; the fc1-dX GEMM
+            # (dh2 = dz1 @ W1) then runs on the fp8 MFMA pipe
+            dz1, dz18 = ext.gemm_nt_8p_gradact_fp8(
+                dy2, w2.t().contiguous(), z1, act, scale8[3:4], amax8[3:4]
+            )
+        elif hip:
             # dX of fc2 with the activation backward fused into the GEMM
             # epilogue: dz1 = (dy @ W2) * act'(z1) — no separate act_bwd pass
             dz1 = ext.gemm_nt_8p_gradact(dy2, w2.t().contiguous(), z1, act)
@@ -177,7 +197,14 @@ class EncoderBlockFn(torch.autograd.Function):
             dz1 = ext.act_bwd(df, z1, act)
         dw2 = _dw_gemm(ext, dy2, f, w2.dtype)
         db2 = _colsum(ext, dy2)
-        dh2 = _dx_gemm(ext, dz1, w1)
+        if fp8:
+            w1t8, sw1 = _quant_e4m3_t(w1)
+            dh2 = torch._scaled_mm(
+                dz18.view(torch.float8_e4m3fn), w1t8.t(), scale_a=scale8[3:4].view(1, 1),
+                scale_b=sw1.view(1, 1), out_dtype=torch.bfloat16,
+            )
+        else:
+            dh2 = _dx_gemm(ext, dz1, w1)
         dw1 = _dw_gemm(ext, dz1, h2_2, w1.dtype)
         db1 = _colsum(ext, dz1)
         # LN2 backward with the MLP residual grad (dy) fused into dx
@@ -205,13 +232,25 @@ class EncoderBlockFn(torch.autograd.Function):
         dqkv2 = dqkv.view(-1, 3 * H)
 
         # QKV projection
-        dh1 = _dx_gemm(ext, dqkv2, wqkv)
+        if fp8:
+            dqkv8 = ext.fp8_cast(dqkv2, scale8[4:5], amax8[4:5])
+            wqkvt8, swq = _quant_e4m3_t(wqkv)
+            dh1 = torch._scaled_mm(
+                dqkv8.view(torch.float8_e4m3fn), wqkvt8.t(),
+                scale_a=scale8[4:5].view(1, 1), scale_b=swq.view(1, 1),
+                out_dtype=torch.bfloat16,
+            )
+        else:
+            dh1 = _dx_gemm(ext, dqkv2, wqkv)
         dwqkv = _dw_gemm(ext, dqkv2, h1_2, wqkv.dtype)
         dbqkv = _colsum(ext, dqkv2)
         # LN1 backward with the attention residual grad (da) fused into dx
         dx, dln1w, dln1b = ext.layernorm_bwd(
             dh1.view(B, L, H), x, ln1w, mean1, rstd1, da3
         )
+        if fp8:
+            # next step's backward scales (delayed)
+            scale8[3:5].copy_(torch.clamp(amax8[3:5] / 448.0, min=1e-12))
         return (dx, dln1w, dln1b, dwqkv, dbqkv, dwproj, dbproj, dln2w, dln2b,
                 dw1, db1, dw2, db2, None, None, None, None, None, None, None)
 

@@ -601,7 +601,8 @@ def test_fp8_fused_block():
     yb, dxb, gb = run(False)
     assert torch.isfinite(y8).all() and torch.isfinite(dx8).all()
     assert rel_err(y8, yb) < 0.06, rel_err(y8, yb)
-    assert rel_err(dx8, dxb) < 0.12, rel_err(dx8, dxb)
+    # dx passes through the fp8 dX GEMMs (dz1, dqkv quantized e4m3)
+    assert rel_err(dx8, dxb) < 0.2, rel_err(dx8, dxb)
     for n in g8:
         assert torch.isfinite(g8[n]).all(), n
 
