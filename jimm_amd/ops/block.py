@@ -54,8 +54,8 @@ def _quant_e4m3_t(w: torch.Tensor):
     """Quantize W^T to e4m3 (row-major (K, N) bytes): the column-major b
     operand a scaled_mm dX GEMM (dz @ W) needs is its .t() view."""
     s = (w.abs().amax().float() / 448.0).clamp(min=1e-12)
-    w8t = (w.t() * s.reciprocal().to(w.dtype)).to(torch.float8_e4m3fn)
-    return w8t, s
+    w8 = (w * s.reciprocal().to(w.dtype)).to(torch.float8_e4m3fn)  # (N, K) row-major
+    return w8.t().contiguous(), s  # (K, N); .t() at the call site is column-major
 
 
 def _mm_fp8q(y8: torch.Tensor, s: torch.Tensor, w: torch.Tensor,
