@@ -95,6 +95,21 @@ __device__ __forceinline__ float act_fwd(float x, int act) {
   }
 }
 
+// e4m3 (OCP) converts on the hardware v_cvt_pk_fp8_f32 pipe. The HIP
+// library __hip_cvt_float_to_fp8 is a ~50-instruction software emulation
+// (measured +77 us on a (73856,1024) LayerNorm); the packed builtin is one
+// VALU op per 2 elements. Saturate-to-finite by clamping to +-448 first
+// (NaN clamps to 448 via IEEE minNum/maxNum; inputs are never NaN here).
+__device__ __forceinline__ unsigned short cvt2_e4m3(float a, float b) {
+  a = fminf(fmaxf(a, -448.f), 448.f);
+  b = fminf(fmaxf(b, -448.f), 448.f);
+  return (unsigned short)(__builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false) & 0xffff);
+}
+
+__device__ __forceinline__ unsigned char cvt_e4m3(float a) {
+  return (unsigned char)(cvt2_e4m3(a, 0.f) & 0xff);
+}
+
 __device__ __forceinline__ float act_grad(float x, int act) {
   switch (act) {
     case ACT_GELU: {

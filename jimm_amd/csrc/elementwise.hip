@@ -65,12 +65,11 @@ __global__ void bias_act_vec_kernel(T* __restrict__ z, const T* __restrict__ bia
     }
     vstore_f32<V>(y + i * V, o);
     if (FP8O) {
-      unsigned char q8[V];
+      unsigned short q8[V / 2];
 #pragma unroll
-      for (int j = 0; j < V; ++j) {
-        tmax = fmaxf(tmax, fabsf(o[j]));
-        q8[j] = (unsigned char)__hip_cvt_float_to_fp8(o[j] * rs8, __HIP_SATFINITE, __HIP_E4M3);
-      }
+      for (int j = 0; j < V; ++j) tmax = fmaxf(tmax, fabsf(o[j]));
+#pragma unroll
+      for (int j = 0; j < V; j += 2) q8[j / 2] = cvt2_e4m3(o[j] * rs8, o[j + 1] * rs8);
       *reinterpret_cast<uint2*>(y8 + i * V) = *reinterpret_cast<uint2*>(q8);
     }
   }
@@ -92,12 +91,11 @@ __global__ void fp8_cast_kernel(const bf16* __restrict__ x, unsigned char* __res
        i += (int64_t)gridDim.x * blockDim.x) {
     float xv[8];
     vload_f32<8>(x + i * 8, xv);
-    unsigned char q8[8];
+    unsigned short q8[4];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      tmax = fmaxf(tmax, fabsf(xv[j]));
-      q8[j] = (unsigned char)__hip_cvt_float_to_fp8(xv[j] * rs8, __HIP_SATFINITE, __HIP_E4M3);
-    }
+    for (int j = 0; j < 8; ++j) tmax = fmaxf(tmax, fabsf(xv[j]));
+#pragma unroll
+    for (int j = 0; j < 8; j += 2) q8[j / 2] = cvt2_e4m3(xv[j] * rs8, xv[j + 1] * rs8);
     *reinterpret_cast<uint2*>(y8 + i * 8) = *reinterpret_cast<uint2*>(q8);
   }
   tmax = wave_reduce_max(tmax);

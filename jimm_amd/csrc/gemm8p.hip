@@ -265,8 +265,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_8p_kernel(
           if (FP8O) {
             // delayed-scaled e4m3 copy (consumed by the fp8 dX GEMM)
             tmax = fmaxf(tmax, fabsf(vy));
-            Y8[(int64_t)m * N + n] =
-                (unsigned char)__hip_cvt_float_to_fp8(vy * rs8, __HIP_SATFINITE, __HIP_E4M3);
+            Y8[(int64_t)m * N + n] = cvt_e4m3(vy * rs8);
           }
         }
       }

@@ -325,9 +325,14 @@ torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x) {
     const char* e = getenv("JIMM_AMD_DW_SPLITM");
     return e ? atoi(e) : 0;
   }();
+  // r02 sweep at M=128*577: small-tile shapes (proj 1024x1024 -> 16 tiles)
+  // are atomic-bound and want ~208 total WGs (430 -> 650 TF/s); larger
+  // shapes peak around <= 20 splits (qkv 48 tiles: 27 -> 20 gave +6%).
+  int heur = tiles <= 16 ? (208 + tiles - 1) / tiles
+                         : (int)std::min<int64_t>((1280 + tiles - 1) / tiles, 20);
   int splitm = splitm_env > 0
                    ? splitm_env
-                   : (int)std::min<int64_t>((1280 + tiles - 1) / tiles, (M + BMS - 1) / BMS);
+                   : (int)std::min<int64_t>(heur, (M + BMS - 1) / BMS);
   if ((int64_t)splitm > (M + BMS - 1) / BMS) splitm = (int)((M + BMS - 1) / BMS);
   if (splitm < 1) splitm = 1;
   int64_t chunk = ((M + splitm - 1) / splitm + BMS - 1) / BMS * BMS;

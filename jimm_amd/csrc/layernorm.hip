@@ -70,16 +70,16 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__
       for (int k = 0; k < VEC; ++k) yv[k] = (xv[k] - mean) * rstd * wv[k] + bv[k];
       vstore_f32<VEC>(yr + i, yv);
       if (FP8O) {
-        unsigned char q8[VEC];
+        unsigned short q8[(VEC + 1) / 2];
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) {
-          tmax = fmaxf(tmax, fabsf(yv[k]));
-          q8[k] = (unsigned char)__hip_cvt_float_to_fp8(yv[k] * rs8, __HIP_SATFINITE, __HIP_E4M3);
-        }
+        for (int k = 0; k < VEC; ++k) tmax = fmaxf(tmax, fabsf(yv[k]));
+#pragma unroll
+        for (int k = 0; k + 1 < VEC; k += 2)
+          q8[k / 2] = cvt2_e4m3(yv[k] * rs8, yv[k + 1] * rs8);
         if (VEC == 8) *reinterpret_cast<uint2*>(y8 + row * H + i) = *reinterpret_cast<uint2*>(q8);
         else {
 #pragma unroll
-          for (int k = 0; k < VEC; ++k) y8[row * H + i + k] = q8[k];
+          for (int k = 0; k < VEC; ++k) y8[row * H + i + k] = cvt_e4m3(yv[k] * rs8);
         }
       }
     }
