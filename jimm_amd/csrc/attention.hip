@@ -365,7 +365,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // ---------------------------------------------------------------------------
 
 template <bool CAUSAL>
-__global__ __launch_bounds__(256) void attn_fwd_small_kernel(
+__global__ __launch_bounds__(320) void attn_fwd_small_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     bf16* __restrict__ o, float* __restrict__ lse, int Lq, int Lk, float scale, int H,
     int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
@@ -376,20 +376,21 @@ __global__ __launch_bounds__(256) void attn_fwd_small_kernel(
   constexpr int PPITCH = 96;  // P-tile pitch: the PV A-frag reads key
                               // columns up to 95 (s=2 covers keys 64..95),
                               // so the row must span the full padded range
-  // K lives entirely in REGISTERS (5 kt x 2 s fragments = 40 VGPR/lane);
-  // only the V block image + the P tile are in LDS: ~15 KiB per wave ->
-  // two 4-wave workgroups per CU (the v1 with a K image was 1 wave/SIMD
-  // and measured SLOWER than the general kernel).
-  constexpr int WSLICE = LP * D + 16 * PPITCH;  // shorts per wave
+  // One 5-wave workgroup per (b, h): wave w owns q-strip w (<= 5 strips at
+  // L <= 80), the V block image is staged ONCE and shared (one barrier).
+  // K lives in REGISTERS per wave (5 kt x 2 s fragments = 40 VGPR/lane).
+  // The previous wave-per-(b,h) version looped the 5 strips serially in
+  // one wave and topped out at 2 waves/SIMD — latency-exposed (68.8 TF/s
+  // at CLIP-text L=77); strip-per-wave runs 5x the waves at the same LDS
+  // footprint per WG (12 KiB V + 5 x 3 KiB P = 27 KiB).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int lo = lane & 15, hi = lane >> 4;
-  short* vb = reinterpret_cast<short*>(smem) + wave * WSLICE;
-  short* pt = vb + LP * D;
+  short* vb = reinterpret_cast<short*>(smem);
+  short* pt = vb + LP * D + wave * 16 * PPITCH;
 
-  const int64_t bh = (int64_t)blockIdx.x * 4 + wave;
-  if (bh >= BH) return;  // wave-uniform; no barriers anywhere in this kernel
+  const int64_t bh = blockIdx.x;
   // zero the P tile once: the PV step reads key columns beyond Lk whose P
   // is never written (their V rows are zero-padded, but garbage LDS can
   // hold NaN bits and NaN * 0 = NaN — seen as an intermittent test failure)
@@ -402,38 +403,42 @@ __global__ __launch_bounds__(256) void attn_fwd_small_kernel(
   const bf16* vp = v + b * v_sb + h * v_sh;
   bf16* op = o + b * o_sb + h * o_sh;
 
-  // ---- stage V into the wave's block image (zero-padded) -----------------
-#pragma unroll
-  for (int r = 0; r < LP * D / (WAVE * 8); ++r) {
-    const int idx = r * WAVE + lane;  // 8-short chunk: (key, d-oct)
-    const int key = idx >> 3;
-    const int d8 = (idx & 7) * 8;
-    bf16x8_t vv_{};
-    if (key < Lk) vv_ = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + d8);
-    *reinterpret_cast<bf16x8_t*>(vb + boff_f<D>(key, d8)) = vv_;
+  // ---- stage V into the workgroup's shared block image (zero-padded) -----
+  {
+    const int t = (int)threadIdx.x;
+    for (int idx = t; idx < LP * D / 8; idx += 320) {  // 8-short chunk: (key, d-oct)
+      const int key = idx >> 3;
+      const int d8 = (idx & 7) * 8;
+      bf16x8_t vv_{};
+      if (key < Lk) vv_ = *reinterpret_cast<const bf16x8_t*>(vp + (int64_t)key * v_sl + d8);
+      *reinterpret_cast<bf16x8_t*>(vb + boff_f<D>(key, d8)) = vv_;
+    }
   }
-  // ---- K fragments straight from global into registers -------------------
-  bf16x8_t kfr[5][2];
-#pragma unroll
-  for (int kt = 0; kt < 5; ++kt) {
-    const int key = 16 * kt + lo;
-#pragma unroll
-    for (int s = 0; s < 2; ++s)
-      kfr[kt][s] = key < Lk
-                       ? *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + 32 * s + hi * 8)
-                       : bf16x8_t{};
-  }
-
   const int nq = (Lq + 15) / 16;
   const int nkt = (Lk + 15) / 16;
-  for (int qs = 0; qs < nq; ++qs) {
+  const int qs = wave;
+  __syncthreads();  // V image visible to all waves
+  if (qs < nq) {
     const int q0 = qs * 16;
+    // ---- K fragments straight from global into registers (causal strips
+    // stop at their diagonal tile) -----------------------------------------
+    const int ktmax0 = CAUSAL ? min(nkt, qs + 1) : nkt;
+    bf16x8_t kfr[5][2];
+#pragma unroll
+    for (int kt = 0; kt < 5; ++kt) {
+      const int key = 16 * kt + lo;
+#pragma unroll
+      for (int s = 0; s < 2; ++s)
+        kfr[kt][s] = (kt < ktmax0 && key < Lk)
+                         ? *reinterpret_cast<const bf16x8_t*>(kp + (int64_t)key * k_sl + 32 * s + hi * 8)
+                         : bf16x8_t{};
+    }
     const int qrow = min(q0 + lo, Lq - 1);
     const bf16x8_t qb0 = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + hi * 8);
     const bf16x8_t qb1 = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 + hi * 8);
 
     // ---- S^T = K . Q^T over <= 5 16-key tiles ----------------------------
-    const int ktmax = CAUSAL ? min(nkt, qs + 1) : nkt;
+    const int ktmax = ktmax0;
     f32x4_t sc[5] = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -561,8 +566,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   // small-L fast path: one wave per (b,h), no barriers (L <= 80, D = 64)
   if (Dr == 64 && Lq == Lk && Lk <= 80) {
     const int64_t BH = (int64_t)B * H;
-    const dim3 sgrid((unsigned)((BH + 3) / 4));
-    const size_t sshmem = 4 * (96 * 64 + 16 * 96) * sizeof(short);
+    const dim3 sgrid((unsigned)BH);  // one 5-wave WG per (b, h)
+    const size_t sshmem = (96 * 64 + 5 * 16 * 96) * sizeof(short);
 #define SMALL_ARGS                                                                         \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                          \
@@ -573,10 +578,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      v.stride(0), v.stride(1), v.stride(2),                                \
                      o.stride(0), o.stride(1), o.stride(2), BH
     if (causal)
-      hipLaunchKernelGGL((attn_fwd_small_kernel<true>), sgrid, dim3(256), sshmem, stream,
+      hipLaunchKernelGGL((attn_fwd_small_kernel<true>), sgrid, dim3(320), sshmem, stream,
                          SMALL_ARGS);
     else
-      hipLaunchKernelGGL((attn_fwd_small_kernel<false>), sgrid, dim3(256), sshmem, stream,
+      hipLaunchKernelGGL((attn_fwd_small_kernel<false>), sgrid, dim3(320), sshmem, stream,
                          SMALL_ARGS);
 #undef SMALL_ARGS
     return {o, lse};
