@@ -372,10 +372,18 @@ __global__ __launch_bounds__(320) void attn_fwd_small_kernel(
     int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t o_sb, int64_t o_sh, int64_t o_sl,
     int64_t BH) {
   constexpr int D = 64;
-  constexpr int LP = 96;      // padded key count (block images need 32-multiples)
-  constexpr int PPITCH = 96;  // P-tile pitch: the PV A-frag reads key
-                              // columns up to 95 (s=2 covers keys 64..95),
-                              // so the row must span the full padded range
+  constexpr int LP = 96;       // padded key count (block images need 32-multiples)
+  constexpr int PPITCH = 88;   // P-tile pitch. The PV A-frag reads key columns
+                               // up to 95 (s=2 covers keys 64..95); cols >= 88
+                               // alias the NEXT row's written (finite) P values,
+                               // which is benign: they multiply V image rows
+                               // 88..95, zero-padded. 88 beats 96 twice: the
+                               // P-tile write stride (44 words) is 2-way bank
+                               // conflicted vs 8-way at 96, and the smaller
+                               // footprint fits 6 WGs/CU instead of 5.
+  constexpr int PSTRIDE = 16 * PPITCH + 8;  // +8 zeroed shorts: the last row's
+                                            // col-95 over-read stays in zeroed
+                                            // LDS even for the last wave
   // One 5-wave workgroup per (b, h): wave w owns q-strip w (<= 5 strips at
   // L <= 80), the V block image is staged ONCE and shared (one barrier).
   // K lives in REGISTERS per wave (5 kt x 2 s fragments = 40 VGPR/lane).
@@ -388,15 +396,17 @@ __global__ __launch_bounds__(320) void attn_fwd_small_kernel(
   const int lane = threadIdx.x % WAVE;
   const int lo = lane & 15, hi = lane >> 4;
   short* vb = reinterpret_cast<short*>(smem);
-  short* pt = vb + LP * D + wave * 16 * PPITCH;
+  short* pt = vb + LP * D + wave * PSTRIDE;
 
   const int64_t bh = blockIdx.x;
   // zero the P tile once: the PV step reads key columns beyond Lk whose P
   // is never written (their V rows are zero-padded, but garbage LDS can
   // hold NaN bits and NaN * 0 = NaN — seen as an intermittent test failure)
 #pragma unroll
-  for (int i = 0; i < 16 * PPITCH / (WAVE * 8); ++i)
-    *reinterpret_cast<bf16x8_t*>(pt + (i * WAVE + lane) * 8) = bf16x8_t{};
+  for (int i = 0; i < (PSTRIDE / 8 + WAVE - 1) / WAVE; ++i) {
+    const int idx = i * WAVE + lane;
+    if (idx < PSTRIDE / 8) *reinterpret_cast<bf16x8_t*>(pt + idx * 8) = bf16x8_t{};
+  }
   const int64_t b = bh / H, h = bh % H;
   const bf16* qp = q + b * q_sb + h * q_sh;
   const bf16* kp = k + b * k_sb + h * k_sh;
@@ -567,7 +577,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   if (Dr == 64 && Lq == Lk && Lk <= 80) {
     const int64_t BH = (int64_t)B * H;
     const dim3 sgrid((unsigned)BH);  // one 5-wave WG per (b, h)
-    const size_t sshmem = (96 * 64 + 5 * 16 * 96) * sizeof(short);
+    const size_t sshmem = (96 * 64 + 5 * (16 * 88 + 8)) * sizeof(short);
 #define SMALL_ARGS                                                                         \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                          \
