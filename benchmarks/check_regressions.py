@@ -20,22 +20,22 @@ from jimm_amd.ops._backend import maybe_enable_tunableop
 # round-2 recorded values (TF/s), with a ~10% regression margin applied
 # (r02: block-image attention + small-L kernel + in-house GEMM default)
 FLOORS_ATTN = {  # (B, H, L, causal): (fwd_tf, bwd_tf)
-    (256, 12, 197, False): (145, 145),
-    (64, 16, 577, False): (255, 220),
-    (256, 8, 77, True): (60, 42),
-    (256, 12, 257, False): (185, 155),
+    (256, 12, 197, False): (150, 155),
+    (64, 16, 577, False): (265, 235),
+    (256, 8, 77, True): (72, 44),     # strip-per-wave small-L kernel (76.4)
+    (256, 12, 257, False): (195, 165),
 }
-FLOOR_STEP_VIT_B1024 = 5250  # img/s (r02: 5430 with in-house GEMM default)
+FLOOR_STEP_VIT_B1024 = 5350  # img/s (r02 final: 5486)
 # in-house GEMM floors on the b1024 model shapes (median TF/s, -10%)
 FLOORS_GEMM = {  # (M, N, K, act): fwd_tf
     (201728, 2304, 768, ""): 790,
     (201728, 3072, 768, "gelu"): 600,
     (201728, 768, 3072, ""): 890,
 }
-FLOORS_DW = {  # (M, N, K): tf
-    (201728, 2304, 768): 620,
-    (201728, 3072, 768): 650,
-    (201728, 768, 3072): 650,
+FLOORS_DW = {  # (M, N, K): tf (post splitm-heuristic rework)
+    (201728, 2304, 768): 630,
+    (201728, 3072, 768): 660,
+    (201728, 768, 3072): 660,
 }
 
 
