@@ -635,6 +635,213 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Small-L fused backward (Lq == Lk <= 80, D = 64): ONE kernel replaces the
+// d2 + dkv + dq trio, whose 64-wide tiles waste 1.7x padding at L = 77
+// (CLIP text). One 5-wave workgroup per (b, h):
+//   phase 1 (wave = 16-query strip): S^T = K.Q^T and dP^T = V.dO^T with K/V
+//     in REGISTERS (40 VGPR each) and Q/dO rows read direct from global;
+//     P = exp(scale*S - lse) (no max pass -- lse saved by the forward);
+//     delta = rowsum(P o dP) (== rowsum(dO o O), so no O re-read);
+//     dS = P o (dP - delta) * scale; the strip's dQ = dS @ K via tr-read
+//     B-fragments of a shared K block image.
+//   phase 2 (wave = 16-key strip): dK = dS^T @ Q and dV = P^T @ dO from
+//     row-major P^T / dS^T LDS tiles (written in-register-order by phase 1)
+//     against tr-read fragments of shared Q / dO block images (the K image
+//     slot is restaged with dO between phases).
+// Tile pitch 88 < 96: over-reads past col 87 alias the next row's finite
+// values and multiply zero-padded image rows (see attn_fwd_small_kernel);
+// tiles are zero-initialized once so causal-masked entries (never written)
+// contribute exact zeros. LDS: 2 images + 3 tiles + pad = 65.3 KiB -> 2
+// workgroups per CU.
+// ---------------------------------------------------------------------------
+
+template <bool CAUSAL>
+__global__ __launch_bounds__(320) void attn_bwd_small_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const bf16* __restrict__ dO, const float* __restrict__ lse, bf16* __restrict__ dq,
+    bf16* __restrict__ dk, bf16* __restrict__ dv, int Lq, int Lk, float scale, int H,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl, int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl, int64_t do_sb, int64_t do_sh, int64_t do_sl,
+    int64_t dq_sb, int64_t dq_sh, int64_t dq_sl, int64_t dk_sb, int64_t dk_sh,
+    int64_t dk_sl, int64_t dv_sb, int64_t dv_sh, int64_t dv_sl) {
+  constexpr int D = 64;
+  constexpr int LP = 96;         // image rows (three 32-row blocks)
+  constexpr int TP = 88;         // tile pitch
+  constexpr int IMG = LP * D;    // shorts per image
+  constexpr int TILE = 80 * TP;  // shorts per tile
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* imgA = reinterpret_cast<short*>(smem);  // K image (ph1) -> dO image (ph2)
+  short* imgQ = imgA + IMG;                      // Q image (ph2)
+  short* PT = imgQ + IMG;                        // P^T  [key][q]
+  short* DST = PT + TILE;                        // dS^T [key][q]
+  short* DSQ = DST + TILE;                       // dS   [q][key] (+8 zeroed pad after)
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int lo = lane & 15, hi = lane >> 4;
+  const int64_t bh = blockIdx.x;
+  const int64_t b = bh / H, h = bh % H;
+  const bf16* qp = q + b * q_sb + h * q_sh;
+  const bf16* kp = k + b * k_sb + h * k_sh;
+  const bf16* vp = v + b * v_sb + h * v_sh;
+  const bf16* dop = dO + b * do_sb + h * do_sh;
+  bf16* dqp = dq + b * dq_sb + h * dq_sh;
+  bf16* dkp = dk + b * dk_sb + h * dk_sh;
+  bf16* dvp = dv + b * dv_sb + h * dv_sh;
+
+  // zero the tiles + pad once (masked entries are never written)
+  for (int i = threadIdx.x; i < (3 * TILE + 8) / 8; i += 320)
+    *reinterpret_cast<bf16x8_t*>(PT + i * 8) = bf16x8_t{};
+  auto stage = [&](short* img, const bf16* src, int64_t sl, int rows) {
+    for (int idx = (int)threadIdx.x; idx < IMG / 8; idx += 320) {
+      const int row = idx >> 3, d8 = (idx & 7) * 8;
+      bf16x8_t vv{};
+      if (row < rows) vv = *reinterpret_cast<const bf16x8_t*>(src + (int64_t)row * sl + d8);
+      *reinterpret_cast<bf16x8_t*>(img + boff<D>(row, d8)) = vv;
+    }
+  };
+  stage(imgA, kp, k_sl, Lk);
+  stage(imgQ, qp, q_sl, Lq);
+  const int nq = (Lq + 15) / 16;
+  const int nkt = (Lk + 15) / 16;
+  __syncthreads();
+
+  // ---- phase 1: per q-strip --------------------------------------------
+  if (wave < nq) {
+    const int qs = wave;
+    const int q0 = 16 * qs;
+    const int ktmax = CAUSAL ? min(nkt, qs + 1) : nkt;
+    bf16x8_t kfr[5][2], vfr[5][2];
+#pragma unroll
+    for (int kt = 0; kt < 5; ++kt) {
+      const int key = 16 * kt + lo;
+      const bool ok = kt < ktmax && key < Lk;
+#pragma unroll
+      for (int sj = 0; sj < 2; ++sj) {
+        kfr[kt][sj] = ok ? *reinterpret_cast<const bf16x8_t*>(
+                               kp + (int64_t)key * k_sl + 32 * sj + hi * 8)
+                         : bf16x8_t{};
+        vfr[kt][sj] = ok ? *reinterpret_cast<const bf16x8_t*>(
+                               vp + (int64_t)key * v_sl + 32 * sj + hi * 8)
+                         : bf16x8_t{};
+      }
+    }
+    const int qrow = min(q0 + lo, Lq - 1);
+    const bf16x8_t qb0 = *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + hi * 8);
+    const bf16x8_t qb1 =
+        *reinterpret_cast<const bf16x8_t*>(qp + (int64_t)qrow * q_sl + 32 + hi * 8);
+    const bf16x8_t db0 =
+        *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qrow * do_sl + hi * 8);
+    const bf16x8_t db1 =
+        *reinterpret_cast<const bf16x8_t*>(dop + (int64_t)qrow * do_sl + 32 + hi * 8);
+    f32x4_t sc[5] = {}, dpc[5] = {};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kt = 0; kt < 5; ++kt) {
+      if (kt >= ktmax) continue;
+      sc[kt] = MFMA16(kfr[kt][0], qb0, sc[kt]);
+      sc[kt] = MFMA16(kfr[kt][1], qb1, sc[kt]);
+      dpc[kt] = MFMA16(vfr[kt][0], db0, dpc[kt]);
+      dpc[kt] = MFMA16(vfr[kt][1], db1, dpc[kt]);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    const int q_idx = q0 + lo;
+    const float lse_q = lse[bh * Lq + qrow];
+    float pv[20];
+    float dsum = 0.f;
+#pragma unroll
+    for (int kt = 0; kt < 5; ++kt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = 16 * kt + 4 * hi + r;
+        const bool ok =
+            kt < ktmax && key < Lk && q_idx < Lq && (!CAUSAL || key <= q_idx);
+        const float p = ok ? __expf(sc[kt][r] * scale - lse_q) : 0.f;
+        pv[4 * kt + r] = p;
+        dsum += p * dpc[kt][r];
+      }
+    }
+    dsum += __shfl_xor(dsum, 16, WAVE);
+    dsum += __shfl_xor(dsum, 32, WAVE);  // delta[q] == rowsum(dO o O)
+#pragma unroll
+    for (int kt = 0; kt < 5; ++kt) {
+      bf16x4_tr ds4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = 16 * kt + 4 * hi + r;
+        const float p = pv[4 * kt + r];
+        const float ds = p * (dpc[kt][r] - dsum) * scale;
+        PT[key * TP + q_idx] = f2bfs(p);
+        DST[key * TP + q_idx] = f2bfs(ds);
+        ds4[r] = f2bfs(ds);
+      }
+      *reinterpret_cast<bf16x4_tr*>(DSQ + q_idx * TP + 16 * kt + 4 * hi) = ds4;
+    }
+    // dQ strip = dS @ K (B = tr fragments of the K image)
+    f32x4_t adq[4] = {};
+    const int smax = (16 * ktmax + 31) / 32;
+#pragma unroll
+    for (int sb = 0; sb < 3; ++sb) {
+      if (sb >= smax) continue;
+      bf16x8_t bfr[4];
+      tr_frag_x4((lds_cp)(const void*)(imgA + sb * 32 * D) + lane * 8, bfr);
+      const bf16x8_t pa =
+          *reinterpret_cast<const bf16x8_t*>(DSQ + (q0 + lo) * TP + 32 * sb + hi * 8);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) adq[dt] = MFMA16(pa, bfr[dt], adq[dt]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qr = q0 + 4 * hi + r;
+      if (qr >= Lq) continue;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+        dqp[(int64_t)qr * dq_sl + 16 * dt + lo] = f2bf(adq[dt][r]);
+    }
+  }
+  __syncthreads();
+  stage(imgA, dop, do_sl, Lq);  // dO image replaces K (phase-1 readers done)
+  __syncthreads();
+
+  // ---- phase 2: per key-strip ------------------------------------------
+  if (wave < nkt) {
+    const int k0 = 16 * wave;
+    const int smin = CAUSAL ? k0 / 32 : 0;
+    const int nsq = (Lq + 31) / 32;
+    f32x4_t adk[4] = {}, adv[4] = {};
+    for (int sb = smin; sb < nsq; ++sb) {
+      bf16x8_t bq[4], bd[4];
+      tr_frag_x4((lds_cp)(const void*)(imgQ + sb * 32 * D) + lane * 8, bq);
+      tr_frag_x4((lds_cp)(const void*)(imgA + sb * 32 * D) + lane * 8, bd);
+      const bf16x8_t ak =
+          *reinterpret_cast<const bf16x8_t*>(DST + (k0 + lo) * TP + 32 * sb + hi * 8);
+      const bf16x8_t av =
+          *reinterpret_cast<const bf16x8_t*>(PT + (k0 + lo) * TP + 32 * sb + hi * 8);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        adk[dt] = MFMA16(ak, bq[dt], adk[dt]);
+        adv[dt] = MFMA16(av, bd[dt], adv[dt]);
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int kr = k0 + 4 * hi + r;
+      if (kr >= Lk) continue;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        dkp[(int64_t)kr * dk_sl + 16 * dt + lo] = f2bf(adk[dt][r]);
+        dvp[(int64_t)kr * dv_sl + 16 * dt + lo] = f2bf(adv[dt][r]);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
@@ -652,6 +859,46 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
     TORCH_CHECK(t->stride(3) == 1, "attn_bwd_fused: innermost dim must be contiguous");
   TORCH_CHECK(lse.is_contiguous() && lse.scalar_type() == torch::kFloat32);
   auto stream = at::hip::getCurrentHIPStream();
+
+  // small-L fused path: one kernel, no delta buffer (Lq == Lk <= 80, D = 64)
+  if (Dr == 64 && Lq == Lk && Lk <= 80) {
+    const dim3 sgrid((unsigned)((int64_t)B * H));
+    const size_t sshmem = (2 * 96 * 64 + 3 * 80 * 88 + 8) * sizeof(short);
+#define SBWD_ARGS                                                                          \
+                     reinterpret_cast<const bf16*>(q.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(k.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(v.data_ptr()),                          \
+                     reinterpret_cast<const bf16*>(dO.data_ptr()), lse.data_ptr<float>(),  \
+                     reinterpret_cast<bf16*>(dq.data_ptr()),                               \
+                     reinterpret_cast<bf16*>(dk.data_ptr()),                               \
+                     reinterpret_cast<bf16*>(dv.data_ptr()), Lq, Lk, (float)scale, H,      \
+                     q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),      \
+                     k.stride(2), v.stride(0), v.stride(1), v.stride(2), dO.stride(0),     \
+                     dO.stride(1), dO.stride(2), dq.stride(0), dq.stride(1), dq.stride(2), \
+                     dk.stride(0), dk.stride(1), dk.stride(2), dv.stride(0), dv.stride(1), \
+                     dv.stride(2)
+    if (causal) {
+      auto kfn = attn_bwd_small_kernel<true>;
+      static bool attr_sb_c = [&] {
+        hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),
+                            hipFuncAttributeMaxDynamicSharedMemorySize, (int)sshmem);
+        return true;
+      }();
+      (void)attr_sb_c;
+      hipLaunchKernelGGL(kfn, sgrid, dim3(320), sshmem, stream, SBWD_ARGS);
+    } else {
+      auto kfn = attn_bwd_small_kernel<false>;
+      static bool attr_sb_n = [&] {
+        hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),
+                            hipFuncAttributeMaxDynamicSharedMemorySize, (int)sshmem);
+        return true;
+      }();
+      (void)attr_sb_n;
+      hipLaunchKernelGGL(kfn, sgrid, dim3(320), sshmem, stream, SBWD_ARGS);
+    }
+#undef SBWD_ARGS
+    return;
+  }
 
   // D = rowsum(dO * O)
   auto Dv = torch::empty({(int64_t)B * H * Lq}, q.options().dtype(torch::kFloat32));
