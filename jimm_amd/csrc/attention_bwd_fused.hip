@@ -653,8 +653,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 // Tile pitch 88 < 96: over-reads past col 87 alias the next row's finite
 // values and multiply zero-padded image rows (see attn_fwd_small_kernel);
 // tiles are zero-initialized once so causal-masked entries (never written)
-// contribute exact zeros. LDS: 2 images + 3 tiles + pad = 65.3 KiB -> 2
-// workgroups per CU.
+// contribute exact zeros.
 // ---------------------------------------------------------------------------
 
 template <bool CAUSAL>
@@ -672,9 +671,10 @@ __global__ __launch_bounds__(320) void attn_bwd_small_kernel(
   constexpr int IMG = LP * D;    // shorts per image
   constexpr int TILE = 80 * TP;  // shorts per tile
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* imgA = reinterpret_cast<short*>(smem);  // K image (ph1) -> dO image (ph2)
+  short* imgA = reinterpret_cast<short*>(smem);  // K image
   short* imgQ = imgA + IMG;                      // Q image (ph2)
-  short* PT = imgQ + IMG;                        // P^T  [key][q]
+  short* imgD = imgQ + IMG;                      // dO image (ph2)
+  short* PT = imgD + IMG;                        // P^T  [key][q]
   short* DST = PT + TILE;                        // dS^T [key][q]
   short* DSQ = DST + TILE;                       // dS   [q][key] (+8 zeroed pad after)
 
@@ -704,6 +704,7 @@ __global__ __launch_bounds__(320) void attn_bwd_small_kernel(
   };
   stage(imgA, kp, k_sl, Lk);
   stage(imgQ, qp, q_sl, Lq);
+  stage(imgD, dop, do_sl, Lq);
   const int nq = (Lq + 15) / 16;
   const int nkt = (Lk + 15) / 16;
   __syncthreads();
@@ -803,9 +804,7 @@ __global__ __launch_bounds__(320) void attn_bwd_small_kernel(
         dqp[(int64_t)qr * dq_sl + 16 * dt + lo] = f2bf(adq[dt][r]);
     }
   }
-  __syncthreads();
-  stage(imgA, dop, do_sl, Lq);  // dO image replaces K (phase-1 readers done)
-  __syncthreads();
+  __syncthreads();  // phase-1 tile writes visible
 
   // ---- phase 2: per key-strip ------------------------------------------
   if (wave < nkt) {
@@ -816,7 +815,7 @@ __global__ __launch_bounds__(320) void attn_bwd_small_kernel(
     for (int sb = smin; sb < nsq; ++sb) {
       bf16x8_t bq[4], bd[4];
       tr_frag_x4((lds_cp)(const void*)(imgQ + sb * 32 * D) + lane * 8, bq);
-      tr_frag_x4((lds_cp)(const void*)(imgA + sb * 32 * D) + lane * 8, bd);
+      tr_frag_x4((lds_cp)(const void*)(imgD + sb * 32 * D) + lane * 8, bd);
       const bf16x8_t ak =
           *reinterpret_cast<const bf16x8_t*>(DST + (k0 + lo) * TP + 32 * sb + hi * 8);
       const bf16x8_t av =
@@ -863,7 +862,7 @@ void attn_bwd_fused(torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Te
   // small-L fused path: one kernel, no delta buffer (Lq == Lk <= 80, D = 64)
   if (Dr == 64 && Lq == Lk && Lk <= 80) {
     const dim3 sgrid((unsigned)((int64_t)B * H));
-    const size_t sshmem = (2 * 96 * 64 + 3 * 80 * 88 + 8) * sizeof(short);
+    const size_t sshmem = (3 * 96 * 64 + 3 * 80 * 88 + 8) * sizeof(short);
 #define SBWD_ARGS                                                                          \
                      reinterpret_cast<const bf16*>(q.data_ptr()),                          \
                      reinterpret_cast<const bf16*>(k.data_ptr()),                          \
