@@ -22,7 +22,7 @@ from jimm_amd.ops._backend import maybe_enable_tunableop
 FLOORS_ATTN = {  # (B, H, L, causal): (fwd_tf, bwd_tf)
     (256, 12, 197, False): (150, 155),
     (64, 16, 577, False): (265, 235),
-    (256, 8, 77, True): (72, 44),     # strip-per-wave small-L kernel (76.4)
+    (256, 8, 77, True): (72, 62),     # strip-per-wave small-L fwd + fused small bwd
     (256, 12, 257, False): (195, 165),
 }
 FLOOR_STEP_VIT_B1024 = 5350  # img/s (r02 final: 5486)
