@@ -55,6 +55,7 @@ std::vector<torch::Tensor> sigmoid_loss_ew(torch::Tensor logits, int64_t diag0);
 bool gemm8p_supported(int64_t M, int64_t N, int64_t K);
 bool gemm_tn8p_supported(int64_t M, int64_t N, int64_t K);
 torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x);
+std::vector<torch::Tensor> gemm_tn_8p_db(torch::Tensor dz, torch::Tensor x);
 torch::Tensor gemm_nt_8p_gradact(torch::Tensor dy, torch::Tensor wt, torch::Tensor z, std::string act);
 torch::Tensor tr16_probe(torch::Tensor src, int64_t mode);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
@@ -92,6 +93,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm8p_supported", &gemm8p_supported, "8-phase 256-tile GEMM shape check");
   m.def("gemm_tn8p_supported", &gemm_tn8p_supported, "TN dW GEMM shape check");
   m.def("gemm_tn_8p", &gemm_tn_8p, "TN weight-grad GEMM, split-M, tr_b16 (K15)");
+  m.def("gemm_tn_8p_db", &gemm_tn_8p_db, "split-M TN dW with fused bias-grad colsum");
   m.def("gemm_nt_8p_gradact", &gemm_nt_8p_gradact, "dX GEMM with fused act-backward epilogue");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 lane-mapping probe");
   m.def("l2norm_fwd", &l2norm_fwd, "row L2-normalize forward (K12)");

@@ -81,10 +81,10 @@ __device__ __forceinline__ void glds16t(const bf16* g, char* lds_dst) {
 // of MFMA operands for fragments f and f+1?  We issue per-phase batches
 // instead — see the asm blocks in the kernel.
 
-template <bool ATOMIC>
+template <bool ATOMIC, bool DBOUT = false>
 __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
     const bf16* __restrict__ DZ, const bf16* __restrict__ X, float* __restrict__ C,
-    int64_t M, int N, int Kw, int64_t chunk_m) {
+    int64_t M, int N, int Kw, int64_t chunk_m, float* __restrict__ DB) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // dz images: [buf]*32 KiB at 0; X images at 64 KiB.
   auto DZs = [&](int buf) { return smem + buf * IMG_BYTES; };
@@ -114,6 +114,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
   if (ntile <= 0) return;
 
   f32x4_t acc[2][4][4] = {};  // [pr][mi][ni]
+  // DBOUT: db[n] = sum_m dz[m][n] folded out of the A fragments already in
+  // registers (the separate colsum kernel re-reads dz from HBM). Only the
+  // k0==0 / wn==0 waves contribute (the same dz fragments are read by every
+  // k-tile and every wn), so the whole-M coverage comes from the k0==0
+  // column of workgroups across splits.
+  float db_acc[2][4] = {};
 
   // Staging decode for this thread.  Fast path (full 64-row tiles): glds
   // with the image permutation on the per-lane SOURCE address — round r
@@ -234,6 +240,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
         ad[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
+        if (DBOUT && wn == 0 && k0 == 0) {
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi) {
+            const bf16* av = reinterpret_cast<const bf16*>(&ad[mi]);
+            float s8 = 0.f;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) s8 += bf2f(av[j]);
+            db_acc[0][mi] += s8;
+          }
+        }
         raw_barrier_tn();
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -269,6 +285,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
         ad[1] = __builtin_shufflevector(a1l, a1h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[2] = __builtin_shufflevector(a2l, a2h, 0, 1, 2, 3, 4, 5, 6, 7);
         ad[3] = __builtin_shufflevector(a3l, a3h, 0, 1, 2, 3, 4, 5, 6, 7);
+        if (DBOUT && wn == 0 && k0 == 0) {
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi) {
+            const bf16* av = reinterpret_cast<const bf16*>(&ad[mi]);
+            float s8 = 0.f;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) s8 += bf2f(av[j]);
+            db_acc[1][mi] += s8;
+          }
+        }
         // kh==1 (phase p3): drain the glds staged at p1 before the barrier
         // that gates the next tile's reads
         if (kh == 1 && more && next_full) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -282,6 +308,21 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
     }
+  }
+
+  if (DBOUT && wn == 0 && k0 == 0) {
+    // A-frag row for lane = lo: db_acc[pr][mi] covers n = n0 + pr*128 +
+    // wm*64 + 16*mi + lo over this wave's m chunks (hi groups); reduce
+    // over hi, one atomic per n per split
+#pragma unroll
+    for (int pr = 0; pr < 2; ++pr)
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        float v_ = db_acc[pr][mi];
+        v_ += __shfl_xor(v_, 16, WAVE);
+        v_ += __shfl_xor(v_, 32, WAVE);
+        if (hi == 0) atomicAdd(DB + n0 + pr * 128 + wm * 64 + 16 * mi + lo, v_);
+      }
   }
 
   // Epilogue: acc[pr][mi][ni] -> C rows n = n0 + pr*128 + wm*64 + 16mi +
@@ -309,8 +350,9 @@ bool gemm_tn8p_supported(int64_t M, int64_t N, int64_t K) {
   return M >= 1 && (N % BN == 0) && (K % BKW == 0);
 }
 
-torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x) {
+static torch::Tensor tn8p_run(torch::Tensor dz, torch::Tensor x, float* dbp) {
   // dz (M, N) bf16, x (M, Kw) bf16 -> dW (N, Kw) fp32 = dz^T @ x
+  // (dbp != nullptr: also fold db[n] = colsum(dz) out of the A fragments)
   TORCH_CHECK(dz.is_cuda() && dz.is_contiguous() && x.is_contiguous());
   TORCH_CHECK(dz.scalar_type() == torch::kBFloat16 && x.scalar_type() == torch::kBFloat16);
   const int64_t M = dz.size(0);
@@ -342,22 +384,35 @@ torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x) {
                       : torch::empty({(int64_t)N, (int64_t)Kw},
                                      dz.options().dtype(torch::kFloat32));
   const size_t shmem = 4 * IMG_BYTES;  // 128 KiB
-#define LAUNCH_TN(AT)                                                                      \
+#define LAUNCH_TN(AT, DBO)                                                                 \
   do {                                                                                     \
-    auto kfn = gemm_tn_8p_kernel<AT>;                                                      \
-    static bool attr_##AT = [&] {                                                          \
+    auto kfn = gemm_tn_8p_kernel<AT, DBO>;                                                 \
+    static bool attr_##AT##DBO = [&] {                                                     \
       hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),                              \
                           hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);         \
       return true;                                                                         \
     }();                                                                                   \
-    (void)attr_##AT;                                                                       \
+    (void)attr_##AT##DBO;                                                                  \
     hipLaunchKernelGGL(kfn, dim3(tiles, splitm), dim3(NTHREADS), shmem, stream,            \
                        reinterpret_cast<const bf16*>(dz.data_ptr()),                       \
                        reinterpret_cast<const bf16*>(x.data_ptr()), C.data_ptr<float>(),   \
-                       M, N, Kw, chunk);                                                   \
+                       M, N, Kw, chunk, dbp);                                              \
   } while (0)
-  if (splitm > 1) LAUNCH_TN(true);
-  else LAUNCH_TN(false);
+  if (dbp && splitm > 1) LAUNCH_TN(true, true);
+  else if (dbp) LAUNCH_TN(false, true);
+  else if (splitm > 1) LAUNCH_TN(true, false);
+  else LAUNCH_TN(false, false);
 #undef LAUNCH_TN
   return C;
+}
+
+torch::Tensor gemm_tn_8p(torch::Tensor dz, torch::Tensor x) {
+  return tn8p_run(dz, x, nullptr);
+}
+
+std::vector<torch::Tensor> gemm_tn_8p_db(torch::Tensor dz, torch::Tensor x) {
+  // fused dW + bias-grad: {dW (N, Kw) fp32, db (N) fp32}
+  auto db = torch::zeros({dz.size(1)}, dz.options().dtype(torch::kFloat32));
+  auto C = tn8p_run(dz, x, db.data_ptr<float>());
+  return {C, db};
 }

@@ -19,7 +19,7 @@ import os
 import torch
 
 from jimm_amd.ops import _backend
-from jimm_amd.ops.hip_linear import _dw_gemm, _dx_gemm, _gemm_mode
+from jimm_amd.ops.hip_linear import _dw_db_gemm, _dw_gemm, _dx_gemm, _gemm_mode
 
 
 def _colsum(ext, dz: torch.Tensor) -> torch.Tensor:
@@ -195,8 +195,7 @@ class EncoderBlockFn(torch.autograd.Function):
         else:
             df = torch.matmul(dy2, w2)
             dz1 = ext.act_bwd(df, z1, act)
-        dw2 = _dw_gemm(ext, dy2, f, w2.dtype)
-        db2 = _colsum(ext, dy2)
+        dw2, db2 = _dw_db_gemm(ext, dy2, f, w2.dtype)
         if fp8:
             w1t8, sw1 = _quant_e4m3_t(w1)
             dh2 = torch._scaled_mm(
@@ -205,8 +204,7 @@ class EncoderBlockFn(torch.autograd.Function):
             )
         else:
             dh2 = _dx_gemm(ext, dz1, w1)
-        dw1 = _dw_gemm(ext, dz1, h2_2, w1.dtype)
-        db1 = _colsum(ext, dz1)
+        dw1, db1 = _dw_db_gemm(ext, dz1, h2_2, w1.dtype)
         # LN2 backward with the MLP residual grad (dy) fused into dx
         da3, dln2w, dln2b = ext.layernorm_bwd(
             dh2.view(B, L, H), a.view(B, L, H), ln2w, mean2, rstd2, dy.contiguous()
@@ -215,8 +213,7 @@ class EncoderBlockFn(torch.autograd.Function):
 
         # attention out-projection
         do2 = _dx_gemm(ext, da2, wproj)
-        dwproj = _dw_gemm(ext, da2, o.transpose(1, 2).reshape(-1, H), wproj.dtype)
-        dbproj = _colsum(ext, da2)
+        dwproj, dbproj = _dw_db_gemm(ext, da2, o.transpose(1, 2).reshape(-1, H), wproj.dtype)
         do_v = do2.view(B, L, nh, H // nh).permute(0, 2, 1, 3)   # (B,nh,L,d) strided view
 
         # fused flash attention backward straight into the strided dqkv
@@ -236,8 +233,7 @@ class EncoderBlockFn(torch.autograd.Function):
         # and it costs dx accuracy. fc1-dX gets its e4m3 operand free from
         # the gradact epilogue, so only that dX runs fp8.)
         dh1 = _dx_gemm(ext, dqkv2, wqkv)
-        dwqkv = _dw_gemm(ext, dqkv2, h1_2, wqkv.dtype)
-        dbqkv = _colsum(ext, dqkv2)
+        dwqkv, dbqkv = _dw_db_gemm(ext, dqkv2, h1_2, wqkv.dtype)
         # LN1 backward with the attention residual grad (da) fused into dx
         dx, dln1w, dln1b = ext.layernorm_bwd(
             dh1.view(B, L, H), x, ln1w, mean1, rstd1, da3

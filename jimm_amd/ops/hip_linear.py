@@ -105,6 +105,26 @@ def _dw_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype) -> torch.Tensor
     return torch.matmul(dz.t(), x2)
 
 
+def _dw_db_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype):
+    """dW AND the bias grad in one kernel: db = colsum(dz) is folded out of
+    the dW kernel's dz fragments (already in registers) instead of a
+    separate full re-read of dz (csrc/gemm_tn8p.hip DBOUT)."""
+    if (
+        _gemm_mode() == "hip"
+        and not _deterministic()
+        and dz.dtype == torch.bfloat16
+        and ext.gemm_tn8p_supported(dz.shape[0], dz.shape[1], x2.shape[1])
+    ):
+        dw, db = ext.gemm_tn_8p_db(dz, x2)
+        return dw.to(out_dtype), db.to(dz.dtype)
+    dw = torch.matmul(dz.t(), x2)
+    if dz.is_cuda and dz.shape[-1] % 8 == 0:
+        db = _backend.ext().colsum(dz).to(dz.dtype)
+    else:
+        db = dz.sum(dim=0)
+    return dw, db
+
+
 def _dx_gemm(ext, dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy.
 
@@ -173,14 +193,17 @@ class _LinearActFn(torch.autograd.Function):
             dz = dy2
         ext = _backend.ext()
         dx = _dx_gemm(ext, dz, w).view(ctx.x_shape) if ctx.needs_input_grad[0] else None
-        dw = _dw_gemm(ext, dz, x2, w.dtype) if ctx.needs_input_grad[1] else None
-        if ctx.has_bias and ctx.needs_input_grad[2]:
-            if dz.shape[-1] % 8 == 0:
-                db = _backend.ext().colsum(dz).to(dz.dtype)
-            else:
-                db = dz.sum(dim=0)
+        if ctx.needs_input_grad[1] and ctx.has_bias and ctx.needs_input_grad[2]:
+            dw, db = _dw_db_gemm(ext, dz, x2, w.dtype)
         else:
-            db = None
+            dw = _dw_gemm(ext, dz, x2, w.dtype) if ctx.needs_input_grad[1] else None
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                if dz.is_cuda and dz.shape[-1] % 8 == 0:
+                    db = _backend.ext().colsum(dz).to(dz.dtype)
+                else:
+                    db = dz.sum(dim=0)
+            else:
+                db = None
         dres = dy.view(ctx.res_shape) if (ctx.has_res and ctx.needs_input_grad[4]) else None
         return dx, dw, db, None, dres
 

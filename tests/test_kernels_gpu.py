@@ -525,6 +525,22 @@ def test_fp8_linear_path():
     assert torch.isfinite(x.grad).all()
 
 
+def test_dw_db_fused():
+    """gemm_tn_8p_db: dW and db match the fp32 references."""
+    from jimm_amd.ops import _backend
+
+    ext = _backend.ext()
+    torch.manual_seed(0)
+    for m, n, k in [(4096, 768, 512), (7000, 1024, 256)]:
+        dz = torch.randn(m, n, device=dev()).bfloat16()
+        x = torch.randn(m, k, device=dev()).bfloat16()
+        dw, db = ext.gemm_tn_8p_db(dz, x)
+        dw_ref = dz.t().float() @ x.float()
+        db_ref = dz.float().sum(0)
+        assert rel_err(dw, dw_ref) < 2e-2, (m, n, k, rel_err(dw, dw_ref))
+        assert rel_err(db, db_ref) < 1e-3, (m, n, k, rel_err(db, db_ref))
+
+
 def test_ln_fp8_producer():
     """layernorm_fwd_fp8: bf16 y identical to layernorm_fwd; y8*scale ~= y;
     amax == max|y|."""
