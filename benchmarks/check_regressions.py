@@ -25,7 +25,7 @@ FLOORS_ATTN = {  # (B, H, L, causal): (fwd_tf, bwd_tf)
     (256, 8, 77, True): (72, 62),     # strip-per-wave small-L fwd + fused small bwd
     (256, 12, 257, False): (195, 165),
 }
-FLOOR_STEP_VIT_B1024 = 5350  # img/s (r02 final: 5486)
+FLOOR_STEP_VIT_B1024 = 5500  # img/s (r02 final: 5753)
 # in-house GEMM floors on the b1024 model shapes (median TF/s, -10%)
 FLOORS_GEMM = {  # (M, N, K, act): fwd_tf
     (201728, 2304, 768, ""): 790,
