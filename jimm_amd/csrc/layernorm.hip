@@ -25,6 +25,88 @@ namespace {
 // this call's |y| max accumulates into amax_out (uint-bits atomicMax, one
 // per wave) — the NEXT step's scale is amax/448.  No extra read passes:
 // the quantization rides the existing LN write.
+// Register-cached forward (H <= 2048, H % 64 == 0): w/b live in registers
+// across the whole row loop (the generic kernel re-reads 8 KB of w/b from
+// L2 per row) and x is kept in registers between the stat pass and the
+// normalize pass (the generic kernel re-reads it). NI = ceil(H/512) 8-float
+// chunks per lane; the last chunk is lane-ragged for H % 512 != 0 (H % 64
+// == 0 keeps chunks all-in or all-out per lane).
+template <typename T, int NI, bool FP8O = false>
+__global__ void ln_fwd_rc_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                                 const float* __restrict__ b, T* __restrict__ y,
+                                 float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                                 int64_t nrows, int H, float eps,
+                                 unsigned char* __restrict__ y8 = nullptr,
+                                 const float* __restrict__ scale8 = nullptr,
+                                 unsigned int* __restrict__ amax_bits = nullptr) {
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  float rs8 = 1.f, tmax = 0.f;
+  if (FP8O) rs8 = 1.f / scale8[0];
+  float wv[NI][8], bv[NI][8];
+  bool act[NI];
+#pragma unroll
+  for (int i = 0; i < NI; ++i) {
+    const int c = lane * 8 + i * WAVE * 8;
+    act[i] = c < H;
+    if (act[i]) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        wv[i][j] = w[c + j];
+        bv[i][j] = b[c + j];
+      }
+    }
+  }
+  for (int64_t row = (int64_t)blockIdx.x * waves_per_block + wave; row < nrows;
+       row += (int64_t)gridDim.x * waves_per_block) {
+    const T* xr = x + row * H;
+    T* yr = y + row * H;
+    float xv[NI][8];
+    float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+    for (int i = 0; i < NI; ++i) {
+      if (!act[i]) continue;
+      vload_f32<8>(xr + lane * 8 + i * WAVE * 8, xv[i]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sum += xv[i][j];
+        sumsq += xv[i][j] * xv[i][j];
+      }
+    }
+    sum = wave_reduce_sum(sum);
+    sumsq = wave_reduce_sum(sumsq);
+    const float mean = sum / H;
+    const float var = sumsq / H - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    if (lane == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+#pragma unroll
+    for (int i = 0; i < NI; ++i) {
+      if (!act[i]) continue;
+      float yv[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) yv[j] = (xv[i][j] - mean) * rstd * wv[i][j] + bv[i][j];
+      vstore_f32<8>(yr + lane * 8 + i * WAVE * 8, yv);
+      if (FP8O) {
+        unsigned short q8[4];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) tmax = fmaxf(tmax, fabsf(yv[j]));
+#pragma unroll
+        for (int j = 0; j < 8; j += 2) q8[j / 2] = cvt2_e4m3(yv[j] * rs8, yv[j + 1] * rs8);
+        *reinterpret_cast<uint2*>(y8 + row * H + lane * 8 + i * WAVE * 8) =
+            *reinterpret_cast<uint2*>(q8);
+      }
+    }
+  }
+  if (FP8O) {
+    tmax = wave_reduce_max(tmax);
+    if (lane == 0) atomicMax(amax_bits, __float_as_uint(tmax));
+  }
+}
+
 template <typename T, int VEC, bool FP8O = false>
 __global__ void ln_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
                               const float* __restrict__ b, T* __restrict__ y,
@@ -259,7 +341,16 @@ void launch_ln_fwd(const T* x, const float* w, const float* b, T* y, float* mean
     hipLaunchKernelGGL((ln_fwd_kernel<T, V>), dim3(grid), dim3(block), 0, stream, x, w, b,
                        y, mean, rstd, nrows, H, eps);
   };
-  if (H % (WAVE * 8) == 0) pick(std::integral_constant<int, 8>{});
+  auto pick_rc = [&](auto ni_tag) {
+    constexpr int NI = decltype(ni_tag)::value;
+    hipLaunchKernelGGL((ln_fwd_rc_kernel<T, NI>), dim3(grid), dim3(block), 0, stream, x, w,
+                       b, y, mean, rstd, nrows, H, eps, nullptr, nullptr, nullptr);
+  };
+  if (H % 64 == 0 && H <= 512) pick_rc(std::integral_constant<int, 1>{});
+  else if (H % 64 == 0 && H <= 1024) pick_rc(std::integral_constant<int, 2>{});
+  else if (H % 64 == 0 && H <= 1536) pick_rc(std::integral_constant<int, 3>{});
+  else if (H % 64 == 0 && H <= 2048) pick_rc(std::integral_constant<int, 4>{});
+  else if (H % (WAVE * 8) == 0) pick(std::integral_constant<int, 8>{});
   else if (H % (WAVE * 4) == 0) pick(std::integral_constant<int, 4>{});
   else if (H % (WAVE * 2) == 0) pick(std::integral_constant<int, 2>{});
   else if (H % WAVE == 0) pick(std::integral_constant<int, 1>{});
@@ -316,12 +407,18 @@ std::vector<torch::Tensor> layernorm_fwd_fp8(torch::Tensor x, torch::Tensor w,
   auto stream = at::hip::getCurrentHIPStream();
   const int block = 256;
   const int grid = (int)std::min<int64_t>((nrows + 3) / 4, 2048);
-  hipLaunchKernelGGL((ln_fwd_kernel<bf16, 8, true>), dim3(grid), dim3(block), 0, stream,
-                     reinterpret_cast<const bf16*>(x.data_ptr()), wf.data_ptr<float>(),
-                     bf.data_ptr<float>(), reinterpret_cast<bf16*>(y.data_ptr()),
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(), nrows, H, 0.f + (float)eps,
-                     y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
-                     reinterpret_cast<unsigned int*>(amax.data_ptr()));
+#define LN8_ARGS                                                                           \
+  dim3(grid), dim3(block), 0, stream, reinterpret_cast<const bf16*>(x.data_ptr()),         \
+      wf.data_ptr<float>(), bf.data_ptr<float>(), reinterpret_cast<bf16*>(y.data_ptr()),   \
+      mean.data_ptr<float>(), rstd.data_ptr<float>(), nrows, H, 0.f + (float)eps,          \
+      y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),                              \
+      reinterpret_cast<unsigned int*>(amax.data_ptr())
+  if (H <= 512) hipLaunchKernelGGL((ln_fwd_rc_kernel<bf16, 1, true>), LN8_ARGS);
+  else if (H <= 1024) hipLaunchKernelGGL((ln_fwd_rc_kernel<bf16, 2, true>), LN8_ARGS);
+  else if (H <= 1536) hipLaunchKernelGGL((ln_fwd_rc_kernel<bf16, 3, true>), LN8_ARGS);
+  else if (H <= 2048) hipLaunchKernelGGL((ln_fwd_rc_kernel<bf16, 4, true>), LN8_ARGS);
+  else hipLaunchKernelGGL((ln_fwd_kernel<bf16, 8, true>), LN8_ARGS);
+#undef LN8_ARGS
   return {y, y8, mean, rstd};
 }
 
