@@ -65,30 +65,6 @@ __device__ __forceinline__ int boff_f(int q, int d) {
 typedef short bf16x4_trf __attribute__((ext_vector_type(4)));
 typedef const __attribute__((address_space(3))) char* lds_cpf;
 
-__device__ __forceinline__ void glds16f(const bf16* g, char* lds_dst) {
-  typedef const __attribute__((address_space(1))) unsigned int* gp_t;
-  typedef __attribute__((address_space(3))) unsigned int* lp_t;
-  __builtin_amdgcn_global_load_lds((gp_t)(const void*)g, (lp_t)(void*)lds_dst, 16, 0, 0);
-}
-
-// boff_f inverse at 16-byte-chunk granularity for glds staging of a
-// ND == 64 block image (512 chunks): lane-linear LDS dest, image
-// permutation applied to the per-lane SOURCE address (gemm_tn8p.hip
-// recipe). Register staging's ds_write_b128 into this layout is
-// inherently 4-way bank-conflicted (only the (q&3)*8-word term survives
-// mod 32 banks — measured 3.5e9 SQ_LDS_BANK_CONFLICT cycles across the
-// attention benches); global_load_lds bypasses the write crossbar.
-__device__ __forceinline__ void boff_f_inv64(int chunk, int& q, int& d) {
-  const int p = chunk * 8;  // shorts
-  const int r1 = p & 2047;
-  const int r2 = r1 & 511;
-  const int r3 = r2 & 63;
-  const int qpos = r2 >> 6;
-  const int qp = ((qpos & 3) << 1) | (qpos >> 2);
-  q = (p >> 11) * 32 + qp * 4 + (r3 >> 4);
-  d = (r1 >> 9) * 16 + (r3 & 15);
-}
-
 __device__ __forceinline__ void trf_x4(lds_cpf base, bf16x8_t (&out)[4]) {
   bf16x4_trf a0l, a0h, a1l, a1h, a2l, a2h, a3l, a3h;
   asm volatile(
@@ -199,78 +175,54 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int ti_max = blockIdx.x + (nactive - 1) * gridDim.x;  // last active tile
   const int kv_end = CAUSAL ? min(Lk, (ti_max + 1) * QBLK) : Lk;
 
-  // T14 staging state. DP == 64: FULL kv tiles stage the V block image by
-  // glds (lane-linear dest, boff_f-permuted source) — register ds_writes
-  // into the block layout are 4-way bank-conflicted; ragged tiles (rows
-  // needing the zero fill) and unaligned V keep the register path.
+  // T14 staging state
   constexpr int NCG = (DP + 63) / 64;  // 64-wide column groups per row
   const int st_row = tid / 4;
   const int st_c0 = (tid % 4) * 16;
   bf16x8_t kreg[NCG][2], vreg[NCG][2];
   bool st_valid;
-  bool v_glds = false;
-  int gq[2] = {}, gd[2] = {};
-  if constexpr (DP == 64) {
-    v_glds = ((reinterpret_cast<uintptr_t>(vp) & 15) == 0) && (v_sl % 8 == 0);
-#pragma unroll
-    for (int r = 0; r < 2; ++r) boff_f_inv64(r * 256 + tid, gq[r], gd[r]);
-  }
-  auto tile_full = [&](int kv0) { return v_glds && kv0 + KVBLK <= Lk; };
   auto load_tile_regs = [&](int kv0) {
     const int key = kv0 + st_row;
     st_valid = key < Lk;
     const int krow = min(key, Lk - 1);
-    const bool skip_v = tile_full(kv0);
 #pragma unroll
     for (int cg = 0; cg < NCG; ++cg) {
       const int c0 = cg * 64 + st_c0;
       if (c0 >= DP) continue;
       kreg[cg][0] = ld8g<DP>(kp + (int64_t)krow * k_sl, c0, Dr);
       kreg[cg][1] = ld8g<DP>(kp + (int64_t)krow * k_sl, c0 + 8, Dr);
-      if (!skip_v) {
-        vreg[cg][0] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0, Dr);
-        vreg[cg][1] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0 + 8, Dr);
-      }
+      vreg[cg][0] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0, Dr);
+      vreg[cg][1] = ld8g<DP>(vp + (int64_t)krow * v_sl, c0 + 8, Dr);
     }
   };
-  auto write_tile = [&](int buf, int kv0) {
+  auto write_tile = [&](int buf) {
     const int bo = buf * IMGSF;
-    const bool vg = tile_full(kv0);
-    if (DP == 64 && vg) {
-      char* vi = reinterpret_cast<char*>(vt_lds + bo);
-#pragma unroll
-      for (int r = 0; r < 2; ++r)
-        glds16f(vp + (int64_t)(kv0 + gq[r]) * v_sl + gd[r], vi + (r * 256 + tid) * 16);
-    }
 #pragma unroll
     for (int cg = 0; cg < NCG; ++cg) {
       const int c0 = cg * 64 + st_c0;
       if (c0 >= DP) continue;
       const bf16x8_t k0 = st_valid ? kreg[cg][0] : bf16x8_t{};
       const bf16x8_t k1 = st_valid ? kreg[cg][1] : bf16x8_t{};
+      const bf16x8_t v0 = st_valid ? vreg[cg][0] : bf16x8_t{};
+      const bf16x8_t v1 = st_valid ? vreg[cg][1] : bf16x8_t{};
       *reinterpret_cast<bf16x8_t*>(k_lds + bo + st_row * LDS_PITCH + c0) = k0;
       *reinterpret_cast<bf16x8_t*>(k_lds + bo + st_row * LDS_PITCH + c0 + 8) = k1;
-      if (!vg) {
-        const bf16x8_t v0 = st_valid ? vreg[cg][0] : bf16x8_t{};
-        const bf16x8_t v1 = st_valid ? vreg[cg][1] : bf16x8_t{};
-        *reinterpret_cast<bf16x8_t*>(vt_lds + bo + boff_f<DP>(st_row, c0)) = v0;
-        *reinterpret_cast<bf16x8_t*>(vt_lds + bo + boff_f<DP>(st_row, c0 + 8)) = v1;
-      }
+      *reinterpret_cast<bf16x8_t*>(vt_lds + bo + boff_f<DP>(st_row, c0)) = v0;
+      *reinterpret_cast<bf16x8_t*>(vt_lds + bo + boff_f<DP>(st_row, c0 + 8)) = v1;
     }
   };
 
   const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
   load_tile_regs(0);
-  write_tile(0, 0);
+  write_tile(0);
   if (ntiles > 1) load_tile_regs(KVBLK);
-  if (v_glds) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   for (int it = 0; it < ntiles; ++it) {
     const int kv0 = it * KVBLK;
     const int sbo = (it & 1) * IMGSF;
     if (it + 1 < ntiles) {
-      write_tile((it + 1) & 1, kv0 + KVBLK);
+      write_tile((it + 1) & 1);
       if (it + 2 < ntiles) load_tile_regs(kv0 + 2 * KVBLK);
     }
 #pragma unroll
@@ -370,9 +322,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       }
     }
     // double-buffered: ONE barrier publishes tile it+1 and retires it
-    // (glds of tile it+1 must land first; the kreg loads for it+2 were
-    // issued after it — their first use waits on them anyway)
-    if (v_glds) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
   }
 
