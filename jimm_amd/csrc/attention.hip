@@ -552,11 +552,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto o_storage = torch::empty({B, Lq, H, Dr}, q.options());
   auto o = o_storage.permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, Lq}, q.options().dtype(torch::kFloat32));
-  // 4 strips only when the q tiles divide evenly (measured: 4 wins at
-  // L=197 — one staging pass; 2 wins elsewhere via 3 waves/SIMD occupancy:
-  // L=577 261->298 TF/s, L=257 184->216, L=77 46->60)
+  // 2 strips everywhere (3 waves/SIMD occupancy): the early-r02 measurement
+  // that had 4 winning at L=197 no longer holds after the block-image /
+  // double-buffer rework (L=197: 160 -> 171 TF/s at nstrip 2; L=257 with 4
+  // is far worse, 207 -> 168). JIMM_AMD_ATTN_NSTRIP overrides for A/B.
   const int ntq = (Lq + QBLK - 1) / QBLK;
-  int nstrip = (DP == 64 && ntq % 4 == 0) ? 4 : 2;
+  int nstrip = 2;
   if (DP == 64) {
     if (const char* e = getenv("JIMM_AMD_ATTN_NSTRIP")) nstrip = (std::string(e) == "2") ? 2 : 4;
   }
