@@ -150,6 +150,8 @@ def _attn_ref(q, k, v, causal, scale):
         (2, 1, 50, 50, False),     # CLIP-B/32 vision
         (1, 1, 130, 130, True),    # ragged, causal
         (1, 1, 64, 64, False),     # single tile
+        (1, 2, 80, 80, True),      # small-L path upper bound
+        (1, 1, 66, 66, True),      # small-L ragged strips
     ],
 )
 def test_attn_fwd(B, H, Lq, Lk, causal):
@@ -313,6 +315,9 @@ def test_attn_fwd_strided_qkv():
         (2, 1, 50, 50, False),     # CLIP-B/32 vision (ragged)
         (1, 1, 130, 130, True),    # ragged, causal
         (2, 2, 1, 256, False),     # MAP head cross-attn (K9, Lq=1)
+        (1, 1, 64, 64, False),     # small-L bwd: exact tile
+        (1, 2, 80, 80, True),      # small-L bwd: upper bound, causal
+        (1, 1, 66, 66, True),      # small-L bwd: ragged strips
     ],
 )
 def test_attn_bwd_fused_kernel(B, H, Lq, Lk, causal):
