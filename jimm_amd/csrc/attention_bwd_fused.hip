@@ -302,6 +302,10 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       // causal diagonal (or beyond Lq) contribute P = dS = 0 — write zeros
       // and skip their MFMAs (at L=77 causal most sub-tiles are waste)
       const int key_min = kv0 + 16 * wave;
+      // whole 16-key strip is padding (ragged last tile, e.g. keys 192..255
+      // at L=197): every P/dS is zero and the dK/dV stores are key-guarded —
+      // skip the strip's S/P/dS and MFMA work outright
+      if (key_min >= Lk) continue;
       bf16x4 ds_stash[4];
 #pragma unroll
       for (int qt = 0; qt < 4; ++qt) {
@@ -554,6 +558,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       if (sidx >= nactive) continue;
       const int q0s = qbase[sidx] + wave * 16;
       if (CAUSAL && kv0 >= qbase[sidx] + BLK) continue;  // tile above diagonal
+      if (q0s >= Lq) continue;  // whole 16-query strip is padding
 
       // ---- S, P, dP, dS per 16-key tile ----------------------------------
       // key sub-tiles above this strip's causal diagonal or beyond Lk give
