@@ -22,12 +22,6 @@ from jimm_amd.ops import _backend
 from jimm_amd.ops.hip_linear import _dw_db_gemm, _dw_gemm, _dx_gemm, _gemm_mode
 
 
-def _colsum(ext, dz: torch.Tensor) -> torch.Tensor:
-    if dz.shape[-1] % 8 == 0:
-        return ext.colsum(dz).to(dz.dtype)
-    return dz.sum(dim=0)
-
-
 def _hip_gemms(M: int, H: int) -> bool:
     """True when the encoder block's GEMMs run on the in-house MFMA kernels
     (JIMM_AMD_GEMM=hip default; all block shapes have N%256==0, K%64==0)."""

@@ -68,18 +68,6 @@ def _gemm_nt_fp8(x2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     )
 
 
-def _gemm_nn_fp8(g: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """dX = g @ w in fp8 (e4m3 per-tensor dynamic scales, bf16 out).
-
-    b operand must be column-major for _scaled_mm: quantize w.t() row-major
-    and pass its transpose view."""
-    g8, sg = _quant_e4m3(g)
-    wt8, sw = _quant_e4m3(w.t().contiguous())
-    return torch._scaled_mm(
-        g8, wt8.t(), scale_a=sg.view(1, 1), scale_b=sw.view(1, 1), out_dtype=torch.bfloat16
-    )
-
-
 def _gemm_mode() -> str:
     # Default "hip" (round 2): the 8-phase 256-tile MFMA GEMM (csrc/gemm8p)
     # with epilogues (bias/act/residual/act-bwd) fused into the GEMM itself,
@@ -128,9 +116,10 @@ def _dw_db_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype):
 def _dx_gemm(ext, dz: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """dX = dz @ w — in-house NT kernel on a pre-transposed weight copy.
 
-    (fp8 dX via _gemm_nn_fp8 was measured NET-NEGATIVE at config 5: the
-    per-call w.t().contiguous() + two quantization passes cost more than
-    the fp8 GEMM saves — backward stays bf16.)"""
+    (Per-call-quantized fp8 dX was measured NET-NEGATIVE at config 5: the
+    w.t().contiguous() + two quantization passes cost more than the fp8
+    GEMM saves. The fused-block path instead runs fc1-dX on fp8 with a
+    producer-emitted e4m3 operand — see ops/block.py.)"""
     if (
         _gemm_mode() == "hip"
         and dz.dtype == torch.bfloat16
