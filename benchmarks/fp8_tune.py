@@ -13,8 +13,9 @@ import torch
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
-M = 128 * 577
-SHAPES = [(3072, 1024), (4096, 1024), (1024, 4096)]
+M = int(os.environ.get("FP8_TUNE_M", 128 * 577))
+SHAPES = [tuple(int(v) for v in p.split("x")) for p in os.environ.get(
+    "FP8_TUNE_SHAPES", "3072x1024,4096x1024,1024x4096").split(",")]
 
 
 def t_ms(fn, iters=30):

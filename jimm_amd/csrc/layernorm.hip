@@ -393,10 +393,12 @@ void launch_ln_bwd(const T* dy, const T* x, const float* w, const float* mean,
 std::vector<torch::Tensor> layernorm_fwd_fp8(torch::Tensor x, torch::Tensor w,
                                              torch::Tensor b, double eps,
                                              torch::Tensor scale8, torch::Tensor amax) {
-  // bf16 in, (y bf16, y8 e4m3-bytes, mean, rstd) out; H % 512 == 0 (VEC 8)
+  // bf16 in, (y bf16, y8 e4m3-bytes, mean, rstd) out; the register-cached
+  // kernel covers H % 64 == 0 up to 2048 (H % 512 != 0 rags the last chunk)
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == torch::kBFloat16);
   const int H = x.size(-1);
-  TORCH_CHECK(H % (WAVE * 8) == 0, "layernorm_fwd_fp8: H % 512 != 0");
+  TORCH_CHECK(H % 64 == 0 && (H <= 2048 || H % (WAVE * 8) == 0),
+              "layernorm_fwd_fp8: H must be a multiple of 64 (and of 512 above 2048)");
   const int64_t nrows = x.numel() / H;
   auto wf = w.contiguous().to(torch::kFloat32);
   auto bf = b.contiguous().to(torch::kFloat32);

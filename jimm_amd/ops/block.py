@@ -100,7 +100,8 @@ class EncoderBlockFn(torch.autograd.Function):
         fp8 = (
             scale8 is not None
             and x.dtype == torch.bfloat16
-            and H % 512 == 0
+            and H % 64 == 0
+            and H <= 2048
             and hasattr(torch, "_scaled_mm")
         )
         if fp8:

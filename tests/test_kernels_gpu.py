@@ -546,16 +546,17 @@ def test_dw_db_fused():
         assert rel_err(db, db_ref) < 1e-3, (m, n, k, rel_err(db, db_ref))
 
 
-def test_ln_fp8_producer():
+@pytest.mark.parametrize("H", [768, 1024])
+def test_ln_fp8_producer(H):
     """layernorm_fwd_fp8: bf16 y identical to layernorm_fwd; y8*scale ~= y;
-    amax == max|y|."""
+    amax == max|y| (H=768 exercises the lane-ragged last chunk)."""
     from jimm_amd.ops import _backend
 
     ext = _backend.ext()
     torch.manual_seed(0)
-    x = torch.randn(64, 197, 1024, device=dev()).bfloat16()
-    w = torch.randn(1024, device=dev()).bfloat16()
-    b = torch.randn(1024, device=dev()).bfloat16()
+    x = torch.randn(64, 197, H, device=dev()).bfloat16()
+    w = torch.randn(H, device=dev()).bfloat16()
+    b = torch.randn(H, device=dev()).bfloat16()
     y_ref, m_ref, r_ref = ext.layernorm_fwd(x, w, b, 1e-6)
     amax0 = y_ref.float().abs().max()
     scale = (amax0 / 448.0).clamp(min=1e-12).reshape(1)
@@ -565,7 +566,7 @@ def test_ln_fp8_producer():
     assert rel_err(y, y_ref) < 1e-3, rel_err(y, y_ref)
     assert torch.allclose(m, m_ref, atol=1e-5) and torch.allclose(r, r_ref, rtol=1e-4)
     deq = y8.view(torch.float8_e4m3fn).float() * scale
-    assert rel_err(deq, y_ref.view(-1, 1024).float()) < 0.04, rel_err(deq, y_ref.float())
+    assert rel_err(deq, y_ref.view(-1, H).float()) < 0.04, rel_err(deq, y_ref.float())
     # kernel tracks amax on fp32 pre-bf16-rounding values: ~0.4% slack
     assert abs(amax.item() - amax0.item()) < 1e-2 * amax0.item()
 
