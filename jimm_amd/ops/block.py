@@ -28,6 +28,9 @@ def _hip_gemms(M: int, H: int) -> bool:
     return _gemm_mode() == "hip" and H % 256 == 0
 
 
+# minimum rows x hidden for the fp8 block path (tests lower it to force fp8)
+_FP8_MIN_MH = 1 << 26
+
 def _fp8_block_state(mod: torch.nn.Module, device: torch.device):
     """Per-block delayed-scaling state, 5 sites: [0] h1->qkv, [1] h2->fc1,
     [2] f->fc2 (forward), [3] dz1->fc1-dX, [4] dqkv->qkv-dX (backward).
@@ -97,16 +100,19 @@ class EncoderBlockFn(torch.autograd.Function):
         # activations. Sites: h1->qkv, h2->fc1, f->fc2; proj keeps the bf16
         # fused-residual GEMM (its input o comes from attention, not from a
         # producer kernel we control cheaply).
-        # fp8 pays only when the GEMMs are big enough to amortize the
-        # per-call weight quantization and the producer-kernel emission:
-        # measured wins at M >= ~74k rows (ViT-L 128x577 +5%, ViT-B
-        # 1024x197 +8%), loses at CLIP-B/32 vision's M = 51200
+        # fp8 pays only when the tower's GEMMs are big enough to amortize
+        # the per-call weight quantization and producer-kernel emission.
+        # M*H separates the measured winners from the losers: ViT-L
+        # 73856x1024 (+5%) and ViT-B 201728x768 (+8%) win; CLIP-B/32
+        # vision 51200x768, CLIP text 78848x512 and SigLIP text 32768x768
+        # all lose — those towers stay on the fused bf16 path even under
+        # set_fp8(True).
         fp8 = (
             scale8 is not None
             and x.dtype == torch.bfloat16
             and H % 64 == 0
             and H <= 2048
-            and B * L > 65536
+            and B * L * H > _FP8_MIN_MH
             and hasattr(torch, "_scaled_mm")
         )
         if fp8:

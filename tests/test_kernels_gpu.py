@@ -598,10 +598,13 @@ def test_fp8_fused_block():
     from jimm_amd.models.common.transformer import EncoderBlock
     from jimm_amd.ops import set_fp8
 
+    import jimm_amd.ops.block as blockmod
+
     torch.manual_seed(2)
     blk = EncoderBlock(512, 8, 2048, hidden_act="gelu", layernorm_epsilon=1e-6).to(dev(), torch.bfloat16)
     x = torch.randn(4, 197, 512, device=dev()).bfloat16()
     dy = torch.randn_like(x) * 0.01
+    blockmod._FP8_MIN_MH = 0  # force the fp8 path at test size
 
     def run(fp8):
         set_fp8(fp8)
@@ -619,7 +622,10 @@ def test_fp8_fused_block():
             set_fp8(False)
         return outs
 
-    y8, dx8, g8 = run(True)
+    try:
+        y8, dx8, g8 = run(True)
+    finally:
+        blockmod._FP8_MIN_MH = 1 << 26
     yb, dxb, gb = run(False)
     assert torch.isfinite(y8).all() and torch.isfinite(dx8).all()
     assert rel_err(y8, yb) < 0.06, rel_err(y8, yb)
@@ -636,7 +642,9 @@ def test_fp8_graph_capture():
     import jimm_amd
     from jimm_amd.ops import set_fp8
     from jimm_amd.train import SyntheticImages, TrainConfig, Trainer
+    import jimm_amd.ops.block as blockmod
 
+    blockmod._FP8_MIN_MH = 0  # force the fp8 path at test size
     torch.manual_seed(0)
     m = jimm_amd.VisionTransformer(
         num_classes=16, img_size=64, patch_size=16, num_layers=2,
@@ -656,6 +664,7 @@ def test_fp8_graph_capture():
         torch.cuda.synchronize()
     finally:
         set_fp8(False)
+        blockmod._FP8_MIN_MH = 1 << 26
     for l in losses:
         v = float(l.item() if hasattr(l, "item") else l)
         assert v == v and abs(v) < 1e4, v
