@@ -5,8 +5,11 @@ identical but lets autograd insert two full-tensor grad-accumulation adds
 per block (the residual branches: x feeds both LN1 and the attention
 residual; a feeds both LN2 and the MLP residual — ~1.1 ms/step for ViT-B
 bs256, profiles/r01_NOTES.md). Here the block is one Function and those
-adds ride the ln_bwd kernel's fused ``addend`` input for free. The forward
-matches the composite blas-mode path kernel-for-kernel.
+adds ride the ln_bwd kernel's fused ``addend`` input for free. The default
+path runs every GEMM on the in-house MFMA kernels with the bias / act /
+residual / act-backward epilogues fused in; under ``set_fp8(True)`` (and
+past the M*H size gate) the forward GEMMs and fc1-dX run e4m3 with
+producer-fused quantization and per-block delayed scaling.
 
 Enabled on GPU when dropout == 0 (the composite path remains the reference
 and the dropout/CPU path). JIMM_AMD_FUSED_BLOCK=0 disables.
@@ -31,18 +34,19 @@ def _hip_gemms(M: int, H: int) -> bool:
 # minimum rows x hidden for the fp8 block path (tests lower it to force fp8)
 _FP8_MIN_MH = 1 << 26
 
+
 def _fp8_block_state(mod: torch.nn.Module, device: torch.device):
-    """Per-block delayed-scaling state, 5 sites: [0] h1->qkv, [1] h2->fc1,
-    [2] f->fc2 (forward), [3] dz1->fc1-dX, [4] dqkv->qkv-dX (backward).
-    scale8 is what this step's producers divide by; amax collected this step
-    becomes next step's scale (scale8[0:3] updated at forward end, [3:5] at
-    backward end). Grad sites start at 1/448 (assume amax~1; adapts in one
+    """Per-block delayed-scaling state, 4 sites: [0] h1->qkv, [1] h2->fc1,
+    [2] f->fc2 (forward), [3] dz1->fc1-dX (backward). scale8 is what this
+    step's producers divide by; amax collected this step becomes next
+    step's scale (scale8[0:3] updated at forward end, [3] at backward
+    end). The grad site starts at 1/448 (assume amax~1; adapts in one
     step) so first-step gradients are not flushed to zero."""
     st = getattr(mod, "_fp8_state", None)
     if st is None or st[0].device != device:
-        scale = torch.ones(5, device=device, dtype=torch.float32)
-        scale[3:5] = 1.0 / 448.0
-        st = (scale, torch.zeros(5, device=device, dtype=torch.float32))
+        scale = torch.ones(4, device=device, dtype=torch.float32)
+        scale[3] = 1.0 / 448.0
+        st = (scale, torch.zeros(4, device=device, dtype=torch.float32))
         mod._fp8_state = st
     return st
 
@@ -245,7 +249,7 @@ class EncoderBlockFn(torch.autograd.Function):
             dh1.view(B, L, H), x, ln1w, mean1, rstd1, da3
         )
         if fp8:
-            # next step's backward scales (delayed)
+            # next step's backward scale (delayed)
             scale8[3:4].copy_(torch.clamp(amax8[3:4] / 448.0, min=1e-12))
         return (dx, dln1w, dln1b, dwqkv, dbqkv, dwproj, dbproj, dln2w, dln2b,
                 dw1, db1, dw2, db2, None, None, None, None, None, None, None)
