@@ -81,7 +81,7 @@ __device__ __forceinline__ void glds16t(const bf16* g, char* lds_dst) {
 // of MFMA operands for fragments f and f+1?  We issue per-phase batches
 // instead — see the asm blocks in the kernel.
 
-template <bool ATOMIC, bool DBOUT = false>
+template <bool ATOMIC, bool DBOUT = false, bool WSOUT = false>
 __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
     const bf16* __restrict__ DZ, const bf16* __restrict__ X, float* __restrict__ C,
     int64_t M, int N, int Kw, int64_t chunk_m, float* __restrict__ DB) {
@@ -337,11 +337,30 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_tn_8p_kernel(
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           const int k = k0 + wn * 64 + 16 * ni + lo;
-          float* dst = C + (int64_t)n * Kw + k;
-          if (ATOMIC) atomicAdd(dst, acc[pr][mi][ni][r]);
+          // WSOUT: each split owns a private C slice (no atomics; a reduce
+          // kernel folds the splits) — A/B alternative to the atomic
+          // combine, selected by JIMM_AMD_DW_WS=1
+          float* dst = C + (WSOUT ? (int64_t)blockIdx.y * N * Kw : 0) +
+                       (int64_t)n * Kw + k;
+          if (ATOMIC && !WSOUT) atomicAdd(dst, acc[pr][mi][ni][r]);
           else *dst = acc[pr][mi][ni][r];
         }
       }
+}
+
+__global__ void ws_reduce_kernel(const float* __restrict__ ws, float* __restrict__ out,
+                                 int64_t n, int splits) {
+  typedef float f32x4_r __attribute__((ext_vector_type(4)));
+  const int64_t nv = n / 4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    f32x4_r acc = *reinterpret_cast<const f32x4_r*>(ws + i * 4);
+    for (int sp = 1; sp < splits; ++sp) {
+      const f32x4_r v = *reinterpret_cast<const f32x4_r*>(ws + (int64_t)sp * n + i * 4);
+      acc += v;
+    }
+    *reinterpret_cast<f32x4_r*>(out + i * 4) = acc;
+  }
 }
 
 }  // namespace
@@ -379,29 +398,46 @@ static torch::Tensor tn8p_run(torch::Tensor dz, torch::Tensor x, float* dbp) {
   if (splitm < 1) splitm = 1;
   int64_t chunk = ((M + splitm - 1) / splitm + BMS - 1) / BMS * BMS;
   splitm = (int)((M + chunk - 1) / chunk);
-  auto C = splitm > 1 ? torch::zeros({(int64_t)N, (int64_t)Kw},
-                                     dz.options().dtype(torch::kFloat32))
-                      : torch::empty({(int64_t)N, (int64_t)Kw},
-                                     dz.options().dtype(torch::kFloat32));
+  static const bool ws_mode = [] {
+    const char* e = getenv("JIMM_AMD_DW_WS");
+    return e && e[0] == '1';
+  }();
+  const bool use_ws = ws_mode && splitm > 1;
+  auto C = (splitm > 1 && !use_ws)
+               ? torch::zeros({(int64_t)N, (int64_t)Kw}, dz.options().dtype(torch::kFloat32))
+               : torch::empty({(int64_t)N, (int64_t)Kw}, dz.options().dtype(torch::kFloat32));
+  torch::Tensor ws;
+  if (use_ws)
+    ws = torch::empty({(int64_t)splitm, (int64_t)N, (int64_t)Kw},
+                      dz.options().dtype(torch::kFloat32));
   const size_t shmem = 4 * IMG_BYTES;  // 128 KiB
-#define LAUNCH_TN(AT, DBO)                                                                 \
+#define LAUNCH_TN(AT, DBO, WSO)                                                            \
   do {                                                                                     \
-    auto kfn = gemm_tn_8p_kernel<AT, DBO>;                                                 \
-    static bool attr_##AT##DBO = [&] {                                                     \
+    auto kfn = gemm_tn_8p_kernel<AT, DBO, WSO>;                                            \
+    static bool attr_##AT##DBO##WSO = [&] {                                                \
       hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),                              \
                           hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);         \
       return true;                                                                         \
     }();                                                                                   \
-    (void)attr_##AT##DBO;                                                                  \
+    (void)attr_##AT##DBO##WSO;                                                             \
     hipLaunchKernelGGL(kfn, dim3(tiles, splitm), dim3(NTHREADS), shmem, stream,            \
                        reinterpret_cast<const bf16*>(dz.data_ptr()),                       \
-                       reinterpret_cast<const bf16*>(x.data_ptr()), C.data_ptr<float>(),   \
-                       M, N, Kw, chunk, dbp);                                              \
+                       reinterpret_cast<const bf16*>(x.data_ptr()),                        \
+                       WSO ? ws.data_ptr<float>() : C.data_ptr<float>(), M, N, Kw, chunk,  \
+                       dbp);                                                               \
   } while (0)
-  if (dbp && splitm > 1) LAUNCH_TN(true, true);
-  else if (dbp) LAUNCH_TN(false, true);
-  else if (splitm > 1) LAUNCH_TN(true, false);
-  else LAUNCH_TN(false, false);
+  if (use_ws) {
+    if (dbp) LAUNCH_TN(false, true, true);
+    else LAUNCH_TN(false, false, true);
+    const int64_t n = (int64_t)N * Kw;
+    const int block = 256;
+    const int grid = (int)std::min<int64_t>((n / 4 + block - 1) / block, 8192);
+    hipLaunchKernelGGL(ws_reduce_kernel, dim3(grid), dim3(block), 0, stream,
+                       ws.data_ptr<float>(), C.data_ptr<float>(), n, splitm);
+  } else if (dbp && splitm > 1) LAUNCH_TN(true, true, false);
+  else if (dbp) LAUNCH_TN(false, true, false);
+  else if (splitm > 1) LAUNCH_TN(true, false, false);
+  else LAUNCH_TN(false, false, false);
 #undef LAUNCH_TN
   return C;
 }
