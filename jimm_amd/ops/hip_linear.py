@@ -80,12 +80,19 @@ def _deterministic() -> bool:
     return os.environ.get("JIMM_AMD_DETERMINISTIC", "0") == "1"
 
 
+def _dw_ws() -> bool:
+    # default workspace+fixed-order-reduce combine (deterministic);
+    # JIMM_AMD_DW_WS=0 selects the atomic combine (A/B only)
+    return os.environ.get("JIMM_AMD_DW_WS", "1") != "0"
+
+
 def _dw_gemm(ext, dz: torch.Tensor, x2: torch.Tensor, out_dtype) -> torch.Tensor:
-    """dW = dz^T @ x2 — in-house split-M TN kernel when supported (fp32
-    accumulate, atomic combine -> rocBLAS fallback in deterministic mode)."""
+    """dW = dz^T @ x2 — in-house split-M TN kernel (fp32 accumulate; the
+    default workspace combine is deterministic, so deterministic mode only
+    falls back when JIMM_AMD_DW_WS=0 forces the atomic combine)."""
     if (
         _gemm_mode() == "hip"
-        and not _deterministic()
+        and (_dw_ws() or not _deterministic())
         and dz.dtype == torch.bfloat16
         and ext.gemm_tn8p_supported(dz.shape[0], dz.shape[1], x2.shape[1])
     ):
