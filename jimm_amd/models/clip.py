@@ -16,7 +16,6 @@ import math
 import torch
 from torch import nn
 
-from jimm_amd import ops
 from jimm_amd.models.common.text import TextTransformer
 from jimm_amd.models.common.vit import VisionTransformerBase
 

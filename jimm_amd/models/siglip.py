@@ -15,7 +15,6 @@ from __future__ import annotations
 import torch
 from torch import nn
 
-from jimm_amd import ops
 from jimm_amd.models.common.text import TextTransformer
 from jimm_amd.models.common.vit import VisionTransformerBase
 

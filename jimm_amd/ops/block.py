@@ -22,7 +22,7 @@ import os
 import torch
 
 from jimm_amd.ops import _backend
-from jimm_amd.ops.hip_linear import _dw_db_gemm, _dw_gemm, _dx_gemm, _gemm_mode
+from jimm_amd.ops.hip_linear import _dw_db_gemm, _dx_gemm, _gemm_mode
 
 
 def _hip_gemms(M: int, H: int) -> bool:
