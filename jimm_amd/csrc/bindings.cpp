@@ -11,7 +11,6 @@ torch::Tensor bias_act_fwd(torch::Tensor z, c10::optional<torch::Tensor> bias, s
 std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, c10::optional<torch::Tensor> bias,
                                             std::string act, torch::Tensor scale8,
                                             torch::Tensor amax);
-torch::Tensor fp8_cast(torch::Tensor x, torch::Tensor scale8, torch::Tensor amax);
 std::vector<torch::Tensor> gemm_nt_8p_gradact_fp8(torch::Tensor dy, torch::Tensor wt,
                                                   torch::Tensor z, std::string act,
                                                   torch::Tensor scale8, torch::Tensor amax);
@@ -69,7 +68,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("addend") = py::none());
   m.def("bias_act_fwd", &bias_act_fwd, "fused bias+activation(+residual) forward");
   m.def("bias_act_fwd_fp8", &bias_act_fwd_fp8, "bias+act forward with fused e4m3 emit");
-  m.def("fp8_cast", &fp8_cast, "delayed-scaled e4m3 cast with fused amax");
   m.def("gemm_nt_8p_gradact_fp8", &gemm_nt_8p_gradact_fp8,
         "gradact dX GEMM with fused e4m3 emit of dz");
   m.def("layernorm_fwd_fp8", &layernorm_fwd_fp8, "LayerNorm forward with fused e4m3 emit");

@@ -79,29 +79,6 @@ __global__ void bias_act_vec_kernel(T* __restrict__ z, const T* __restrict__ bia
   }
 }
 
-// delayed-scaled e4m3 cast with fused amax collection (producer for the
-// fp8 dX GEMMs whose bf16 source is written by kernels we do not control,
-// e.g. the strided attention-backward dqkv buffer)
-__global__ void fp8_cast_kernel(const bf16* __restrict__ x, unsigned char* __restrict__ y8,
-                                const float* __restrict__ scale8,
-                                unsigned int* __restrict__ amax_bits, int64_t nv) {
-  const float rs8 = 1.f / scale8[0];
-  float tmax = 0.f;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    float xv[8];
-    vload_f32<8>(x + i * 8, xv);
-    unsigned short q8[4];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) tmax = fmaxf(tmax, fabsf(xv[j]));
-#pragma unroll
-    for (int j = 0; j < 8; j += 2) q8[j / 2] = cvt2_e4m3(xv[j] * rs8, xv[j + 1] * rs8);
-    *reinterpret_cast<uint2*>(y8 + i * 8) = *reinterpret_cast<uint2*>(q8);
-  }
-  tmax = wave_reduce_max(tmax);
-  if ((threadIdx.x % WAVE) == 0) atomicMax(amax_bits, __float_as_uint(tmax));
-}
-
 template <typename T>
 __global__ void act_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ z,
                                T* __restrict__ dz, int64_t n, int act) {
@@ -374,22 +351,6 @@ torch::Tensor colsum(torch::Tensor dz) {
     TORCH_CHECK(false, "colsum: unsupported dtype");
   }
   return ws.sum(0);  // (gy*rstep, N) fp32 reduce — tiny
-}
-
-torch::Tensor fp8_cast(torch::Tensor x, torch::Tensor scale8, torch::Tensor amax) {
-  // bf16 -> delayed-scaled e4m3 bytes + amax collection (one read, one write)
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == torch::kBFloat16);
-  TORCH_CHECK(x.numel() % 8 == 0);
-  auto y8 = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
-  const int64_t nv = x.numel() / 8;
-  const int block = 256;
-  const int grid = (int)std::min<int64_t>((nv + block - 1) / block, kMaxGrid);
-  auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(fp8_cast_kernel, dim3(grid), dim3(block), 0, stream,
-                     reinterpret_cast<const bf16*>(x.data_ptr()),
-                     y8.data_ptr<unsigned char>(), scale8.data_ptr<float>(),
-                     reinterpret_cast<unsigned int*>(amax.data_ptr()), nv);
-  return y8;
 }
 
 std::vector<torch::Tensor> bias_act_fwd_fp8(torch::Tensor z, c10::optional<torch::Tensor> bias,
