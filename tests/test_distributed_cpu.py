@@ -431,5 +431,31 @@ def _bucket_overlap_stress(rank, world):
             assert torch.allclose(p.grad, rg / world, atol=1e-4)
 
 
+def _comm_stats_world2(rank, world):
+    """Trainer.comm_stats surfaces the per-step collective traffic
+    (SURVEY §5 metrics): all-reduce bytes equal the full grad payload and
+    the contrastive gather bytes are nonzero for clip-task steps."""
+    import jimm_amd
+    from jimm_amd.train import SyntheticImageText, TrainConfig, Trainer
+
+    torch.manual_seed(0)
+    m = jimm_amd.CLIP(
+        embed_dim=16, image_resolution=32, vision_layers=1, vision_width=32,
+        vision_patch_size=16, context_length=8, vocab_size=64,
+        transformer_width=32, transformer_heads=2, transformer_layers=1,
+    ).float()
+    tr = Trainer(m, TrainConfig(task="clip", lr=1e-3))
+    data = SyntheticImageText(2, 32, 8, 64, torch.device("cpu"), dtype=torch.float32, seed=rank)
+    tr.train_step(next(iter(data)))
+    st = tr.comm_stats()
+    assert st["allreduce_bytes"] == st["grad_bytes_total"] > 0, st
+    assert st["allreduce_buckets"] >= 1
+    assert st["gather_bytes"] > 0, st  # image+text embedding all-gathers
+
+
+def test_comm_stats_world2():
+    spawn(_comm_stats_world2, port=29527)
+
+
 def test_bucket_overlap_stress():
     spawn(_bucket_overlap_stress, world=4, port=29520)
