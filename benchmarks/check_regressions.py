@@ -32,10 +32,10 @@ FLOORS_GEMM = {  # (M, N, K, act): fwd_tf
     (201728, 3072, 768, "gelu"): 600,
     (201728, 768, 3072, ""): 890,
 }
-FLOORS_DW = {  # (M, N, K): tf (post splitm-heuristic rework)
-    (201728, 2304, 768): 630,
-    (201728, 3072, 768): 660,
-    (201728, 768, 3072): 660,
+FLOORS_DW = {  # (M, N, K): tf (post splitm rework + workspace combine)
+    (201728, 2304, 768): 640,
+    (201728, 3072, 768): 730,
+    (201728, 768, 3072): 730,
 }
 
 
