@@ -281,6 +281,25 @@ def test_model_train_step_gpu():
     assert all(math.isfinite(l) for l in losses), losses
 
 
+def test_model_head_dim_72_gpu():
+    """A head_dim-72 model (SigLIP-so400m class: hidden % heads = 72) runs
+    the padded-DP flash kernels end to end: one train step, finite loss
+    and grads."""
+    from jimm_amd.train import SyntheticImages, TrainConfig, Trainer
+
+    torch.manual_seed(0)
+    m = jimm_amd.VisionTransformer(
+        num_classes=10, img_size=64, patch_size=16, num_layers=2,
+        hidden_size=144, num_heads=2, mlp_dim=512,
+    ).to(dev(), torch.bfloat16)
+    tr = Trainer(m, TrainConfig(task="vit", lr=1e-3))
+    data = SyntheticImages(8, 64, 10, dev(), dtype=torch.bfloat16)
+    out = tr.train_step(next(iter(data)))
+    assert math.isfinite(out["loss"].item())
+    for p in m.parameters():
+        assert torch.isfinite(p.grad if p.grad is not None else p).all()
+
+
 def test_gpu_vs_cpu_model_parity():
     """ViT-Tiny logits: GPU bf16 HIP path vs CPU fp32 reference path."""
     torch.manual_seed(0)
