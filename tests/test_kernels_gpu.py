@@ -290,8 +290,8 @@ def test_model_head_dim_72_gpu():
     torch.manual_seed(0)
     m = jimm_amd.VisionTransformer(
         num_classes=10, img_size=64, patch_size=16, num_layers=2,
-        hidden_size=144, num_heads=2, mlp_dim=512,
-    ).to(dev(), torch.bfloat16)
+        hidden_size=576, num_heads=8, mlp_dim=1024,
+    ).to(dev(), torch.bfloat16)  # head_dim = 72, H % 64 == 0 (LN constraint)
     tr = Trainer(m, TrainConfig(task="vit", lr=1e-3))
     data = SyntheticImages(8, 64, 10, dev(), dtype=torch.bfloat16)
     out = tr.train_step(next(iter(data)))
