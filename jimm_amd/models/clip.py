@@ -72,9 +72,13 @@ class CLIP(nn.Module):
         self.text_projection = nn.Linear(transformer_width, embed_dim, bias=False)
         self.logit_scale = nn.Parameter(torch.tensor(math.log(1 / 0.07)))
 
-    def gradient_checkpointing_enable(self) -> None:
+    def gradient_checkpointing_enable(self, every_n: int = 1) -> None:
+        """Recompute encoder blocks in backward; every_n > 1 checkpoints
+        only every n-th block (selective)."""
         self.vision_model.encoder.gradient_checkpointing = True
         self.text_model.encoder.gradient_checkpointing = True
+        self.vision_model.encoder.checkpoint_every = every_n
+        self.text_model.encoder.checkpoint_every = every_n
 
     def encode_image(self, images: torch.Tensor) -> torch.Tensor:
         from jimm_amd.parallel.tp import row_parallel_linear

@@ -146,12 +146,20 @@ class Encoder(nn.Module):
         if getattr(self, "gradient_checkpointing", False) and torch.is_grad_enabled():
             from torch.utils.checkpoint import checkpoint
 
-            for layer in self.layers:
-                # recompute the block's forward during backward: activation
-                # memory per block drops from ~16 to ~1 tensors of (B,L,H)
-                # (needed for e.g. SigLIP at 4096 pairs/GPU = the 32k-global
-                # BASELINE config on 8 GPUs; without it b4096 OOMs 288 GB)
-                x = checkpoint(layer, x, use_reentrant=False)
+            # recompute checkpointed blocks' forwards during backward:
+            # activation memory per block drops from ~16 to ~1 tensors of
+            # (B,L,H) (needed for e.g. SigLIP at 4096 pairs/GPU = the
+            # 32k-global BASELINE config on 8 GPUs; without it b4096 OOMs
+            # 288 GB). checkpoint_every = n > 1 is the selective variant:
+            # only every n-th block recomputes — 1/n of the recompute cost
+            # for ~(1 - 1/n) of the memory win's complement; pick n by how
+            # much headroom the batch leaves.
+            every = int(getattr(self, "checkpoint_every", 1) or 1)
+            for i, layer in enumerate(self.layers):
+                if i % every == 0:
+                    x = checkpoint(layer, x, use_reentrant=False)
+                else:
+                    x = layer(x)
             return x
         for layer in self.layers:
             x = layer(x)

@@ -68,11 +68,15 @@ class SigLIP(nn.Module):
         self.logit_scale = nn.Parameter(torch.tensor(1.0))
         self.logit_bias = nn.Parameter(torch.tensor(0.0))
 
-    def gradient_checkpointing_enable(self) -> None:
+    def gradient_checkpointing_enable(self, every_n: int = 1) -> None:
         """Recompute encoder blocks in backward (both towers) — trades ~35%
-        step time for ~10x less activation memory (huge-batch training)."""
+        step time for ~10x less activation memory (huge-batch training).
+        every_n > 1 checkpoints only every n-th block (selective: 1/n the
+        recompute cost when the batch leaves some memory headroom)."""
         self.vision_model.encoder.gradient_checkpointing = True
         self.text_model.encoder.gradient_checkpointing = True
+        self.vision_model.encoder.checkpoint_every = every_n
+        self.text_model.encoder.checkpoint_every = every_n
 
     def encode_image(self, images: torch.Tensor) -> torch.Tensor:
         return self.vision_model(images)  # no visual projection (siglip.py:123-133)

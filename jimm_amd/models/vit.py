@@ -56,8 +56,12 @@ class VisionTransformer(nn.Module):
         else:
             self.classifier = None
 
-    def gradient_checkpointing_enable(self) -> None:
+    def gradient_checkpointing_enable(self, every_n: int = 1) -> None:
+        """Recompute encoder blocks in backward; every_n > 1 checkpoints
+        only every n-th block (selective: 1/n the recompute cost for a
+        partial memory win)."""
         self.vision.encoder.gradient_checkpointing = True
+        self.vision.encoder.checkpoint_every = every_n
 
     def forward(self, images: torch.Tensor) -> torch.Tensor:
         x = self.vision(images)  # (B, H) CLS- or MAP-pooled

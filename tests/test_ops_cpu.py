@@ -139,26 +139,28 @@ def test_eval_step_contrastive():
 
 
 def test_gradient_checkpointing_matches():
-    """Checkpointed blocks give the same loss/grads as the plain path."""
+    """Checkpointed blocks (full and selective every-n) give the same
+    loss/grads as the plain path."""
     import jimm_amd
 
     torch.manual_seed(0)
-    def run(ckpt):
+    def run(ckpt, every_n=1):
         torch.manual_seed(1)
         m = jimm_amd.VisionTransformer(num_classes=5, img_size=32, patch_size=16,
-                                       num_layers=2, num_heads=2, mlp_dim=64, hidden_size=32)
+                                       num_layers=4, num_heads=2, mlp_dim=64, hidden_size=32)
         if ckpt:
-            m.gradient_checkpointing_enable()
+            m.gradient_checkpointing_enable(every_n=every_n)
         x = torch.randn(2, 3, 32, 32)
         out = m(x)
         out.square().sum().backward()
         return out.detach(), [p.grad.clone() for p in m.parameters()]
 
     o1, g1 = run(False)
-    o2, g2 = run(True)
-    assert torch.allclose(o1, o2, atol=1e-6)
-    for a, b in zip(g1, g2):
-        assert torch.allclose(a, b, atol=1e-5)
+    for every_n in (1, 2, 3):
+        o2, g2 = run(True, every_n)
+        assert torch.allclose(o1, o2, atol=1e-6), every_n
+        for a, b in zip(g1, g2):
+            assert torch.allclose(a, b, atol=1e-5), every_n
 
 
 def test_lr_schedule():
