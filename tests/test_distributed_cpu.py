@@ -457,5 +457,30 @@ def test_comm_stats_world2():
     spawn(_comm_stats_world2, port=29527)
 
 
+def test_bench_torchrun_world2():
+    """The driver's exact launch path: torchrun --nproc-per-node 2 bench.py
+    on CPU/gloo — rendezvous on 127.0.0.1, per-rank data shards, barrier +
+    max-over-ranks timing, exactly one JSON line from rank 0."""
+    import json
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29531", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batch", "2"],
+        capture_output=True, text=True, timeout=600,
+        cwd=os.path.join(os.path.dirname(__file__), ".."),
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, lines  # exactly one JSON line (rank 0)
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 2 and out["config"]["parallelism"] == "dp2"
+    assert out["config"]["global_batch"] == 4  # 2 ranks x per-rank batch 2
+    assert out["value"] > 0 and out["steps"] == 2
+
+
 def test_bucket_overlap_stress():
     spawn(_bucket_overlap_stress, world=4, port=29520)
