@@ -480,6 +480,15 @@ def test_bench_torchrun_world2():
     assert out["n_gpus"] == 2 and out["config"]["parallelism"] == "dp2"
     assert out["config"]["global_batch"] == 4  # 2 ranks x per-rank batch 2
     assert out["value"] > 0 and out["steps"] == 2
+    # full driver contract: every key the round harness parses
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in out, key
+    for key in ("model", "global_batch", "img_size", "parallelism"):
+        assert key in out["config"], key
+    assert out["scaling"] == "weak" and out["higher_is_better"] is True
+    assert out["data"] == "synthetic"
 
 
 def test_bucket_overlap_stress():
