@@ -390,7 +390,7 @@ __global__ __launch_bounds__(320) void attn_fwd_small_kernel(
   // The previous wave-per-(b,h) version looped the 5 strips serially in
   // one wave and topped out at 2 waves/SIMD — latency-exposed (68.8 TF/s
   // at CLIP-text L=77); strip-per-wave runs 5x the waves at the same LDS
-  // footprint per WG (12 KiB V + 5 x 3 KiB P = 27 KiB).
+  // footprint per WG (12 KiB V image + 5 x 2.77 KiB P tiles = 25.8 KiB).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
