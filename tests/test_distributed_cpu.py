@@ -294,6 +294,41 @@ def test_full_model_tp_siglip_world4():
     spawn(_check_full_model_tp_siglip, world=4, port=29518)
 
 
+def _check_full_model_tp_vit(rank, world):
+    from jimm_amd.parallel.tp import shard_vit
+
+    def make():
+        return jimm_amd.VisionTransformer(num_classes=7, img_size=32, patch_size=16,
+                                          num_layers=2, num_heads=4, mlp_dim=128,
+                                          hidden_size=64, pooling="MAP")
+
+    torch.manual_seed(0)
+    model = make()
+    ref = make()
+    ref.load_state_dict(model.state_dict())
+
+    g = torch.Generator().manual_seed(7)
+    imgs = torch.randn(2, 3, 32, 32, generator=g)
+
+    imgs_r = imgs.clone().requires_grad_(True)
+    lr = ref(imgs_r)
+    lr.sum().backward()
+
+    shard_vit(model, None)
+    imgs_t = imgs.clone().requires_grad_(True)
+    lt = model(imgs_t)
+    lt.sum().backward()
+
+    assert torch.allclose(lt, lr, atol=1e-4), (lt - lr).abs().max()
+    assert torch.allclose(imgs_t.grad, imgs_r.grad, atol=1e-4)
+
+
+def test_full_model_tp_vit_world4():
+    """ViT (MAP pooling) full-model TP vs the unsharded oracle — completes
+    the TP test matrix (CLIP/SigLIP covered above)."""
+    spawn(_check_full_model_tp_vit, world=4, port=29528)
+
+
 # ---------------------------------------------------------------------------
 def _check_trainer_dp_invariance(rank, world):
     """Full Trainer step at DP=2 matches a single-process run on the same
