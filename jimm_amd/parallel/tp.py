@@ -19,7 +19,8 @@ all-reduce bwd) before the column-parallel weights, `reduce_from_tp`
 
 SURVEY notes C2 is optional for parity (models fit easily in 288 GB HBM —
 DP-first); it is provided for the reference's pure tensor-sharded inference
-mode and validated by the gloo world-2 CPU test.
+mode and validated against the unsharded oracle at gloo world 2 and 4 for
+all three model families (tests/test_distributed_cpu.py).
 """
 
 from __future__ import annotations
